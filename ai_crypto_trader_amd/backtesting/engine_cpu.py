@@ -1,0 +1,248 @@
+"""CPU reference backtest engine (numpy, float32).
+
+Implements EXACTLY the per-candle state machine specified in strategy.py —
+it is the golden reference the HIP backtest kernel
+(ops/hip/backtest.hip) is validated against, and the engine behind the
+"CPU backtest_engine (plumbing, no GPU)" baseline config.
+
+Replaces the reference's per-candle loop (strategy_tester.py:190-300)
+with deterministic parameterized strategies instead of a per-candle OpenAI
+call, and computes indicators per candle position rather
+than once per window (the reference's known shortcut, strategy_tester.py:63-125).
+
+Vectorized over lanes = (param-set x symbol); loops over candles — the
+same loop order as one GPU lane.
+"""
+
+from __future__ import annotations
+
+import numpy as np
+
+from .strategy import FEE, MAX_WIN, NPARAM, WARMUP
+
+NMETRIC = 10
+METRIC_NAMES = [
+    "final_equity", "n_trades", "wins", "gross_profit", "gross_loss",
+    "max_drawdown", "sum_ret", "sum_ret2", "sharpe", "fitness",
+]
+
+ANNUALIZE = np.float32(np.sqrt(525_600.0))   # 1m candles per year
+EPS = np.float32(1e-9)
+
+
+def run_backtest_cpu(
+    candles: np.ndarray,       # (nsym, T, 4) f32 [close, high, low, volume]
+    population: np.ndarray,    # (P, NPARAM) f32
+    *,
+    initial_equity: float = 1.0,
+    record_equity: bool = False,
+):
+    """Run every param-set against every symbol.
+
+    Returns metrics (P, nsym, NMETRIC) f32; optionally the full equity
+    curves (P, nsym, T) f32 for plotting/validation.
+    """
+    f32 = np.float32
+    candles = np.asarray(candles, dtype=f32)
+    population = np.asarray(population, dtype=f32)
+    assert candles.ndim == 3 and candles.shape[2] == 4
+    assert population.ndim == 2 and population.shape[1] == NPARAM
+    nsym, T, _ = candles.shape
+    P = population.shape[0]
+    L = P * nsym
+
+    # Per-lane parameter broadcast: lane l = (param p, symbol s), s fastest.
+    par = np.repeat(population, nsym, axis=0)          # (L, NPARAM)
+    rsi_p = np.maximum(par[:, 0].astype(np.int32), 1)
+    rsi_os, rsi_ob = par[:, 1], par[:, 2]
+    a_f = (f32(2.0) / (par[:, 3] + f32(1.0))).astype(f32)
+    a_s = (f32(2.0) / (par[:, 4] + f32(1.0))).astype(f32)
+    a_sig = (f32(2.0) / (par[:, 5] + f32(1.0))).astype(f32)
+    bb_w = np.clip(par[:, 6].astype(np.int32), 2, MAX_WIN)
+    bb_k, bb_bth, bb_sth = par[:, 7], par[:, 8], par[:, 9]
+    entry_v = par[:, 10].astype(np.int32)
+    exit_v = par[:, 11].astype(np.int32)
+    size_pct, sl_pct, tp_pct = par[:, 12], par[:, 13], par[:, 14]
+    trail_pct, trail_act = par[:, 15], par[:, 16]
+
+    # Indicator state.
+    ema_f = np.empty(L, f32)
+    ema_s = np.empty(L, f32)
+    sig = np.zeros(L, f32)
+    avg_gain = np.zeros(L, f32)
+    avg_loss = np.zeros(L, f32)
+    ring = np.zeros((L, MAX_WIN), f32)
+    # f64 rolling sums: f32 incremental sums random-walk drift over ~1M
+    # candles and sum2/n - mean^2 cancels catastrophically; f32*f32 products
+    # are exact in f64, so these stay exact. The HIP kernel does the same.
+    bb_sum = np.zeros(L, np.float64)
+    bb_sum2 = np.zeros(L, np.float64)
+    inv_w = 1.0 / bb_w.astype(np.float64)
+    prev_close = np.zeros(L, f32)
+
+    # Trading state.
+    cash = np.full(L, initial_equity, f32)
+    units = np.zeros(L, f32)
+    in_pos = np.zeros(L, bool)
+    entry_cost = np.zeros(L, f32)
+    entry_price = np.zeros(L, f32)
+    stop = np.zeros(L, f32)
+    tp = np.zeros(L, f32)
+    peak = np.zeros(L, f32)
+    equity = np.full(L, initial_equity, f32)
+    max_eq = np.full(L, initial_equity, f32)
+    max_dd = np.zeros(L, f32)
+    n_trades = np.zeros(L, f32)
+    wins = np.zeros(L, f32)
+    gross_p = np.zeros(L, f32)
+    gross_l = np.zeros(L, f32)
+    sum_ret = np.zeros(L, f32)
+    sum_ret2 = np.zeros(L, f32)
+
+    curves = np.empty((L, T), f32) if record_equity else None
+
+    close_all = np.ascontiguousarray(candles[:, :, 0])   # (nsym, T)
+    high_all = np.ascontiguousarray(candles[:, :, 1])
+    low_all = np.ascontiguousarray(candles[:, :, 2])
+    sym_idx = np.tile(np.arange(nsym), P)                # lane -> symbol
+    lanes = np.arange(L)
+
+    for t in range(T):
+        close = close_all[sym_idx, t]
+        high = high_all[sym_idx, t]
+        low = low_all[sym_idx, t]
+
+        # --- 1. indicators -------------------------------------------------
+        if t == 0:
+            ema_f[:] = close
+            ema_s[:] = close
+            change = np.zeros(L, f32)
+        else:
+            ema_f += a_f * (close - ema_f)
+            ema_s += a_s * (close - ema_s)
+            change = close - prev_close
+        macd = ema_f - ema_s
+        sig += a_sig * (macd - sig)
+        macd_hist = macd - sig
+
+        gain = np.maximum(change, f32(0.0))
+        loss = np.maximum(-change, f32(0.0))
+        avg_gain += (gain - avg_gain) / rsi_p.astype(f32)
+        avg_loss += (loss - avg_loss) / rsi_p.astype(f32)
+        rsi = f32(100.0) - f32(100.0) / (
+            f32(1.0) + avg_gain / np.maximum(avg_loss, EPS)
+        )
+
+        ridx = t % bb_w                                   # per-lane ring slot
+        old = ring[lanes, ridx].astype(np.float64)
+        c64 = close.astype(np.float64)
+        bb_sum += c64 - old
+        bb_sum2 += c64 * c64 - old * old
+        ring[lanes, ridx] = close
+        inv_cnt = np.where(t + 1 < bb_w, 1.0 / (t + 1.0), inv_w)
+        mean64 = bb_sum * inv_cnt
+        var64 = np.maximum(bb_sum2 * inv_cnt - mean64 * mean64, 0.0)
+        mean = mean64.astype(f32)
+        std = np.sqrt(var64.astype(f32))
+        band = bb_k * std
+        bb_pos = (close - (mean - band)) / np.maximum(f32(2.0) * band, EPS)
+
+        prev_close = close
+
+        # --- 2. votes ------------------------------------------------------
+        if t >= WARMUP:
+            buy = (
+                (rsi < rsi_os).astype(np.int32)
+                + (macd_hist > 0).astype(np.int32)
+                + (bb_pos < bb_bth).astype(np.int32)
+            )
+            sell = (
+                (rsi > rsi_ob).astype(np.int32)
+                + (macd_hist < 0).astype(np.int32)
+                + (bb_pos > bb_sth).astype(np.int32)
+            )
+            net = buy - sell
+        else:
+            net = np.zeros(L, np.int32)
+
+        # --- 3. position management ---------------------------------------
+        pos = in_pos
+        peak = np.where(pos, np.maximum(peak, high), peak)
+        trail_on = pos & (trail_pct > 0) & (
+            peak >= entry_price * (f32(1.0) + trail_act)
+        )
+        stop = np.where(
+            trail_on, np.maximum(stop, peak * (f32(1.0) - trail_pct)), stop
+        )
+
+        hit_sl = pos & (low <= stop)
+        hit_tp = pos & ~hit_sl & (high >= tp)
+        hit_sig = pos & ~hit_sl & ~hit_tp & (net <= -exit_v)
+        exiting = hit_sl | hit_tp | hit_sig
+        exit_price = np.where(hit_sl, stop, np.where(hit_tp, tp, close))
+
+        proceeds = units * exit_price * (f32(1.0) - f32(FEE))
+        pnl = proceeds - entry_cost
+        cash = np.where(exiting, cash + proceeds, cash)
+        n_trades += exiting
+        wins += exiting & (pnl > 0)
+        gross_p += np.where(exiting, np.maximum(pnl, f32(0.0)), f32(0.0))
+        gross_l += np.where(exiting, np.maximum(-pnl, f32(0.0)), f32(0.0))
+        units = np.where(exiting, f32(0.0), units)
+        in_pos = pos & ~exiting
+
+        entering = (~pos) & (t >= WARMUP) & (net >= entry_v)
+        cost = np.minimum(size_pct * equity, cash)
+        new_units = cost * (f32(1.0) - f32(FEE)) / close
+        cash = np.where(entering, cash - cost, cash)
+        units = np.where(entering, new_units, units)
+        entry_cost = np.where(entering, cost, entry_cost)
+        entry_price = np.where(entering, close, entry_price)
+        stop = np.where(entering, close * (f32(1.0) - sl_pct), stop)
+        tp = np.where(entering, close * (f32(1.0) + tp_pct), tp)
+        peak = np.where(entering, close, peak)
+        in_pos = in_pos | entering
+
+        # --- 4. mark to market --------------------------------------------
+        new_eq = cash + units * close
+        r = new_eq / equity - f32(1.0)
+        sum_ret += r
+        sum_ret2 += r * r
+        equity = new_eq
+        max_eq = np.maximum(max_eq, equity)
+        max_dd = np.maximum(max_dd, (max_eq - equity) / max_eq)
+        if record_equity:
+            curves[:, t] = equity
+
+    metrics = finalize_metrics(
+        T, equity, n_trades, wins, gross_p, gross_l, max_dd, sum_ret, sum_ret2
+    ).reshape(P, nsym, NMETRIC)
+    if record_equity:
+        return metrics, curves.reshape(P, nsym, T)
+    return metrics
+
+
+def finalize_metrics(
+    T, equity, n_trades, wins, gross_p, gross_l, max_dd, sum_ret, sum_ret2
+) -> np.ndarray:
+    """Shared final-stats formula (strategy_evaluation.py:32-228 semantics:
+    annualized Sharpe from per-candle returns, win-rate, profit factor via
+    gross sums, running-peak drawdown). Fitness mirrors the GA objective of
+    strategy_evolution_service.py:525-694 (sharpe + win-rate - drawdown
+    penalty; no-trade strategies are penalized)."""
+    f32 = np.float32
+    n = f32(max(T, 1))
+    mean = sum_ret / n
+    var = np.maximum(sum_ret2 / n - mean * mean, f32(0.0))
+    sharpe = mean / np.maximum(np.sqrt(var), EPS) * ANNUALIZE
+    sharpe = np.where(n_trades > 0, sharpe, f32(0.0))
+    win_rate = wins / np.maximum(n_trades, f32(1.0))
+    fitness = np.where(
+        n_trades > 0,
+        sharpe + win_rate - f32(2.0) * max_dd,
+        f32(-1.0),
+    )
+    return np.stack(
+        [equity, n_trades, wins, gross_p, gross_l, max_dd, sum_ret,
+         sum_ret2, sharpe.astype(f32), fitness.astype(f32)], axis=-1
+    ).astype(f32)

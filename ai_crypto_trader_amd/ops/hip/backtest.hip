@@ -1,0 +1,247 @@
+// GPU backtest engine — the headline kernel.
+//
+// Replaces the reference's per-candle Python loop
+// (backtesting/strategy_tester.py:190-300, services/strategy_evaluation.py:777-878)
+// with one CDNA4 lane per (param-set x symbol) marching candles:
+//   - a block = 256 lanes = 256 param-sets of one symbol
+//   - candle tiles ([close,high,low,volume] f32x4) staged in LDS, read
+//     broadcast by all lanes (conflict-free: same address)
+//   - Bollinger rolling window lives in an LDS ring (stride MAX_WIN+1 so
+//     lane accesses spread across banks)
+//   - EMA/MACD/Wilder-RSI are O(1) register recurrences per candle
+//   - the position state machine (SL/TP/trailing/vote exits — semantics of
+//     trade_executor_service.py:55-399 + binance_ml_strategy.py:489-543) is
+//     branchless selects, matching backtesting/engine_cpu.py bit-for-bit
+//     modulo FMA contraction (disabled here via pragma for golden tests).
+//
+// Grid: nsym * ceil(P/256) blocks, XCD-affine map keeps the blocks of one
+// symbol on one XCD so their shared candle stream stays in that XCD's L2.
+
+#include "common.hpp"
+
+#define BT_BLOCK 256
+#define BT_TILE 256
+#define BT_MAXWIN 32           // == strategy.py MAX_WIN
+#define BT_NPARAM 17
+#define BT_NMETRIC 10
+#define BT_WARMUP 128          // == strategy.py WARMUP
+#define BT_FEE 0.001f
+#define BT_EPS 1e-9f
+#define BT_ANNUALIZE 724.9827573f   // float32(sqrt(525600))
+
+namespace {
+
+struct LaneParams {
+    float rsi_p, rsi_os, rsi_ob;
+    float a_f, a_s, a_sig;
+    int bb_w;
+    float bb_k, bb_bth, bb_sth;
+    int entry_v, exit_v;
+    float size_pct, sl_pct, tp_pct, trail_pct, trail_act;
+};
+
+__global__ void __launch_bounds__(BT_BLOCK) backtest_kernel(
+    const float* __restrict__ candles,   // (nsym, T, 4)
+    const float* __restrict__ pop,       // (P, NPARAM)
+    float* __restrict__ metrics,         // (P, nsym, NMETRIC)
+    int nsym, int T, int P, int chunks_per_sym, float initial_equity)
+{
+#pragma clang fp contract(off)           // match the numpy f32 reference
+    __shared__ float tile[BT_TILE][4];
+    __shared__ float ring[BT_BLOCK * (BT_MAXWIN + 1)];
+
+    // block -> (symbol, param chunk); same-symbol blocks share an XCD when
+    // the shape allows (dispatcher places block b on XCD b%8).
+    int bid = blockIdx.x;
+    int sym, chunk;
+    int nblocks = nsym * chunks_per_sym;
+    if ((nsym & 7) == 0 && (nblocks & 7) == 0) {
+        int xcd = bid & 7, j = bid >> 3;
+        sym = xcd + 8 * (j / chunks_per_sym);
+        chunk = j % chunks_per_sym;
+    } else {
+        sym = bid / chunks_per_sym;
+        chunk = bid % chunks_per_sym;
+    }
+
+    const int tid = threadIdx.x;
+    const int p = chunk * BT_BLOCK + tid;
+    const bool active = p < P;
+    const int pl = active ? p : 0;       // inactive lanes shadow lane 0
+
+    // ---- load + derive params ------------------------------------------
+    LaneParams q;
+    {
+        const float* pr = pop + (long)pl * BT_NPARAM;
+        q.rsi_p = pr[0]; q.rsi_os = pr[1]; q.rsi_ob = pr[2];
+        q.a_f = 2.0f / (pr[3] + 1.0f);
+        q.a_s = 2.0f / (pr[4] + 1.0f);
+        q.a_sig = 2.0f / (pr[5] + 1.0f);
+        q.bb_w = min(max((int)pr[6], 2), BT_MAXWIN);
+        q.bb_k = pr[7]; q.bb_bth = pr[8]; q.bb_sth = pr[9];
+        q.entry_v = (int)pr[10]; q.exit_v = (int)pr[11];
+        q.size_pct = pr[12]; q.sl_pct = pr[13]; q.tp_pct = pr[14];
+        q.trail_pct = pr[15]; q.trail_act = pr[16];
+        q.rsi_p = fmaxf(floorf(q.rsi_p), 1.0f);
+    }
+    float* myring = ring + tid * (BT_MAXWIN + 1);
+    for (int i = 0; i <= BT_MAXWIN; ++i) myring[i] = 0.0f;
+
+    // ---- state ----------------------------------------------------------
+    float ema_f = 0.f, ema_s = 0.f, sig = 0.f;
+    float avg_gain = 0.f, avg_loss = 0.f;
+    // f64 rolling sums (engine_cpu.py rationale: f32 drifts + cancels)
+    double bb_sum = 0.0, bb_sum2 = 0.0;
+    const double inv_w = 1.0 / (double)q.bb_w;
+    float prev_close = 0.f;
+
+    float cash = initial_equity, units = 0.f;
+    bool in_pos = false;
+    float entry_cost = 0.f, entry_price = 0.f;
+    float stop = 0.f, tp = 0.f, peak = 0.f;
+    float equity = initial_equity, max_eq = initial_equity, max_dd = 0.f;
+    float n_trades = 0.f, wins = 0.f, gross_p = 0.f, gross_l = 0.f;
+    float sum_ret = 0.f, sum_ret2 = 0.f;
+
+    const float4* sym_candles =
+        reinterpret_cast<const float4*>(candles + (long)sym * T * 4);
+
+    for (int t0 = 0; t0 < T; t0 += BT_TILE) {
+        __syncthreads();
+        if (t0 + tid < T) {
+            float4 c = sym_candles[t0 + tid];
+            tile[tid][0] = c.x;   // close
+            tile[tid][1] = c.y;   // high
+            tile[tid][2] = c.z;   // low
+            tile[tid][3] = c.w;   // volume
+        }
+        __syncthreads();
+        const int tend = min(BT_TILE, T - t0);
+        for (int tt = 0; tt < tend; ++tt) {
+            const int t = t0 + tt;
+            const float close = tile[tt][0];
+            const float high = tile[tt][1];
+            const float low = tile[tt][2];
+
+            // --- 1. indicators ---------------------------------------
+            float change;
+            if (t == 0) {
+                ema_f = close; ema_s = close; change = 0.0f;
+            } else {
+                ema_f += q.a_f * (close - ema_f);
+                ema_s += q.a_s * (close - ema_s);
+                change = close - prev_close;
+            }
+            float macd = ema_f - ema_s;
+            sig += q.a_sig * (macd - sig);
+            float macd_hist = macd - sig;
+
+            float gain = fmaxf(change, 0.0f);
+            float loss = fmaxf(-change, 0.0f);
+            avg_gain += (gain - avg_gain) / q.rsi_p;
+            avg_loss += (loss - avg_loss) / q.rsi_p;
+            float rsi =
+                100.0f - 100.0f / (1.0f + avg_gain / fmaxf(avg_loss, BT_EPS));
+
+            int ridx = t % q.bb_w;
+            double old = (double)myring[ridx];
+            double c64 = (double)close;
+            bb_sum += c64 - old;
+            bb_sum2 += c64 * c64 - old * old;
+            myring[ridx] = close;
+            double inv_cnt = (t + 1 < q.bb_w) ? 1.0 / (t + 1.0) : inv_w;
+            double mean64 = bb_sum * inv_cnt;
+            double var64 = fmax(bb_sum2 * inv_cnt - mean64 * mean64, 0.0);
+            float mean = (float)mean64;
+            float std_ = sqrtf((float)var64);
+            float band = q.bb_k * std_;
+            float bb_pos =
+                (close - (mean - band)) / fmaxf(2.0f * band, BT_EPS);
+
+            prev_close = close;
+
+            // --- 2. votes --------------------------------------------
+            int net = 0;
+            if (t >= BT_WARMUP) {
+                int buy = (rsi < q.rsi_os) + (macd_hist > 0.0f) +
+                          (bb_pos < q.bb_bth);
+                int sell = (rsi > q.rsi_ob) + (macd_hist < 0.0f) +
+                           (bb_pos > q.bb_sth);
+                net = buy - sell;
+            }
+
+            // --- 3. position management ------------------------------
+            if (in_pos) {
+                peak = fmaxf(peak, high);
+                bool trail_on = (q.trail_pct > 0.0f) &&
+                                (peak >= entry_price * (1.0f + q.trail_act));
+                if (trail_on)
+                    stop = fmaxf(stop, peak * (1.0f - q.trail_pct));
+                bool hit_sl = low <= stop;
+                bool hit_tp = !hit_sl && high >= tp;
+                bool hit_sig = !hit_sl && !hit_tp && net <= -q.exit_v;
+                if (hit_sl || hit_tp || hit_sig) {
+                    float exit_price = hit_sl ? stop : (hit_tp ? tp : close);
+                    float proceeds = units * exit_price * (1.0f - BT_FEE);
+                    float pnl = proceeds - entry_cost;
+                    cash += proceeds;
+                    n_trades += 1.0f;
+                    wins += (pnl > 0.0f) ? 1.0f : 0.0f;
+                    gross_p += fmaxf(pnl, 0.0f);
+                    gross_l += fmaxf(-pnl, 0.0f);
+                    units = 0.0f;
+                    in_pos = false;
+                }
+            } else if (t >= BT_WARMUP && net >= q.entry_v) {
+                float cost = fminf(q.size_pct * equity, cash);
+                units = cost * (1.0f - BT_FEE) / close;
+                cash -= cost;
+                entry_cost = cost;
+                entry_price = close;
+                stop = close * (1.0f - q.sl_pct);
+                tp = close * (1.0f + q.tp_pct);
+                peak = close;
+                in_pos = true;
+            }
+
+            // --- 4. mark to market -----------------------------------
+            float new_eq = cash + units * close;
+            float r = new_eq / equity - 1.0f;
+            sum_ret += r;
+            sum_ret2 += r * r;
+            equity = new_eq;
+            max_eq = fmaxf(max_eq, equity);
+            max_dd = fmaxf(max_dd, (max_eq - equity) / max_eq);
+        }
+    }
+
+    if (!active) return;
+
+    // ---- finalize (engine_cpu.finalize_metrics semantics) ---------------
+    float n = (float)max(T, 1);
+    float mean_r = sum_ret / n;
+    float var_r = fmaxf(sum_ret2 / n - mean_r * mean_r, 0.0f);
+    float sharpe = mean_r / fmaxf(sqrtf(var_r), BT_EPS) * BT_ANNUALIZE;
+    if (!(n_trades > 0.0f)) sharpe = 0.0f;
+    float win_rate = wins / fmaxf(n_trades, 1.0f);
+    float fitness = (n_trades > 0.0f)
+                        ? sharpe + win_rate - 2.0f * max_dd
+                        : -1.0f;
+
+    float* out = metrics + ((long)p * nsym + sym) * BT_NMETRIC;
+    out[0] = equity; out[1] = n_trades; out[2] = wins;
+    out[3] = gross_p; out[4] = gross_l; out[5] = max_dd;
+    out[6] = sum_ret; out[7] = sum_ret2; out[8] = sharpe; out[9] = fitness;
+}
+
+}  // namespace
+
+extern "C" void launch_backtest(const float* candles, const float* pop,
+                                float* metrics, int nsym, int T, int P,
+                                float initial_equity, hipStream_t stream) {
+    int chunks = (P + BT_BLOCK - 1) / BT_BLOCK;
+    dim3 grid(nsym * chunks);
+    hipLaunchKernelGGL(backtest_kernel, grid, dim3(BT_BLOCK), 0, stream,
+                       candles, pop, metrics, nsym, T, P, chunks,
+                       initial_equity);
+}

@@ -1,0 +1,138 @@
+"""Rolling technical indicators: GPU kernel wrapper + CPU golden reference.
+
+CPU reference implements the exact recurrences of
+ops/hip/indicators.hip sequentially over the full stream (no chunking),
+which equals the chunked GPU output because the EMA/Wilder warmup error
+(1-alpha)^512 underflows f32 (see kernel header). Indicator formulas follow
+the reference's TechnicalAnalyzer (binance_ml_strategy.py:40-182).
+"""
+
+from __future__ import annotations
+
+import numpy as np
+
+from . import require_hip_ops
+
+NIND = 13
+IND_NAMES = [
+    "ema12", "ema26", "macd", "macd_signal", "macd_hist", "rsi14",
+    "bb_mid", "bb_up", "bb_lo", "atr14", "stoch_k", "williams_r", "vwap20",
+]
+
+
+def indicators_cpu(candles: np.ndarray) -> np.ndarray:
+    """(nsym, T, 4) [close, high, low, vol] f32 -> (nsym, T, NIND) f32."""
+    f32 = np.float32
+    candles = np.asarray(candles, dtype=f32)
+    nsym, T, _ = candles.shape
+    out = np.zeros((nsym, T, NIND), dtype=f32)
+    a12, a26, a9 = f32(2.0 / 13.0), f32(2.0 / 27.0), f32(2.0 / 10.0)
+    W, S = 20, 14
+
+    close = candles[:, :, 0]
+    high = candles[:, :, 1]
+    low = candles[:, :, 2]
+    vol = candles[:, :, 3]
+
+    ema12 = close[:, 0].copy()
+    ema26 = close[:, 0].copy()
+    sig = np.zeros(nsym, f32)
+    avg_gain = np.zeros(nsym, f32)
+    avg_loss = np.zeros(nsym, f32)
+    atr = np.zeros(nsym, f32)
+    bb_sum = np.zeros(nsym, np.float64)    # f64: see engine_cpu.py rationale
+    bb_sum2 = np.zeros(nsym, np.float64)
+    pv_sum = np.zeros(nsym, f32)
+    vol_sum = np.zeros(nsym, f32)
+    ring_c = np.zeros((nsym, W), f32)
+    ring_h = np.zeros((nsym, S), f32)
+    ring_l = np.zeros((nsym, S), f32)
+    ring_pv = np.zeros((nsym, W), f32)
+    ring_v = np.zeros((nsym, W), f32)
+    prev_close = close[:, 0].copy()
+
+    for t in range(T):
+        c, h, lo_, v = close[:, t], high[:, t], low[:, t], vol[:, t]
+        if t == 0:
+            change = np.zeros(nsym, f32)
+        else:
+            ema12 += a12 * (c - ema12)
+            ema26 += a26 * (c - ema26)
+            change = c - prev_close
+        macd = ema12 - ema26
+        sig += a9 * (macd - sig)
+
+        gain = np.maximum(change, f32(0))
+        loss = np.maximum(-change, f32(0))
+        avg_gain += (gain - avg_gain) / f32(14.0)
+        avg_loss += (loss - avg_loss) / f32(14.0)
+        rsi = f32(100.0) - f32(100.0) / (
+            f32(1.0) + avg_gain / np.maximum(avg_loss, f32(1e-9))
+        )
+        tr = np.maximum(
+            h - lo_, np.maximum(np.abs(h - prev_close),
+                                np.abs(lo_ - prev_close))
+        )
+        atr += (tr - atr) / f32(14.0)
+        prev_close = c
+
+        ri = t % W
+        old = ring_c[:, ri].astype(np.float64)
+        c64 = c.astype(np.float64)
+        bb_sum += c64 - old
+        bb_sum2 += c64 * c64 - old * old
+        ring_c[:, ri] = c
+        bcnt = float(min(t + 1, W))
+        mean64 = bb_sum / bcnt
+        var64 = np.maximum(bb_sum2 / bcnt - mean64 * mean64, 0.0)
+        mean = mean64.astype(f32)
+        sd = np.sqrt(var64.astype(f32))
+
+        rs = t % S
+        ring_h[:, rs] = h
+        ring_l[:, rs] = lo_
+        scnt = min(t + 1, S)
+        hmax = ring_h[:, :scnt].max(axis=1)
+        lmin = ring_l[:, :scnt].min(axis=1)
+        rng = np.maximum(hmax - lmin, f32(1e-9))
+        stoch = (c - lmin) / rng * f32(100.0)
+        williams = f32(-100.0) * (hmax - c) / rng
+
+        tp = (h + lo_ + c) * f32(1.0 / 3.0)
+        pv = tp * v
+        pv_sum += pv - ring_pv[:, ri]
+        vol_sum += v - ring_v[:, ri]
+        ring_pv[:, ri] = pv
+        ring_v[:, ri] = v
+        vwap = pv_sum / np.maximum(vol_sum, f32(1e-9))
+
+        out[:, t, 0] = ema12
+        out[:, t, 1] = ema26
+        out[:, t, 2] = macd
+        out[:, t, 3] = sig
+        out[:, t, 4] = macd - sig
+        out[:, t, 5] = rsi
+        out[:, t, 6] = mean
+        out[:, t, 7] = mean + f32(2.0) * sd
+        out[:, t, 8] = mean - f32(2.0) * sd
+        out[:, t, 9] = atr
+        out[:, t, 10] = stoch
+        out[:, t, 11] = williams
+        out[:, t, 12] = vwap
+    return out
+
+
+def indicators_gpu(candles) -> "torch.Tensor":
+    """(nsym, T, 4) f32 cuda -> (nsym, T, NIND) f32 cuda."""
+    import torch
+
+    ops = require_hip_ops()
+    assert candles.is_cuda and candles.dtype == torch.float32
+    candles = candles.contiguous()
+    nsym, T, _ = candles.shape
+    out = torch.empty(
+        (nsym, T, NIND), dtype=torch.float32, device=candles.device
+    )
+    stream = torch.cuda.current_stream(candles.device).cuda_stream
+    ops.indicators(candles.data_ptr(), out.data_ptr(), nsym, T, NIND, stream)
+    return out

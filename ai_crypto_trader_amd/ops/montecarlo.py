@@ -1,0 +1,175 @@
+"""Monte-Carlo correlated-GBM ops: GPU wrapper + CPU golden reference.
+
+Replaces monte_carlo_service.py:197-336: GBM paths, percentiles, VaR/CVaR,
+probability of profit, per-path max drawdown. The CPU reference
+reimplements the exact Philox4x32-10 + Box-Muller stream of the kernel in
+vectorized numpy so small-scale tests are bit-comparable (modulo
+transcendental rounding), and an analytic-moment test covers the large-N
+statistics.
+"""
+
+from __future__ import annotations
+
+import numpy as np
+
+from . import require_hip_ops
+
+# --- Philox4x32-10 in numpy (matches ops/hip/common.hpp) -------------------
+
+_M0 = np.uint64(0xD2511F53)
+_M1 = np.uint64(0xCD9E8D57)
+_W0 = np.uint32(0x9E3779B9)
+_W1 = np.uint32(0xBB67AE85)
+
+
+def philox4x32_np(seed: int, ctr_lo: np.ndarray, ctr_hi: np.ndarray):
+    """Vectorized Philox4x32-10. ctr_lo/ctr_hi: uint64 arrays."""
+    u32 = np.uint32
+    u64 = np.uint64
+    k0 = u32(seed & 0xFFFFFFFF)
+    k1 = u32((seed >> 32) & 0xFFFFFFFF)
+    ctr_lo = ctr_lo.astype(u64)
+    ctr_hi = ctr_hi.astype(u64)
+    c0 = ctr_lo.astype(u32)
+    c1 = (ctr_lo >> u64(32)).astype(u32)
+    c2 = ctr_hi.astype(u32)
+    c3 = (ctr_hi >> u64(32)).astype(u32)
+    k0 = np.full_like(c0, k0)
+    k1 = np.full_like(c0, k1)
+    for _ in range(10):
+        p0 = _M0 * c0.astype(u64)
+        p1 = _M1 * c2.astype(u64)
+        hi0 = (p0 >> u64(32)).astype(u32)
+        lo0 = p0.astype(u32)
+        hi1 = (p1 >> u64(32)).astype(u32)
+        lo1 = p1.astype(u32)
+        n0 = hi1 ^ c1 ^ k0
+        n1 = lo1
+        n2 = hi0 ^ c3 ^ k1
+        n3 = lo0
+        c0, c1, c2, c3 = n0, n1, n2, n3
+        k0 = k0 + _W0
+        k1 = k1 + _W1
+    return c0, c1, c2, c3
+
+
+def _u32_to_unit(v: np.ndarray) -> np.ndarray:
+    # matches device: ((float)v + 1.0f) * 2.3283064e-10f  (f32 throughout)
+    return (v.astype(np.float32) + np.float32(1.0)) * np.float32(2.3283064e-10)
+
+
+def philox_normal4_np(seed: int, ctr_lo: np.ndarray, ctr_hi: np.ndarray):
+    """Matches device philox_normal4: returns (..., 4) float32 normals."""
+    a, b, c, d = philox4x32_np(seed, ctr_lo, ctr_hi)
+
+    def bm(x, y):
+        u1 = _u32_to_unit(x).astype(np.float32)
+        u2 = _u32_to_unit(y).astype(np.float32)
+        r = np.sqrt(np.float32(-2.0) * np.log(u1))
+        ang = np.float32(2.0 * np.pi) * u2
+        return r * np.cos(ang), r * np.sin(ang)
+
+    z0, z1 = bm(a, b)
+    z2, z3 = bm(c, d)
+    return np.stack([z0, z1, z2, z3], axis=-1).astype(np.float32)
+
+
+# --- CPU reference path generator -----------------------------------------
+
+def mc_paths_cpu(
+    chol: np.ndarray,        # (A, A) Cholesky factor of the correlation
+    mu: np.ndarray,          # (A,) annualized drift
+    sigma: np.ndarray,       # (A,) annualized vol
+    weights: np.ndarray,     # (A,) portfolio weights (sum to 1)
+    *,
+    n_steps: int,
+    n_paths: int,
+    dt: float,
+    s0: float = 1.0,
+    seed: int = 0,
+):
+    """Reference for mc_paths_kernel: same Philox stream, same update
+    order. Returns (final_value, max_dd) each (n_paths,) f32."""
+    A = chol.shape[0]
+    f32 = np.float32
+    cvol = (sigma[:, None] * chol * np.sqrt(dt)).astype(f32)   # (a, k)
+    cvol_k_major = np.ascontiguousarray(cvol.T)                # (k, a)
+    drift = ((mu - 0.5 * sigma**2) * dt).astype(f32)
+    wS0 = (weights * s0).astype(f32)
+
+    logS = np.zeros((n_paths, A), f32)
+    v0 = f32(weights.sum() * s0)
+    vmax = np.full(n_paths, v0, f32)
+    mdd = np.zeros(n_paths, f32)
+    V = np.full(n_paths, v0, f32)
+    paths = np.arange(n_paths, dtype=np.uint64)
+    for step in range(n_steps):
+        for k4 in range(A // 4):
+            ctr_hi = np.full(
+                n_paths, (np.uint64(step) << np.uint64(32)) | np.uint64(k4),
+                dtype=np.uint64,
+            )
+            z4 = philox_normal4_np(seed, paths, ctr_hi)        # (P, 4)
+            for dz in range(4):
+                k = k4 * 4 + dz
+                logS += np.outer(z4[:, dz], cvol_k_major[k])
+        logS += drift
+        V = (wS0 * np.exp(logS)).sum(axis=1).astype(f32)
+        vmax = np.maximum(vmax, V)
+        mdd = np.maximum(mdd, (vmax - V) / vmax)
+    return V, mdd
+
+
+def mc_paths_gpu(
+    chol, mu, sigma, weights, *, n_steps: int, n_paths: int, dt: float,
+    s0: float = 1.0, seed: int = 0, device="cuda",
+):
+    """GPU path generation; returns (final_value, max_dd) torch tensors."""
+    import torch
+
+    ops = require_hip_ops()
+    A = int(chol.shape[0])
+    f32 = np.float32
+    cvol = (np.asarray(sigma)[:, None] * np.asarray(chol) * np.sqrt(dt))
+    cvol_k_major = np.ascontiguousarray(cvol.T, dtype=f32)
+    drift = ((np.asarray(mu) - 0.5 * np.asarray(sigma) ** 2) * dt).astype(f32)
+    wS0 = (np.asarray(weights) * s0).astype(f32)
+    v0 = float(wS0.sum())
+
+    t_cvol = torch.from_numpy(cvol_k_major).to(device)
+    t_drift = torch.from_numpy(drift).to(device)
+    t_w = torch.from_numpy(wS0).to(device)
+    fv = torch.empty(n_paths, dtype=torch.float32, device=device)
+    dd = torch.empty(n_paths, dtype=torch.float32, device=device)
+    stream = torch.cuda.current_stream(fv.device).cuda_stream
+    ops.mc_paths(
+        t_cvol.data_ptr(), t_drift.data_ptr(), t_w.data_ptr(), 0,
+        fv.data_ptr(), dd.data_ptr(), A, n_steps, n_paths, v0, seed, stream,
+    )
+    return fv, dd
+
+
+def risk_stats(final_values, v0: float, confidences=(0.95, 0.99)):
+    """VaR/CVaR/percentiles/prob-profit from per-path final values
+    (monte_carlo_service.py:304-325 semantics). Works on numpy arrays or
+    torch tensors (stays on device for torch)."""
+    import torch
+
+    is_torch = isinstance(final_values, torch.Tensor)
+    fv = final_values
+    n = fv.shape[0]
+    sorted_fv = torch.sort(fv).values if is_torch else np.sort(fv)
+    out = {}
+    for conf in confidences:
+        idx = int((1.0 - conf) * n)
+        var_level = sorted_fv[idx]
+        tail = sorted_fv[: max(idx, 1)]
+        cvar_level = tail.mean()
+        out[f"var_{int(conf * 100)}"] = float(v0 - var_level)
+        out[f"cvar_{int(conf * 100)}"] = float(v0 - cvar_level)
+    for pct in (5, 25, 50, 75, 95):
+        idx = min(n - 1, int(pct / 100.0 * n))
+        out[f"p{pct}"] = float(sorted_fv[idx])
+    out["mean"] = float(fv.mean())
+    out["prob_profit"] = float((fv > v0).sum() / n)
+    return out

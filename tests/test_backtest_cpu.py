@@ -1,0 +1,93 @@
+"""CPU backtest engine behavior tests (the golden reference itself)."""
+
+import numpy as np
+import pytest
+
+from ai_crypto_trader_amd.backtesting.engine_cpu import (
+    METRIC_NAMES, NMETRIC, run_backtest_cpu,
+)
+from ai_crypto_trader_amd.backtesting.strategy import (
+    DEFAULT_PARAMS, NPARAM, WARMUP, clip_params, dict_to_params,
+    random_population,
+)
+from ai_crypto_trader_amd.data.synthetic import candles_chl_v, generate_ohlcv
+
+
+def test_shapes(small_market):
+    pop = random_population(8, seed=3)
+    m = run_backtest_cpu(small_market, pop)
+    assert m.shape == (8, small_market.shape[0], NMETRIC)
+    assert np.isfinite(m).all()
+
+
+def test_deterministic(small_market):
+    pop = random_population(4, seed=3)
+    m1 = run_backtest_cpu(small_market, pop)
+    m2 = run_backtest_cpu(small_market, pop)
+    np.testing.assert_array_equal(m1, m2)
+
+
+def test_no_trades_before_warmup():
+    # a market that ends right at warmup must produce zero trades
+    c = candles_chl_v(generate_ohlcv(WARMUP, 1, seed=0))
+    m = run_backtest_cpu(c, random_population(16, seed=1))
+    assert (m[..., 1] == 0).all()
+    assert (m[..., 9] == -1.0).all()      # no-trade fitness penalty
+
+
+def test_equity_conservation(small_market):
+    """Equity only changes via fees and price moves: a strategy that never
+    trades keeps equity exactly 1.0."""
+    p = dict_to_params({"entry_votes": 3, "rsi_oversold": 5.0,
+                        "bb_buy_th": 0.0})
+    # entry requires 3 net buy votes with impossible thresholds -> no trades
+    p = clip_params(p[None, :])
+    m = run_backtest_cpu(small_market, p)
+    no_trade = m[..., 1] == 0
+    assert (m[..., 0][no_trade] == 1.0).all()
+
+
+def test_stop_loss_bounds_loss():
+    """With a tight stop-loss, no single trade loses much more than
+    sl_pct + 2*fee (modulo gap-through-stop fills)."""
+    ohlcv = generate_ohlcv(5000, 1, seed=11, sigma=1.5)
+    c = candles_chl_v(ohlcv)
+    p = dict_to_params({"stop_loss_pct": 0.01, "take_profit_pct": 0.4,
+                        "entry_votes": 1, "position_size_pct": 1.0})
+    m, curves = run_backtest_cpu(c, clip_params(p[None, :]),
+                                 record_equity=True)
+    eq = curves[0, 0]
+    rel_drop = np.diff(eq) / eq[:-1]
+    # single-candle loss bounded by stop distance + fees + wick slack
+    assert rel_drop.min() > -0.08
+
+
+def test_fees_reduce_equity():
+    """Round-trip trades pay 2x fee: heavy trading on a flat market must
+    strictly lose money."""
+    rng = np.random.default_rng(5)
+    T = 2000
+    close = 1.0 + 0.001 * rng.standard_normal(T).astype(np.float32)
+    candles = np.stack(
+        [close, close * 1.0005, close * 0.9995, np.ones_like(close)], axis=-1
+    )[None].astype(np.float32)
+    p = dict_to_params({"entry_votes": 1, "exit_votes": 1,
+                        "position_size_pct": 1.0})
+    m = run_backtest_cpu(candles, clip_params(p[None, :]))
+    if m[0, 0, 1] > 3:       # if it traded a lot, fees must show
+        assert m[0, 0, 0] < 1.0
+
+
+def test_metric_consistency(small_market):
+    pop = random_population(32, seed=9)
+    m = run_backtest_cpu(small_market, pop)
+    wins, trades = m[..., 2], m[..., 1]
+    assert (wins <= trades).all()
+    assert (m[..., 5] >= 0).all() and (m[..., 5] <= 1).all()   # drawdown
+    gp, gl = m[..., 3], m[..., 4]
+    assert (gp >= 0).all() and (gl >= 0).all()
+    # equity change ~= gross_profit - gross_loss (open position slack)
+    closed = trades > 0
+    np.testing.assert_allclose(
+        (m[..., 0] - 1.0)[closed], (gp - gl)[closed], atol=0.05
+    )

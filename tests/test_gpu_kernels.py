@@ -1,0 +1,194 @@
+"""GPU numerics tests: every HIP kernel vs its CPU/torch fp32 reference.
+
+All tests here require an MI355X (run via `pytest -m gpu`). They fail
+loudly (never skip) if the HIP extension is missing on a GPU machine.
+"""
+
+import numpy as np
+import pytest
+
+torch = pytest.importorskip("torch")
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def dev():
+    assert torch.cuda.is_available()
+    return torch.device("cuda:0")
+
+
+def test_extension_loads_on_gpu():
+    from ai_crypto_trader_amd.ops import require_hip_ops
+    mod = require_hip_ops()
+    assert mod.GFX_ARCH == "gfx950"
+
+
+def test_mfma_f32_layout(dev):
+    """Validates the mfma_f32_16x16x4f32 fragment mapping (covar.hip)
+    against torch.matmul — asymmetric inputs so transposes can't hide."""
+    from ai_crypto_trader_amd.ops import require_hip_ops
+    ops = require_hip_ops()
+    M, N, K = 32, 48, 64
+    g = torch.Generator(device="cpu").manual_seed(0)
+    A = torch.randn(M, K, generator=g).float()
+    B = torch.randn(K, N, generator=g).float() + torch.arange(N).float() / N
+    Ad, Bd = A.to(dev), B.to(dev)
+    C = torch.empty(M, N, device=dev, dtype=torch.float32)
+    stream = torch.cuda.current_stream(dev).cuda_stream
+    ops.mfma_gemm_test(Ad.data_ptr(), Bd.data_ptr(), C.data_ptr(),
+                       M, N, K, stream)
+    torch.cuda.synchronize()
+    ref = A @ B
+    torch.testing.assert_close(C.cpu(), ref, rtol=1e-5, atol=1e-5)
+
+
+def test_backtest_gpu_matches_cpu(dev, small_market):
+    from ai_crypto_trader_amd.backtesting.engine_cpu import run_backtest_cpu
+    from ai_crypto_trader_amd.backtesting.strategy import random_population
+    from ai_crypto_trader_amd.ops.backtest import run_backtest_gpu
+
+    pop = random_population(64, seed=21)
+    ref = run_backtest_cpu(small_market, pop)
+
+    c_t = torch.from_numpy(small_market).to(dev)
+    p_t = torch.from_numpy(pop).to(dev)
+    got = run_backtest_gpu(c_t, p_t)
+    torch.cuda.synchronize()
+    got = got.cpu().numpy()
+
+    # integer-valued metrics must agree exactly (same decisions per candle)
+    np.testing.assert_array_equal(got[..., 1], ref[..., 1])   # n_trades
+    np.testing.assert_array_equal(got[..., 2], ref[..., 2])   # wins
+    np.testing.assert_allclose(got[..., 0], ref[..., 0], rtol=2e-5)
+    np.testing.assert_allclose(got[..., 5], ref[..., 5], rtol=1e-3,
+                               atol=1e-6)
+    np.testing.assert_allclose(got[..., 9], ref[..., 9], rtol=1e-3,
+                               atol=1e-3)
+
+
+def test_backtest_gpu_nondivisible_pop(dev, small_market):
+    """P not a multiple of the 256-lane block must still work."""
+    from ai_crypto_trader_amd.backtesting.engine_cpu import run_backtest_cpu
+    from ai_crypto_trader_amd.backtesting.strategy import random_population
+    from ai_crypto_trader_amd.ops.backtest import run_backtest_gpu
+
+    pop = random_population(37, seed=5)
+    ref = run_backtest_cpu(small_market, pop)
+    got = run_backtest_gpu(
+        torch.from_numpy(small_market).to(dev), torch.from_numpy(pop).to(dev)
+    )
+    torch.cuda.synchronize()
+    np.testing.assert_array_equal(got.cpu().numpy()[..., 1], ref[..., 1])
+
+
+def test_mc_gpu_matches_cpu_reference(dev):
+    from ai_crypto_trader_amd.ops.montecarlo import mc_paths_cpu, mc_paths_gpu
+
+    A, n_steps, n_paths = 8, 8, 4096
+    rng = np.random.default_rng(3)
+    corr = np.full((A, A), 0.3) + 0.7 * np.eye(A)
+    chol = np.linalg.cholesky(corr)
+    mu = rng.uniform(0.0, 0.2, A)
+    sigma = rng.uniform(0.2, 0.8, A)
+    w = np.full(A, 1.0 / A)
+    ref_fv, ref_dd = mc_paths_cpu(
+        chol, mu, sigma, w, n_steps=n_steps, n_paths=n_paths,
+        dt=1 / 252, seed=9,
+    )
+    fv, dd = mc_paths_gpu(
+        chol, mu, sigma, w, n_steps=n_steps, n_paths=n_paths,
+        dt=1 / 252, seed=9, device=dev,
+    )
+    torch.cuda.synchronize()
+    np.testing.assert_allclose(fv.cpu().numpy(), ref_fv, rtol=5e-4)
+    np.testing.assert_allclose(dd.cpu().numpy(), ref_dd, rtol=5e-3,
+                               atol=1e-5)
+
+
+def test_mc_gpu_analytic_moments(dev):
+    from ai_crypto_trader_amd.ops.montecarlo import mc_paths_gpu, risk_stats
+
+    A, n_steps, n_paths = 64, 32, 400_000
+    rho = 0.4
+    corr = np.full((A, A), rho) + (1 - rho) * np.eye(A)
+    chol = np.linalg.cholesky(corr)
+    mu = np.full(A, 0.1)
+    sigma = np.full(A, 0.5)
+    w = np.full(A, 1.0 / A)
+    dt = 1 / 252
+    fv, dd = mc_paths_gpu(
+        chol, mu, sigma, w, n_steps=n_steps, n_paths=n_paths, dt=dt,
+        seed=11, device=dev,
+    )
+    torch.cuda.synchronize()
+    horizon = n_steps * dt
+    assert abs(float(fv.mean()) - np.exp(0.1 * horizon)) < 0.005
+    sig_p = 0.5 * np.sqrt((1 + (A - 1) * rho) / A)
+    assert abs(float(torch.log(fv).std()) - sig_p * np.sqrt(horizon)) < 0.005
+    s = risk_stats(fv, v0=1.0)
+    assert s["cvar_95"] > s["var_95"] > 0
+    assert (dd >= 0).all()
+
+
+def test_cov_gpu_matches_cpu(dev):
+    from ai_crypto_trader_amd.ops.covar import cov_cpu, cov_gpu
+
+    rng = np.random.default_rng(4)
+    X = (rng.standard_normal((20_000, 64)) * 0.01).astype(np.float32)
+    ref = cov_cpu(X)
+    got = cov_gpu(torch.from_numpy(X).to(dev))
+    torch.cuda.synchronize()
+    np.testing.assert_allclose(got.cpu().numpy(), ref, rtol=1e-3,
+                               atol=1e-10)
+
+
+def test_cov_gpu_odd_T_and_small_N(dev):
+    from ai_crypto_trader_amd.ops.covar import cov_cpu, cov_gpu
+
+    rng = np.random.default_rng(6)
+    X = (rng.standard_normal((1037, 16)) * 0.02).astype(np.float32)
+    got = cov_gpu(torch.from_numpy(X).to(dev))
+    torch.cuda.synchronize()
+    np.testing.assert_allclose(got.cpu().numpy(), cov_cpu(X), rtol=1e-3,
+                               atol=1e-10)
+
+
+def test_indicators_gpu_matches_cpu(dev, small_market):
+    from ai_crypto_trader_amd.ops.indicators import (
+        indicators_cpu, indicators_gpu,
+    )
+
+    ref = indicators_cpu(small_market)
+    got = indicators_gpu(torch.from_numpy(small_market).to(dev))
+    torch.cuda.synchronize()
+    got = got.cpu().numpy()
+    # chunk-warmup reconvergence: equal to sequential within f32 noise
+    np.testing.assert_allclose(got, ref, rtol=1e-3, atol=2e-4)
+
+
+def test_ga_gpu_invariants(dev):
+    from ai_crypto_trader_amd.backtesting.strategy import (
+        PARAM_BOUNDS, random_population,
+    )
+    from ai_crypto_trader_amd.ops.ga import ga_evolve_gpu, population_diversity
+
+    P = 512
+    pop = torch.from_numpy(random_population(P, seed=1)).to(dev)
+    g = torch.Generator(device="cpu").manual_seed(0)
+    fitness = torch.randn(P, generator=g).float().to(dev)
+    child = ga_evolve_gpu(pop, fitness, elite_k=8, seed=3, gen=0)
+    torch.cuda.synchronize()
+    child_np = child.cpu().numpy()
+    order = torch.argsort(fitness, descending=True)[:8]
+    np.testing.assert_array_equal(
+        child_np[:8], pop[order].cpu().numpy())        # elitism
+    lo, hi = PARAM_BOUNDS[:, 0], PARAM_BOUNDS[:, 1]
+    assert (child_np >= lo - 1e-5).all()
+    assert (child_np <= hi + 1e-5).all()
+    assert (child_np[:, 4] > child_np[:, 3]).all()
+    assert population_diversity(child_np) > 0.001
+    # determinism
+    child2 = ga_evolve_gpu(pop, fitness, elite_k=8, seed=3, gen=0)
+    torch.cuda.synchronize()
+    np.testing.assert_array_equal(child_np, child2.cpu().numpy())
