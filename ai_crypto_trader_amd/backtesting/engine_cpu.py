@@ -53,7 +53,8 @@ def run_backtest_cpu(
 
     # Per-lane parameter broadcast: lane l = (param p, symbol s), s fastest.
     par = np.repeat(population, nsym, axis=0)          # (L, NPARAM)
-    rsi_p = np.maximum(par[:, 0].astype(np.int32), 1)
+    # reciprocal multiply (not divide) — matches the HIP kernel exactly
+    inv_rsi_p = (f32(1.0) / np.maximum(np.floor(par[:, 0]), f32(1.0))).astype(f32)
     rsi_os, rsi_ob = par[:, 1], par[:, 2]
     a_f = (f32(2.0) / (par[:, 3] + f32(1.0))).astype(f32)
     a_s = (f32(2.0) / (par[:, 4] + f32(1.0))).astype(f32)
@@ -127,8 +128,8 @@ def run_backtest_cpu(
 
         gain = np.maximum(change, f32(0.0))
         loss = np.maximum(-change, f32(0.0))
-        avg_gain += (gain - avg_gain) / rsi_p.astype(f32)
-        avg_loss += (loss - avg_loss) / rsi_p.astype(f32)
+        avg_gain += (gain - avg_gain) * inv_rsi_p
+        avg_loss += (loss - avg_loss) * inv_rsi_p
         rsi = f32(100.0) - f32(100.0) / (
             f32(1.0) + avg_gain / np.maximum(avg_loss, EPS)
         )

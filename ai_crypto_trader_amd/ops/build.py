@@ -25,6 +25,7 @@ SOURCES = [
     "covar.hip",
     "indicators.hip",
     "lstm.hip",
+    "rl_env.hip",
     "bindings.cpp",
 ]
 

@@ -4,15 +4,16 @@
 // (backtesting/strategy_tester.py:190-300, services/strategy_evaluation.py:777-878)
 // with one CDNA4 lane per (param-set x symbol) marching candles:
 //   - a block = 256 lanes = 256 param-sets of one symbol
-//   - candle tiles ([close,high,low,volume] f32x4) staged in LDS, read
-//     broadcast by all lanes (conflict-free: same address)
-//   - Bollinger rolling window lives in an LDS ring (stride MAX_WIN+1 so
-//     lane accesses spread across banks)
+//   - candle tiles staged in LDS with a MAX_WIN-candle HALO: close history
+//     is shared by every lane of the block, so the Bollinger "ring buffer"
+//     is just a broadcast read of close[t-W] from the halo tile — no
+//     per-lane LDS state at all (v1 used 32 KB/block of per-lane rings and
+//     was LDS-occupancy-bound at 4 waves/SIMD)
 //   - EMA/MACD/Wilder-RSI are O(1) register recurrences per candle
 //   - the position state machine (SL/TP/trailing/vote exits — semantics of
-//     trade_executor_service.py:55-399 + binance_ml_strategy.py:489-543) is
-//     branchless selects, matching backtesting/engine_cpu.py bit-for-bit
-//     modulo FMA contraction (disabled here via pragma for golden tests).
+//     trade_executor_service.py:55-399 + binance_ml_strategy.py:489-543)
+//     matches backtesting/engine_cpu.py bit-for-bit (fp-contract off, same
+//     operation order, f64 Bollinger sums both sides).
 //
 // Grid: nsym * ceil(P/256) blocks, XCD-affine map keeps the blocks of one
 // symbol on one XCD so their shared candle stream stays in that XCD's L2.
@@ -22,6 +23,8 @@
 #define BT_BLOCK 256
 #define BT_TILE 256
 #define BT_MAXWIN 32           // == strategy.py MAX_WIN
+#define BT_HALO BT_MAXWIN
+#define BT_SPAN (BT_TILE + BT_HALO)
 #define BT_NPARAM 17
 #define BT_NMETRIC 10
 #define BT_WARMUP 128          // == strategy.py WARMUP
@@ -32,7 +35,7 @@
 namespace {
 
 struct LaneParams {
-    float rsi_p, rsi_os, rsi_ob;
+    float inv_rsi_p, rsi_os, rsi_ob;
     float a_f, a_s, a_sig;
     int bb_w;
     float bb_k, bb_bth, bb_sth;
@@ -47,8 +50,8 @@ __global__ void __launch_bounds__(BT_BLOCK) backtest_kernel(
     int nsym, int T, int P, int chunks_per_sym, float initial_equity)
 {
 #pragma clang fp contract(off)           // match the numpy f32 reference
-    __shared__ float tile[BT_TILE][4];
-    __shared__ float ring[BT_BLOCK * (BT_MAXWIN + 1)];
+    __shared__ float chist[BT_SPAN];     // close history incl. halo
+    __shared__ float hl[BT_SPAN][2];     // high, low
 
     // block -> (symbol, param chunk); same-symbol blocks share an XCD when
     // the shape allows (dispatcher places block b on XCD b%8).
@@ -73,7 +76,8 @@ __global__ void __launch_bounds__(BT_BLOCK) backtest_kernel(
     LaneParams q;
     {
         const float* pr = pop + (long)pl * BT_NPARAM;
-        q.rsi_p = pr[0]; q.rsi_os = pr[1]; q.rsi_ob = pr[2];
+        q.inv_rsi_p = 1.0f / fmaxf(floorf(pr[0]), 1.0f);
+        q.rsi_os = pr[1]; q.rsi_ob = pr[2];
         q.a_f = 2.0f / (pr[3] + 1.0f);
         q.a_s = 2.0f / (pr[4] + 1.0f);
         q.a_sig = 2.0f / (pr[5] + 1.0f);
@@ -82,10 +86,7 @@ __global__ void __launch_bounds__(BT_BLOCK) backtest_kernel(
         q.entry_v = (int)pr[10]; q.exit_v = (int)pr[11];
         q.size_pct = pr[12]; q.sl_pct = pr[13]; q.tp_pct = pr[14];
         q.trail_pct = pr[15]; q.trail_act = pr[16];
-        q.rsi_p = fmaxf(floorf(q.rsi_p), 1.0f);
     }
-    float* myring = ring + tid * (BT_MAXWIN + 1);
-    for (int i = 0; i <= BT_MAXWIN; ++i) myring[i] = 0.0f;
 
     // ---- state ----------------------------------------------------------
     float ema_f = 0.f, ema_s = 0.f, sig = 0.f;
@@ -108,20 +109,27 @@ __global__ void __launch_bounds__(BT_BLOCK) backtest_kernel(
 
     for (int t0 = 0; t0 < T; t0 += BT_TILE) {
         __syncthreads();
-        if (t0 + tid < T) {
-            float4 c = sym_candles[t0 + tid];
-            tile[tid][0] = c.x;   // close
-            tile[tid][1] = c.y;   // high
-            tile[tid][2] = c.z;   // low
-            tile[tid][3] = c.w;   // volume
+        // stage [t0 - HALO, t0 + TILE) with zero-fill left of t=0
+        for (int i = tid; i < BT_SPAN; i += BT_BLOCK) {
+            const int t = t0 - BT_HALO + i;
+            if (t >= 0 && t < T) {
+                float4 c = sym_candles[t];
+                chist[i] = c.x;
+                hl[i][0] = c.y;
+                hl[i][1] = c.z;
+            } else {
+                chist[i] = 0.0f;
+                hl[i][0] = 0.0f;
+                hl[i][1] = 0.0f;
+            }
         }
         __syncthreads();
         const int tend = min(BT_TILE, T - t0);
         for (int tt = 0; tt < tend; ++tt) {
             const int t = t0 + tt;
-            const float close = tile[tt][0];
-            const float high = tile[tt][1];
-            const float low = tile[tt][2];
+            const float close = chist[tt + BT_HALO];
+            const float high = hl[tt + BT_HALO][0];
+            const float low = hl[tt + BT_HALO][1];
 
             // --- 1. indicators ---------------------------------------
             float change;
@@ -138,17 +146,17 @@ __global__ void __launch_bounds__(BT_BLOCK) backtest_kernel(
 
             float gain = fmaxf(change, 0.0f);
             float loss = fmaxf(-change, 0.0f);
-            avg_gain += (gain - avg_gain) / q.rsi_p;
-            avg_loss += (loss - avg_loss) / q.rsi_p;
+            avg_gain += (gain - avg_gain) * q.inv_rsi_p;
+            avg_loss += (loss - avg_loss) * q.inv_rsi_p;
             float rsi =
                 100.0f - 100.0f / (1.0f + avg_gain / fmaxf(avg_loss, BT_EPS));
 
-            int ridx = t % q.bb_w;
-            double old = (double)myring[ridx];
+            // Bollinger: close[t - W] comes from the shared halo tile
+            // (== the zero-initialized per-lane ring of engine_cpu.py)
+            double old = (double)chist[tt + BT_HALO - q.bb_w];
             double c64 = (double)close;
             bb_sum += c64 - old;
             bb_sum2 += c64 * c64 - old * old;
-            myring[ridx] = close;
             double inv_cnt = (t + 1 < q.bb_w) ? 1.0 / (t + 1.0) : inv_w;
             double mean64 = bb_sum * inv_cnt;
             double var64 = fmax(bb_sum2 * inv_cnt - mean64 * mean64, 0.0);

@@ -25,10 +25,23 @@ extern "C" void launch_mc_paths(const float*, const float*, const float*,
 extern "C" void launch_cov(const float*, float*, int, int, hipStream_t);
 extern "C" void launch_indicators(const float*, float*, int, int, int,
                                   hipStream_t);
-extern "C" void launch_lstm_cell_fwd(const float*, const float*, float*,
-                                     float*, float*, int, int, hipStream_t);
+extern "C" void launch_lstm_seq_fwd(const void*, const void*, const float*,
+                                    void*, void*, float*, int, int, int,
+                                    hipStream_t);
+extern "C" void launch_lstm_seq_bwd(const float*, const void*, const float*,
+                                    const void*, void*, int, int, int,
+                                    hipStream_t);
 extern "C" void launch_mfma_gemm_test(const void*, const void*, float*, int,
                                       int, int, hipStream_t);
+extern "C" void launch_mfma_gemm_test_bf16(const void*, const void*, float*,
+                                           int, int, int, hipStream_t);
+extern "C" void launch_env_reset(const float*, float*, float*, int, int, int,
+                                 int, uint64_t, uint64_t, hipStream_t);
+extern "C" void launch_env_step(const float*, float*, const int*, float*,
+                                float*, float*, int, int, int, int, float,
+                                uint64_t, uint64_t, hipStream_t);
+extern "C" void launch_gae(const float*, const float*, const float*, float*,
+                           float*, int, int, float, float, hipStream_t);
 
 static void check(hipError_t e, const char* what) {
     if (e != hipSuccess)
@@ -113,6 +126,83 @@ PYBIND11_MODULE(_hip_ops, m) {
                                     reinterpret_cast<float*>(c), M, N, K,
                                     as_stream(stream));
               check(hipGetLastError(), "mfma_gemm_test launch");
+          });
+
+    m.def("mfma_gemm_test_bf16",
+          [](uintptr_t a, uintptr_t b, uintptr_t c, int M, int N, int K,
+             uintptr_t stream) {
+              launch_mfma_gemm_test_bf16(reinterpret_cast<const void*>(a),
+                                         reinterpret_cast<const void*>(b),
+                                         reinterpret_cast<float*>(c), M, N,
+                                         K, as_stream(stream));
+              check(hipGetLastError(), "mfma_gemm_test_bf16 launch");
+          });
+
+    m.def("lstm_seq_fwd",
+          [](uintptr_t xproj, uintptr_t Wt, uintptr_t bias, uintptr_t h_out,
+             uintptr_t gates_out, uintptr_t c_out, int B, int T, int H,
+             uintptr_t stream) {
+              launch_lstm_seq_fwd(reinterpret_cast<const void*>(xproj),
+                                  reinterpret_cast<const void*>(Wt),
+                                  reinterpret_cast<const float*>(bias),
+                                  reinterpret_cast<void*>(h_out),
+                                  reinterpret_cast<void*>(gates_out),
+                                  reinterpret_cast<float*>(c_out), B, T, H,
+                                  as_stream(stream));
+              check(hipGetLastError(), "lstm_seq_fwd launch");
+          });
+
+    m.def("lstm_seq_bwd",
+          [](uintptr_t dh_up, uintptr_t gates, uintptr_t c_sav, uintptr_t W,
+             uintptr_t dgates_out, int B, int T, int H, uintptr_t stream) {
+              launch_lstm_seq_bwd(reinterpret_cast<const float*>(dh_up),
+                                  reinterpret_cast<const void*>(gates),
+                                  reinterpret_cast<const float*>(c_sav),
+                                  reinterpret_cast<const void*>(W),
+                                  reinterpret_cast<void*>(dgates_out), B, T,
+                                  H, as_stream(stream));
+              check(hipGetLastError(), "lstm_seq_bwd launch");
+          });
+
+    m.def("env_reset",
+          [](uintptr_t candles, uintptr_t state, uintptr_t obs, int nsym,
+             int T, int n_envs, int ep_len, uint64_t seed, uint64_t epoch,
+             uintptr_t stream) {
+              launch_env_reset(reinterpret_cast<const float*>(candles),
+                               reinterpret_cast<float*>(state),
+                               reinterpret_cast<float*>(obs), nsym, T,
+                               n_envs, ep_len, seed, epoch,
+                               as_stream(stream));
+              check(hipGetLastError(), "env_reset launch");
+          });
+
+    m.def("env_step",
+          [](uintptr_t candles, uintptr_t state, uintptr_t actions,
+             uintptr_t obs, uintptr_t reward, uintptr_t done, int nsym,
+             int T, int n_envs, int ep_len, float fee, uint64_t seed,
+             uint64_t epoch, uintptr_t stream) {
+              launch_env_step(reinterpret_cast<const float*>(candles),
+                              reinterpret_cast<float*>(state),
+                              reinterpret_cast<const int*>(actions),
+                              reinterpret_cast<float*>(obs),
+                              reinterpret_cast<float*>(reward),
+                              reinterpret_cast<float*>(done), nsym, T,
+                              n_envs, ep_len, fee, seed, epoch,
+                              as_stream(stream));
+              check(hipGetLastError(), "env_step launch");
+          });
+
+    m.def("gae",
+          [](uintptr_t rewards, uintptr_t values, uintptr_t dones,
+             uintptr_t adv, uintptr_t returns, int T, int E, float gamma,
+             float lam, uintptr_t stream) {
+              launch_gae(reinterpret_cast<const float*>(rewards),
+                         reinterpret_cast<const float*>(values),
+                         reinterpret_cast<const float*>(dones),
+                         reinterpret_cast<float*>(adv),
+                         reinterpret_cast<float*>(returns), T, E, gamma, lam,
+                         as_stream(stream));
+              check(hipGetLastError(), "gae launch");
           });
 
     m.def("device_synchronize", []() { check(hipDeviceSynchronize(), "sync"); });
