@@ -130,9 +130,10 @@ def run_backtest_cpu(
         loss = np.maximum(-change, f32(0.0))
         avg_gain += (gain - avg_gain) * inv_rsi_p
         avg_loss += (loss - avg_loss) * inv_rsi_p
-        rsi = f32(100.0) - f32(100.0) / (
-            f32(1.0) + avg_gain / np.maximum(avg_loss, EPS)
-        )
+        # division-free RSI votes: rsi = 100*ag/(ag+al'), so
+        # rsi < thr  <=>  100*ag < thr*(ag+al')   (al' = max(al, eps) > 0)
+        rsi_num = f32(100.0) * avg_gain
+        rsi_den = avg_gain + np.maximum(avg_loss, EPS)
 
         ridx = t % bb_w                                   # per-lane ring slot
         old = ring[lanes, ridx].astype(np.float64)
@@ -146,21 +147,24 @@ def run_backtest_cpu(
         mean = mean64.astype(f32)
         std = np.sqrt(var64.astype(f32))
         band = bb_k * std
-        bb_pos = (close - (mean - band)) / np.maximum(f32(2.0) * band, EPS)
+        # division-free BB votes: bb_pos = num/den with den > 0, so
+        # bb_pos < thr  <=>  num < thr*den
+        bb_num = close - (mean - band)
+        bb_den = np.maximum(f32(2.0) * band, EPS)
 
         prev_close = close
 
         # --- 2. votes ------------------------------------------------------
         if t >= WARMUP:
             buy = (
-                (rsi < rsi_os).astype(np.int32)
+                (rsi_num < rsi_os * rsi_den).astype(np.int32)
                 + (macd_hist > 0).astype(np.int32)
-                + (bb_pos < bb_bth).astype(np.int32)
+                + (bb_num < bb_bth * bb_den).astype(np.int32)
             )
             sell = (
-                (rsi > rsi_ob).astype(np.int32)
+                (rsi_num > rsi_ob * rsi_den).astype(np.int32)
                 + (macd_hist < 0).astype(np.int32)
-                + (bb_pos > bb_sth).astype(np.int32)
+                + (bb_num > bb_sth * bb_den).astype(np.int32)
             )
             net = buy - sell
         else:

@@ -148,8 +148,10 @@ __global__ void __launch_bounds__(BT_BLOCK) backtest_kernel(
             float loss = fmaxf(-change, 0.0f);
             avg_gain += (gain - avg_gain) * q.inv_rsi_p;
             avg_loss += (loss - avg_loss) * q.inv_rsi_p;
-            float rsi =
-                100.0f - 100.0f / (1.0f + avg_gain / fmaxf(avg_loss, BT_EPS));
+            // division-free RSI votes (engine_cpu.py): rsi<thr <=>
+            // 100*ag < thr*(ag+al')
+            float rsi_num = 100.0f * avg_gain;
+            float rsi_den = avg_gain + fmaxf(avg_loss, BT_EPS);
 
             // Bollinger: close[t - W] comes from the shared halo tile
             // (== the zero-initialized per-lane ring of engine_cpu.py)
@@ -157,24 +159,27 @@ __global__ void __launch_bounds__(BT_BLOCK) backtest_kernel(
             double c64 = (double)close;
             bb_sum += c64 - old;
             bb_sum2 += c64 * c64 - old * old;
-            double inv_cnt = (t + 1 < q.bb_w) ? 1.0 / (t + 1.0) : inv_w;
+            double inv_cnt = inv_w;
+            if (t < BT_MAXWIN && t + 1 < q.bb_w)   // uniformly skipped t>=32
+                inv_cnt = 1.0 / (t + 1.0);
             double mean64 = bb_sum * inv_cnt;
             double var64 = fmax(bb_sum2 * inv_cnt - mean64 * mean64, 0.0);
             float mean = (float)mean64;
             float std_ = sqrtf((float)var64);
             float band = q.bb_k * std_;
-            float bb_pos =
-                (close - (mean - band)) / fmaxf(2.0f * band, BT_EPS);
+            // division-free BB votes (engine_cpu.py): pos<thr <=> num<thr*den
+            float bb_num = close - (mean - band);
+            float bb_den = fmaxf(2.0f * band, BT_EPS);
 
             prev_close = close;
 
             // --- 2. votes --------------------------------------------
             int net = 0;
             if (t >= BT_WARMUP) {
-                int buy = (rsi < q.rsi_os) + (macd_hist > 0.0f) +
-                          (bb_pos < q.bb_bth);
-                int sell = (rsi > q.rsi_ob) + (macd_hist < 0.0f) +
-                           (bb_pos > q.bb_sth);
+                int buy = (rsi_num < q.rsi_os * rsi_den) +
+                          (macd_hist > 0.0f) + (bb_num < q.bb_bth * bb_den);
+                int sell = (rsi_num > q.rsi_ob * rsi_den) +
+                           (macd_hist < 0.0f) + (bb_num > q.bb_sth * bb_den);
                 net = buy - sell;
             }
 

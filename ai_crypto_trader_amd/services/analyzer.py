@@ -1,0 +1,173 @@
+"""Analyzer service (reference parity: services/ai_analyzer_service.py +
+services/ai_trader.py).
+
+The reference sends every market update to GPT-4 (ai_trader.py:36-189,
+network-bound, 60 s/symbol throttle). Here the default decision engine is
+LocalAnalyst — a deterministic, explainable voting analyst implementing
+the same decision contract (BUY/SELL/HOLD + confidence + explanation +
+factor_weights, README.md:516-575) from the same inputs (market update +
+social + news + nn predictions) with no network. An OpenAI-compatible
+adapter can be plugged in behind the same interface when credentials and
+egress exist (ai_trader.py parity seam).
+"""
+
+from __future__ import annotations
+
+import time
+
+from ..bus.schema import Channels, Keys, TradingSignal
+from .base import Service
+
+
+class LocalAnalyst:
+    """Deterministic explainable analyst: weighted factor votes.
+
+    Mirrors the reference's factor_weights explanation block
+    (ai_trader.py:108-167) — each factor contributes a score in [-1, 1];
+    decision = sign of the weighted sum with confidence = |score| mapped
+    through the factor agreement."""
+
+    VERSION = "rule-1.0"
+    MODEL_ID = "local_analyst"
+
+    def __init__(self, weights: dict | None = None):
+        self.weights = weights or {
+            "momentum": 0.25, "oscillators": 0.25, "trend": 0.20,
+            "social_sentiment": 0.15, "nn_prediction": 0.15,
+        }
+
+    def analyze(self, market: dict, social: dict | None = None,
+                news: dict | None = None,
+                nn_pred: dict | None = None) -> dict:
+        f = {}
+        # momentum: multi-horizon price changes
+        mom = (market.get("price_change_1m", 0) * 0.5 +
+               market.get("price_change_5m", 0) * 0.3 +
+               market.get("price_change_15m", 0) * 0.2)
+        f["momentum"] = max(-1.0, min(1.0, mom / 1.0))
+        # oscillators: RSI/stoch/williams consensus
+        rsi = market.get("rsi", 50)
+        stoch = market.get("stoch_k", 50)
+        willr = market.get("williams_r", -50)
+        osc = ((30 - rsi) / 30 if rsi < 30 else
+               (70 - rsi) / 30 if rsi > 70 else 0.0)
+        osc += 0.5 * ((20 - stoch) / 20 if stoch < 20 else
+                      (80 - stoch) / 20 if stoch > 80 else 0.0)
+        osc += 0.5 * ((-80 - willr) / 20 if willr < -80 else
+                      (-20 - willr) / 20 if willr > -20 else 0.0)
+        f["oscillators"] = max(-1.0, min(1.0, osc))
+        # trend
+        t = {"uptrend": 1.0, "downtrend": -1.0}.get(
+            market.get("trend", "neutral"), 0.0)
+        f["trend"] = t * market.get("trend_strength", 0.0) / 100.0
+        # social sentiment in [0,1] -> [-1,1]
+        if social:
+            s = social.get("weighted_sentiment", 0.5)
+            f["social_sentiment"] = (s - 0.5) * 2.0
+        else:
+            f["social_sentiment"] = 0.0
+        if news:
+            f["social_sentiment"] = (
+                f["social_sentiment"] +
+                (news.get("sentiment", 0.5) - 0.5) * 2.0) / 2.0
+        # nn prediction
+        if nn_pred:
+            f["nn_prediction"] = max(-1.0, min(
+                1.0, nn_pred.get("predicted_change_pct", 0.0) / 1.0))
+        else:
+            f["nn_prediction"] = 0.0
+
+        score = sum(self.weights[k] * f[k] for k in self.weights)
+        agreement = sum(
+            1 for k in self.weights
+            if f[k] * score > 0 and abs(f[k]) > 0.05)
+        confidence = min(0.99, abs(score) * 1.6 + 0.08 * agreement)
+        decision = "BUY" if score > 0.12 else (
+            "SELL" if score < -0.12 else "HOLD")
+        risk = "low" if abs(score) > 0.5 else (
+            "high" if abs(f["oscillators"]) > 0.8 else "medium")
+        key = sorted(self.weights, key=lambda k: -abs(f[k] * self.weights[k]))
+        return {
+            "decision": decision,
+            "confidence": round(confidence, 4),
+            "reasoning": f"weighted factor score {score:+.3f} "
+                         f"({agreement} factors agree)",
+            "risk_level": risk,
+            "key_indicators": key[:3],
+            "explanation": {k: round(v, 4) for k, v in f.items()},
+            "factor_weights": self.weights,
+            "model_version": self.VERSION,
+            "model_id": self.MODEL_ID,
+        }
+
+    def should_take_trade(self, analysis: dict,
+                          min_confidence: float = 0.7) -> bool:
+        """ai_trader.py:368-387 gate."""
+        return (analysis["decision"] == "BUY"
+                and analysis["confidence"] >= min_confidence)
+
+
+class AnalyzerService(Service):
+    name = "ai_analyzer"
+
+    def __init__(self, bus, config=None, analyst: LocalAnalyst | None = None):
+        super().__init__(bus, config)
+        self.analyst = analyst or LocalAnalyst()
+        self.last_analysis: dict[str, float] = {}
+        self.social_cache: dict[str, dict] = {}
+        self.signals_published = 0
+
+    def run_tasks(self):
+        return [self._consume_market(), self._consume_social()]
+
+    async def _consume_social(self):
+        sub = self.bus.subscribe(Channels.SOCIAL_UPDATES)
+
+        def on_msg(_, msg):
+            sym = msg.get("symbol")
+            if sym:
+                self.social_cache[sym] = msg.get("data", {})
+
+        await self.consume(sub, on_msg)
+
+    async def _consume_market(self):
+        sub = self.bus.subscribe(Channels.MARKET_UPDATES)
+        interval = self.config.trading.ai_analysis_interval
+
+        async def on_msg(_, market):
+            sym = market.get("symbol")
+            if sym is None:
+                return
+            now = time.monotonic()
+            if now - self.last_analysis.get(sym, -1e9) < interval:
+                return       # per-symbol throttle (ai_analyzer:388-393)
+            self.last_analysis[sym] = now
+            news = await self.bus.hget(Keys.NEWS_ANALYSIS, sym)
+            nn = await self.bus.get_json(Keys.nn_prediction(sym, "1m"))
+            import json as _json
+            news_d = _json.loads(news) if news else None
+            analysis = self.analyst.analyze(
+                market, self.social_cache.get(sym), news_d, nn)
+            sig = TradingSignal(
+                symbol=sym,
+                decision=analysis["decision"],
+                confidence=analysis["confidence"],
+                reasoning=analysis["reasoning"],
+                risk_level=analysis["risk_level"],
+                key_indicators=analysis["key_indicators"],
+                explanation=analysis["explanation"],
+                factor_weights=analysis["factor_weights"],
+                model_version=analysis["model_version"],
+                model_id=analysis["model_id"],
+                market_data=market,
+            )
+            await self.bus.publish(Channels.TRADING_SIGNALS, sig.to_dict())
+            self.signals_published += 1
+            self.metrics.signals.labels(sym, analysis["decision"]).inc()
+            self.metrics.ai_confidence.labels(sym).set(
+                analysis["confidence"])
+
+        await self.consume(sub, on_msg)
+
+    async def run(self):
+        pass

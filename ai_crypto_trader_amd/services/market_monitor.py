@@ -1,0 +1,139 @@
+"""Market monitor service (reference parity:
+services/market_monitor_service.py — websocket/feed intake, per-symbol
+multi-timeframe indicators (:219-298), combined indicators (:476-485),
+volume profile on a 5-minute cadence (:303-372), publishes
+`market_updates` + `trading_opportunities`, maintains the
+`current_prices` hash (:541-545))."""
+
+from __future__ import annotations
+
+import numpy as np
+
+from ..bus.schema import Channels, Keys, MarketUpdate
+from ..ops.indicators import indicators_cpu
+from ..utils.indicator_combinations import calculate_indicator_combinations
+from ..utils.volume_profile import VolumeProfileAnalyzer
+from .base import Service
+
+HIST = 2048            # rolling candles kept per symbol
+
+
+class MarketMonitorService(Service):
+    name = "market_monitor"
+
+    def __init__(self, bus, feed, config=None):
+        super().__init__(bus, config)
+        self.feed = feed
+        self.vp = VolumeProfileAnalyzer()
+        self.hist: dict[str, np.ndarray] = {}
+        self.hist_len: dict[str, int] = {}
+        self.last_vp: dict[str, dict] = {}
+        self.last_vp_t: dict[str, int] = {}
+        self.updates_published = 0
+
+    def _push(self, c) -> np.ndarray | None:
+        h = self.hist.get(c.symbol)
+        if h is None:
+            h = np.zeros((HIST, 4), np.float32)
+            self.hist[c.symbol] = h
+            self.hist_len[c.symbol] = 0
+        n = self.hist_len[c.symbol]
+        if n == HIST:
+            h[:-1] = h[1:]
+            n -= 1
+        h[n] = (c.close, c.high, c.low, c.volume)
+        self.hist_len[c.symbol] = n + 1
+        return h[: n + 1]
+
+    @staticmethod
+    def _resample(win: np.ndarray, k: int) -> np.ndarray:
+        """1m -> k-minute candles (close=last, high=max, low=min, vol=sum)."""
+        T = win.shape[0] // k * k
+        if T == 0:
+            return win[:1][None][0].reshape(1, 4)
+        w = win[-T:].reshape(-1, k, 4)
+        out = np.empty((w.shape[0], 4), np.float32)
+        out[:, 0] = w[:, -1, 0]
+        out[:, 1] = w[:, :, 1].max(axis=1)
+        out[:, 2] = w[:, :, 2].min(axis=1)
+        out[:, 3] = w[:, :, 3].sum(axis=1)
+        return out
+
+    def _indic_last(self, win: np.ndarray) -> np.ndarray:
+        # bound the per-update cost: 160 candles cover every window/EMA
+        # warmup the live path needs (offline exact paths use the GPU
+        # indicator kernel over full history instead)
+        return indicators_cpu(win[None, -160:])[0, -1]
+
+    def build_update(self, symbol: str, win: np.ndarray) -> MarketUpdate:
+        last = self._indic_last(win)
+        close = float(win[-1, 0])
+        i3 = self._indic_last(self._resample(win, 3)) if win.shape[0] >= 6 \
+            else last
+        i5 = self._indic_last(self._resample(win, 5)) if win.shape[0] >= 10 \
+            else last
+
+        def pct(nback):
+            if win.shape[0] <= nback:
+                return 0.0
+            return float((close / win[-1 - nback, 0] - 1.0) * 100.0)
+
+        # trend from SMA20/SMA50 (market_monitor_service.py:262-280)
+        closes = win[:, 0]
+        sma20 = closes[-20:].mean() if len(closes) >= 20 else closes.mean()
+        sma50 = closes[-50:].mean() if len(closes) >= 50 else closes.mean()
+        if close > sma20 > sma50:
+            trend = "uptrend"
+        elif close < sma20 < sma50:
+            trend = "downtrend"
+        else:
+            trend = "neutral"
+        tstr = float(min(abs(close / sma20 - 1) * 0.6 +
+                         abs(close / sma50 - 1) * 0.4, 1.0) * 100.0)
+
+        bb_up, bb_lo = float(last[7]), float(last[8])
+        bb_pos = (close - bb_lo) / max(bb_up - bb_lo, 1e-9)
+
+        u = MarketUpdate(
+            symbol=symbol,
+            current_price=close,
+            avg_volume=float(win[-20:, 3].mean()),
+            rsi=float(last[5]), rsi_3m=float(i3[5]), rsi_5m=float(i5[5]),
+            stoch_k=float(last[10]),
+            macd=float(last[2]), macd_3m=float(i3[2]), macd_5m=float(i5[2]),
+            williams_r=float(last[11]),
+            bb_position=float(bb_pos),
+            trend=trend, trend_strength=tstr,
+            price_change_1m=pct(1), price_change_3m=pct(3),
+            price_change_5m=pct(5), price_change_15m=pct(15),
+        )
+        d = u.to_dict()
+        d["combined_indicators"] = calculate_indicator_combinations(d)
+        return d
+
+    async def run(self):
+        tcfg = self.config.trading
+        async for c in self.feed:
+            if not self.running:
+                break
+            win = self._push(c)
+            if win.shape[0] < 30:
+                continue
+            d = self.build_update(c.symbol, win)
+            # volume profile every 5 candles (reference: 5-min cadence)
+            t_last = self.last_vp_t.get(c.symbol, -10)
+            if c.t - t_last >= 5:
+                self.last_vp[c.symbol] = self.vp.analyze(win[-240:])
+                self.last_vp_t[c.symbol] = c.t
+            d["volume_profile"] = self.last_vp.get(c.symbol, {})
+
+            await self.bus.publish(Channels.MARKET_UPDATES, d)
+            await self.bus.hset(Keys.CURRENT_PRICES, c.symbol,
+                                d["current_price"])
+            self.updates_published += 1
+            # opportunity gate (market_monitor_service.py:563-574)
+            if abs(d["price_change_1m"]) >= tcfg.min_price_change_pct and \
+                    d["avg_volume"] * d["current_price"] >= \
+                    tcfg.min_volume_usdc:
+                await self.bus.publish(Channels.TRADING_OPPORTUNITIES, d)
+        self.running = False

@@ -1,0 +1,241 @@
+"""Market regime service (reference parity:
+services/market_regime_service.py + services/utils/market_regime_detector.py).
+
+Features per window: return, volatility, trend strength (rolling slope),
+RSI, MACD, BB width (market_regime_detector.py:64-110). Detectors: KMeans
+/ GMM (torch, GPU-capable) and the reference's heuristic cluster->label
+mapping by mean return & volatility (:226-296), plus the rule-based
+fallback (market_regime_service.py:503-604). Publishes `strategy_switch`
+on regime changes; maintains `current_market_regime` /
+`market_regime_history` keys."""
+
+from __future__ import annotations
+
+import time
+
+import numpy as np
+import torch
+
+from ..bus.schema import Channels, Keys, StrategySwitch
+from .base import Service
+
+REGIMES = ("bull", "bear", "ranging", "volatile")
+
+
+def extract_features(closes: np.ndarray, win: int = 32) -> np.ndarray:
+    """(T,) closes -> (n_win, 6) features [ret, vol, slope, rsi, macd, bbw]
+    over trailing windows (market_regime_detector.py:64-110 semantics)."""
+    T = len(closes)
+    n = T - win
+    if n <= 0:
+        return np.zeros((0, 6), np.float32)
+    out = np.zeros((n, 6), np.float32)
+    lr = np.diff(np.log(closes))
+    for i in range(n):
+        w = closes[i:i + win]
+        r = lr[i:i + win - 1]
+        out[i, 0] = w[-1] / w[0] - 1.0
+        out[i, 1] = r.std()
+        x = np.arange(win)
+        out[i, 2] = np.polyfit(x, w / w.mean(), 1)[0] * win
+        g = np.maximum(np.diff(w), 0).mean()
+        l = np.maximum(-np.diff(w), 0).mean()
+        out[i, 3] = 100 * g / max(g + l, 1e-12)
+        ema_f = w[-12:].mean()
+        ema_s = w[-26:].mean() if win >= 26 else w.mean()
+        out[i, 4] = (ema_f - ema_s) / w[-1]
+        out[i, 5] = w.std() / max(w.mean(), 1e-9)
+    return out
+
+
+class KMeansTorch:
+    """Plain torch KMeans (runs on GPU when tensors are cuda) — the
+    detector's default model (market_regime_detector.py:138)."""
+
+    def __init__(self, k: int = 4, iters: int = 50, seed: int = 0):
+        self.k = k
+        self.iters = iters
+        self.seed = seed
+        self.centers: torch.Tensor | None = None
+
+    def fit(self, X: torch.Tensor):
+        g = torch.Generator(device="cpu").manual_seed(self.seed)
+        idx = torch.randperm(X.shape[0], generator=g)[: self.k]
+        self.centers = X[idx.to(X.device)].clone()
+        for _ in range(self.iters):
+            d = torch.cdist(X, self.centers)
+            assign = d.argmin(dim=1)
+            for c in range(self.k):
+                m = assign == c
+                if m.any():
+                    self.centers[c] = X[m].mean(dim=0)
+        return self
+
+    def predict(self, X: torch.Tensor) -> torch.Tensor:
+        return torch.cdist(X, self.centers).argmin(dim=1)
+
+
+class GMMTorch:
+    """Diagonal-covariance GMM via EM (market_regime_detector.py:144)."""
+
+    def __init__(self, k: int = 4, iters: int = 60, seed: int = 0):
+        self.k = k
+        self.iters = iters
+        self.seed = seed
+
+    def fit(self, X: torch.Tensor):
+        km = KMeansTorch(self.k, 20, self.seed).fit(X)
+        self.mu = km.centers.clone()
+        self.var = torch.ones_like(self.mu) * X.var(dim=0, keepdim=True)
+        self.pi = torch.full((self.k,), 1.0 / self.k, device=X.device)
+        for _ in range(self.iters):
+            logp = self._log_prob(X)                       # (N, k)
+            logr = logp - torch.logsumexp(logp, dim=1, keepdim=True)
+            r = logr.exp()
+            nk = r.sum(dim=0) + 1e-9
+            self.mu = (r.T @ X) / nk[:, None]
+            self.var = (r.T @ (X ** 2)) / nk[:, None] - self.mu ** 2
+            self.var = torch.clamp(self.var, min=1e-8)
+            self.pi = nk / nk.sum()
+        return self
+
+    def _log_prob(self, X):
+        d = X[:, None, :] - self.mu[None]
+        return (torch.log(self.pi)[None]
+                - 0.5 * ((d ** 2) / self.var[None]).sum(-1)
+                - 0.5 * torch.log(self.var[None]).sum(-1))
+
+    def predict(self, X):
+        return self._log_prob(X).argmax(dim=1)
+
+    def predict_proba(self, X):
+        lp = self._log_prob(X)
+        return (lp - torch.logsumexp(lp, 1, keepdim=True)).exp()
+
+
+def label_clusters(X: np.ndarray, assign: np.ndarray, k: int) -> dict:
+    """Heuristic cluster -> regime mapping by mean return & volatility
+    (market_regime_detector.py:226-296)."""
+    stats = []
+    for c in range(k):
+        m = assign == c
+        if not m.any():
+            stats.append((0.0, 0.0))
+            continue
+        stats.append((float(X[m, 0].mean()), float(X[m, 1].mean())))
+    vols = [s[1] for s in stats]
+    vol_hi = np.percentile(vols, 75)
+    labels = {}
+    for c, (ret, vol) in enumerate(stats):
+        if vol >= vol_hi and vol > 0:
+            labels[c] = "volatile"
+        elif ret > 0.002:
+            labels[c] = "bull"
+        elif ret < -0.002:
+            labels[c] = "bear"
+        else:
+            labels[c] = "ranging"
+    return labels
+
+
+def rule_based_regime(closes: np.ndarray) -> tuple[str, float]:
+    """Rule-based thresholds (market_regime_service.py:503-604)."""
+    if len(closes) < 30:
+        return "ranging", 0.0
+    r = np.diff(np.log(closes[-100:]))
+    ret = closes[-1] / closes[-min(len(closes), 100)] - 1.0
+    vol = float(r.std() * np.sqrt(525_600))
+    if vol > 1.2:
+        return "volatile", min(vol / 2.0, 1.0)
+    if ret > 0.01:
+        return "bull", min(abs(ret) * 20, 1.0)
+    if ret < -0.01:
+        return "bear", min(abs(ret) * 20, 1.0)
+    return "ranging", 0.3
+
+
+class MarketRegimeService(Service):
+    name = "market_regime"
+
+    def __init__(self, bus, config=None, device="cpu"):
+        super().__init__(bus, config)
+        self.device = device
+        self.prices: dict[str, list[float]] = {}
+        self.model = None
+        self.labels: dict[int, str] = {}
+        self.current = "ranging"
+        self.history: list[dict] = []
+
+    def run_tasks(self):
+        return [self._consume_market(), self._regime_loop()]
+
+    async def _consume_market(self):
+        sub = self.bus.subscribe(Channels.MARKET_UPDATES)
+
+        def on_msg(_, m):
+            if m.get("symbol"):
+                h = self.prices.setdefault(m["symbol"], [])
+                h.append(m["current_price"])
+                if len(h) > 4096:
+                    del h[:2048]
+
+        await self.consume(sub, on_msg)
+
+    def _primary_closes(self) -> np.ndarray | None:
+        if not self.prices:
+            return None
+        sym = self.config.trading.symbols[0] \
+            if self.config.trading.symbols[0] in self.prices \
+            else next(iter(self.prices))
+        h = self.prices[sym]
+        return np.asarray(h, np.float64) if len(h) >= 64 else None
+
+    def detect(self, closes: np.ndarray) -> tuple[str, float]:
+        method = self.config.regime.method
+        if method == "rule" or len(closes) < 200:
+            return rule_based_regime(closes)
+        X_np = extract_features(closes)
+        if len(X_np) < 4 * self.config.regime.n_regimes:
+            return rule_based_regime(closes)
+        X = torch.from_numpy(X_np)
+        if self.device != "cpu":
+            X = X.to(self.device)
+        Xn = (X - X.mean(0)) / (X.std(0) + 1e-9)
+        cls = GMMTorch if method == "gmm" else KMeansTorch
+        self.model = cls(self.config.regime.n_regimes,
+                         seed=self.config.seed).fit(Xn)
+        assign = self.model.predict(Xn).cpu().numpy()
+        self.labels = label_clusters(X_np, assign, self.config.regime.n_regimes)
+        regime = self.labels.get(int(assign[-1]), "ranging")
+        conf = float((assign == assign[-1]).mean())
+        return regime, conf
+
+    async def _regime_loop(self):
+        while self.running:
+            closes = self._primary_closes()
+            if closes is not None:
+                closes = closes[-self.config.regime.lookback:]
+                try:
+                    regime, conf = self.detect(closes)
+                except Exception as e:
+                    self.log.warning("detect failed: %s", e)
+                    regime, conf = rule_based_regime(closes)
+                vol = float(np.diff(np.log(closes[-100:])).std()
+                            * np.sqrt(525_600)) if len(closes) > 2 else 0.0
+                entry = {"regime": regime, "confidence": conf,
+                         "volatility": min(vol, 2.0), "at": time.time()}
+                await self.bus.set(Keys.CURRENT_MARKET_REGIME, entry)
+                self.history.append(entry)
+                await self.bus.set(Keys.MARKET_REGIME_HISTORY,
+                                   self.history[-100:])
+                if regime != self.current:
+                    await self.bus.publish(
+                        Channels.STRATEGY_SWITCH,
+                        StrategySwitch(regime, f"for-{self.current}",
+                                       f"for-{regime}",
+                                       "regime change").to_dict())
+                    self.current = regime
+            await self.sleep(5.0)
+
+    async def run(self):
+        pass

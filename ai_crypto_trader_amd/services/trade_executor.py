@@ -1,0 +1,247 @@
+"""Trade executor service (reference parity:
+services/trade_executor_service.py:401-1428 + TrailingStopManager :55-399).
+
+Subscribes `trading_signals` + `strategy_update`, gates on confidence and
+portfolio-VaR / correlation conditions (:719-787), executes through the
+exchange seam (FakeExchange by default), manages SL/TP/trailing stops, and
+maintains the `holdings` / `active_trades` / `trailing_stops` keys."""
+
+from __future__ import annotations
+
+import json
+import time
+
+from ..bus.schema import Channels, Keys
+from ..utils.exchange import ExchangeInterface
+from .base import Service
+
+
+class TrailingStopManager:
+    """4 trailing strategies (trade_executor_service.py:168-247):
+    percent_based / atr_based / volatility_based / fixed_amount, with an
+    activation threshold and monotonic stop raises (:333-371)."""
+
+    def __init__(self, strategy: str = "percent_based",
+                 trail_pct: float = 0.02, activation_pct: float = 0.01,
+                 fixed_amount: float = 0.0, atr_mult: float = 2.0):
+        self.strategy = strategy
+        self.trail_pct = trail_pct
+        self.activation_pct = activation_pct
+        self.fixed_amount = fixed_amount
+        self.atr_mult = atr_mult
+        self.stops: dict[str, dict] = {}
+
+    def register(self, symbol: str, entry_price: float, stop_price: float,
+                 atr: float = 0.0):
+        self.stops[symbol] = {
+            "entry": entry_price, "stop": stop_price, "peak": entry_price,
+            "atr": atr, "active": False, "updates": 0,
+        }
+
+    def remove(self, symbol: str):
+        self.stops.pop(symbol, None)
+
+    def update(self, symbol: str, price: float,
+               recent_vol: float = 0.0) -> float | None:
+        """Returns a new (raised) stop price or None."""
+        s = self.stops.get(symbol)
+        if s is None:
+            return None
+        s["peak"] = max(s["peak"], price)
+        if not s["active"]:
+            if s["peak"] >= s["entry"] * (1 + self.activation_pct):
+                s["active"] = True
+            else:
+                return None
+        if self.strategy == "percent_based":
+            cand = s["peak"] * (1 - self.trail_pct)
+        elif self.strategy == "atr_based":
+            cand = s["peak"] - self.atr_mult * s["atr"]
+        elif self.strategy == "volatility_based":
+            cand = s["peak"] * (1 - max(recent_vol * 2.0, 0.005))
+        else:  # fixed_amount
+            cand = s["peak"] - self.fixed_amount
+        if cand > s["stop"]:
+            s["stop"] = cand
+            s["updates"] += 1
+            return cand
+        return None
+
+    def status(self) -> dict:
+        return {k: dict(v) for k, v in self.stops.items()}
+
+
+class TradeExecutorService(Service):
+    name = "trade_executor"
+
+    def __init__(self, bus, exchange: ExchangeInterface, config=None):
+        super().__init__(bus, config)
+        self.exchange = exchange
+        self.trailing = TrailingStopManager()
+        self.active: dict[str, dict] = {}     # symbol -> trade record
+        self.strategy_params: dict = {}
+        self.trades_done = 0
+
+    def run_tasks(self):
+        return [self._consume_signals(), self._monitor_trades()]
+
+    # --- gates (trade_executor_service.py:719-787) -----------------------
+    async def check_trading_conditions(self, signal: dict) -> tuple[bool, str]:
+        if len(self.active) >= self.config.trading.max_positions:
+            return False, "max_positions"
+        if signal["symbol"] in self.active:
+            return False, "already_in_position"
+        risk = await self.bus.get_json(Keys.PORTFOLIO_RISK)
+        if risk and risk.get("portfolio_var_pct", 0.0) > \
+                self.config.risk.max_portfolio_var_pct * 100:
+            return False, "portfolio_var_exceeded"
+        if risk:
+            corr_block = risk.get("high_correlation_symbols", [])
+            if signal["symbol"] in corr_block:
+                return False, "correlation_limit"
+        return True, "ok"
+
+    async def _get_risk_info(self, symbol: str) -> dict:
+        sl = await self.bus.get_json(Keys.ADAPTIVE_STOP_LOSSES) or {}
+        return sl.get(symbol, {})
+
+    async def execute_buy(self, signal: dict):
+        sym = signal["symbol"]
+        price = self.exchange.get_ticker(sym)["price"]
+        if price <= 0:
+            return
+        balances = self.exchange.get_balances()
+        quote = self.config.trading.quote_asset
+        cash = balances.get(quote, 0.0)
+        risk_info = signal.get("risk_info", {})
+        pos_pct = risk_info.get("optimal_position_pct",
+                                self.config.risk.fixed_position_pct)
+        # social risk adjustment (trade_executor_service.py:799-814)
+        sra = await self.bus.hget(Keys.SOCIAL_RISK_ADJUSTMENTS, sym)
+        if sra:
+            adj = json.loads(sra)
+            pos_pct *= adj.get("position_multiplier", 1.0)
+        cost = min(cash * pos_pct, cash)
+        if cost < 1e-6:
+            return
+        qty = cost / price
+        order = self.exchange.create_order(sym, "BUY", "MARKET", qty)
+        if order.status != "FILLED":
+            return
+        stop_pct = risk_info.get("adaptive_stop_pct",
+                                 self.config.risk.base_stop_loss_pct)
+        stop = price * (1 - stop_pct)
+        tp = price * (1 + 2 * stop_pct)        # 2:1 RR (PositionSizer :251)
+        # STOP_LOSS_LIMIT at stop*0.99 limit (reference :980)
+        self.exchange.create_order(sym, "SELL", "STOP_LOSS_LIMIT", qty,
+                                   price=stop * 0.99, stop_price=stop)
+        self.trailing.register(sym, price, stop)
+        self.active[sym] = {
+            "symbol": sym, "qty": qty, "entry_price": price,
+            "stop_price": stop, "tp_price": tp,
+            "opened_at": time.time(),
+            "signal_confidence": signal.get("confidence", 0.0),
+        }
+        self.trades_done += 1
+        self.metrics.executions.labels(sym, "BUY").inc()
+        await self._write_state()
+        await self.bus.publish(Channels.TRADE_EXECUTIONS, {
+            "symbol": sym, "side": "BUY", "qty": qty, "price": price,
+        })
+
+    async def execute_sell(self, sym: str, reason: str):
+        trade = self.active.pop(sym, None)
+        if trade is None:
+            return
+        price = self.exchange.get_ticker(sym)["price"]
+        self.exchange.create_order(sym, "SELL", "MARKET", trade["qty"])
+        self.trailing.remove(sym)
+        self.trades_done += 1
+        self.metrics.executions.labels(sym, "SELL").inc()
+        await self._write_state()
+        await self.bus.publish(Channels.TRADE_EXECUTIONS, {
+            "symbol": sym, "side": "SELL", "qty": trade["qty"],
+            "price": price, "reason": reason,
+            "pnl_pct": (price / trade["entry_price"] - 1) * 100,
+        })
+
+    async def _write_state(self):
+        """holdings / active_trades / trailing_stops keys
+        (trade_executor_service.py:709-714, :1212, :1126)."""
+        balances = self.exchange.get_balances()
+        prices = {s: self.exchange.get_ticker(s)["price"]
+                  for s in self.active}
+        total = balances.get(self.config.trading.quote_asset, 0.0)
+        holdings = {}
+        for asset, qty in balances.items():
+            if asset == self.config.trading.quote_asset:
+                holdings[asset] = {"qty": qty, "value": qty}
+            else:
+                p = prices.get(asset + self.config.trading.quote_asset, 0.0)
+                holdings[asset] = {"qty": qty, "value": qty * p}
+                total += qty * p
+        await self.bus.set(Keys.HOLDINGS, {
+            "holdings": holdings, "total_value": total,
+            "timestamp": time.time(),
+        })
+        await self.bus.set(Keys.ACTIVE_TRADES, self.active)
+        await self.bus.set(Keys.TRAILING_STOPS, self.trailing.status())
+        self.metrics.portfolio_value.set(total)
+        self.metrics.active_trades.set(len(self.active))
+
+    # --- loops -----------------------------------------------------------
+    async def _consume_signals(self):
+        sub = self.bus.subscribe(Channels.TRADING_SIGNALS,
+                                 Channels.RISK_ENRICHED_SIGNALS,
+                                 Channels.STRATEGY_UPDATE)
+        min_conf = self.config.trading.min_confidence
+
+        async def on_msg(chan, msg):
+            if chan == Channels.STRATEGY_UPDATE:
+                # 'reload' hot-swap trigger (strategy_evolution:356)
+                params = await self.bus.get_json(Keys.STRATEGY_PARAMS)
+                if params:
+                    self.strategy_params = params
+                    self.log.info("strategy params hot-swapped")
+                return
+            if not isinstance(msg, dict) or "decision" not in msg:
+                return
+            sym = msg.get("symbol")
+            if msg["decision"] == "BUY" and msg.get("confidence", 0) \
+                    >= min_conf:
+                ok, why = await self.check_trading_conditions(msg)
+                if ok:
+                    await self.execute_buy(msg)
+                else:
+                    self.log.debug("buy blocked: %s", why)
+            elif msg["decision"] == "SELL" and sym in self.active:
+                await self.execute_sell(sym, "signal")
+
+        await self.consume(sub, on_msg)
+
+    async def _monitor_trades(self):
+        """SL/TP/trailing monitor loop (reference :1104-1217)."""
+        await self._write_state()          # publish holdings at startup
+        last_state = 0.0
+        while self.running:
+            if time.monotonic() - last_state > 2.0:
+                last_state = time.monotonic()
+                await self._write_state()
+            for sym in list(self.active):
+                trade = self.active.get(sym)
+                if trade is None:
+                    continue
+                price = self.exchange.get_ticker(sym)["price"]
+                if price <= 0:
+                    continue
+                new_stop = self.trailing.update(sym, price)
+                if new_stop is not None:
+                    trade["stop_price"] = new_stop
+                if price <= trade["stop_price"]:
+                    await self.execute_sell(sym, "stop_loss")
+                elif price >= trade["tp_price"]:
+                    await self.execute_sell(sym, "take_profit")
+            await self.sleep(0.1)
+
+    async def run(self):
+        pass

@@ -1,0 +1,205 @@
+"""Control-plane integration tests: bus, schema, services, end-to-end
+pipeline on the in-process bus (CPU)."""
+
+import asyncio
+
+import numpy as np
+import pytest
+
+from ai_crypto_trader_amd.bus.message_bus import InProcessBus
+from ai_crypto_trader_amd.bus.schema import (
+    Channels, Keys, MarketUpdate, SocialUpdate, TradingSignal,
+)
+from ai_crypto_trader_amd.config import AppConfig
+from ai_crypto_trader_amd.data.feed import SyntheticFeed
+from ai_crypto_trader_amd.data.synthetic import candles_chl_v, generate_ohlcv
+from ai_crypto_trader_amd.utils.circuit_breaker import (
+    CircuitBreaker, CircuitOpenError,
+)
+from ai_crypto_trader_amd.utils.exchange import FakeExchange
+from ai_crypto_trader_amd.utils.indicator_combinations import (
+    calculate_indicator_combinations,
+)
+from ai_crypto_trader_amd.utils.rate_limiter import SlidingWindowLimiter
+from ai_crypto_trader_amd.utils.volume_profile import VolumeProfileAnalyzer
+
+
+# ------------------------------ bus ----------------------------------------
+
+def test_bus_pubsub_and_keys():
+    async def go():
+        bus = InProcessBus()
+        sub = bus.subscribe("market_updates", "trading_*")
+        await bus.publish("market_updates", {"a": 1})
+        await bus.publish("trading_signals", {"b": 2})
+        await bus.publish("other", {"c": 3})
+        c1, m1 = await sub.get(timeout=1)
+        c2, m2 = await sub.get(timeout=1)
+        assert (c1, m1) == ("market_updates", {"a": 1})
+        assert (c2, m2) == ("trading_signals", {"b": 2})
+        assert sub.queue.empty()
+
+        await bus.set("k", {"x": 1})
+        assert await bus.get_json("k") == {"x": 1}
+        await bus.hset("h", "f", 2.5)
+        assert await bus.hget("h", "f") == "2.5"
+        assert "k" in await bus.keys("*")
+
+    asyncio.run(go())
+
+
+def test_schema_payload_shapes():
+    u = MarketUpdate(symbol="BTCUSDC", current_price=1.0, avg_volume=2.0)
+    d = u.to_dict()
+    # reference field names (market_monitor_service.py:445-524)
+    for k in ("symbol", "current_price", "avg_volume", "rsi", "macd",
+              "bb_position", "trend", "trend_strength", "price_change_1m"):
+        assert k in d
+    s = SocialUpdate(symbol="BTCUSDC").to_dict()
+    assert "data" in s and "metrics" in s["data"]
+    assert "sentiment" in s["data"]["metrics"]
+    t = TradingSignal(symbol="X", decision="BUY", confidence=0.9).to_dict()
+    for k in ("decision", "confidence", "reasoning", "risk_level",
+              "explanation", "factor_weights", "model_version"):
+        assert k in t
+
+
+# ------------------------------ utils --------------------------------------
+
+def test_circuit_breaker_state_machine():
+    br = CircuitBreaker("t", failure_threshold=2, recovery_timeout=0.05)
+
+    def boom():
+        raise ValueError()
+
+    for _ in range(2):
+        with pytest.raises(ValueError):
+            br.call(boom)
+    assert br.state.value == "open"
+    with pytest.raises(CircuitOpenError):
+        br.call(lambda: 1)
+    import time
+    time.sleep(0.06)
+    assert br.call(lambda: 42) == 42        # half-open -> closed
+    assert br.state.value == "closed"
+
+
+def test_rate_limiter():
+    lim = SlidingWindowLimiter(3, window_s=60)
+    assert all(lim.allow() for _ in range(3))
+    assert not lim.allow()
+    assert lim.remaining() == 0
+
+
+def test_fake_exchange_fills():
+    ex = FakeExchange(initial_balance=1000.0)
+    ex.set_price("BTCUSDC", 100.0)
+    o = ex.create_order("BTCUSDC", "BUY", "MARKET", 5.0)
+    assert o.status == "FILLED"
+    assert ex.get_balances()["USDC"] == pytest.approx(500.0)
+    assert ex.get_balances()["BTC"] == pytest.approx(5.0 * 0.999)
+    # stop-loss fills when price crosses
+    ex.create_order("BTCUSDC", "SELL", "STOP_LOSS_LIMIT",
+                    ex.get_balances()["BTC"], price=89.0, stop_price=90.0)
+    ex.set_price("BTCUSDC", 89.5)
+    assert ex.get_balances().get("BTC", 0.0) == 0.0
+    assert ex.portfolio_value() < 1000.0     # fees + stop loss
+
+
+def test_indicator_combinations_shape():
+    u = MarketUpdate(symbol="S", current_price=1.0, avg_volume=1.0,
+                     rsi=25.0, stoch_k=15.0, williams_r=-85.0,
+                     macd=0.1, macd_3m=0.1, macd_5m=0.2).to_dict()
+    c = calculate_indicator_combinations(u)
+    assert c["trend_confirmation"]["signal"] == "bullish"
+    assert c["oscillator_consensus"]["signal"] == "oversold"
+    for key in ("triple_ma", "double_rsi", "market_regime",
+                "reversal_probability", "breakout_confirmation",
+                "divergence"):
+        assert key in c
+
+
+def test_volume_profile():
+    c = candles_chl_v(generate_ohlcv(500, 1, seed=3))[0]
+    vp = VolumeProfileAnalyzer().analyze(c)
+    assert vp["value_area_low"] <= vp["poc"] <= vp["value_area_high"]
+    assert -1 <= vp["volume_delta"] <= 1
+    assert vp["signal"] in ("above_value_area", "below_value_area",
+                            "at_poc", "inside_value_area")
+
+
+# --------------------------- pipeline --------------------------------------
+
+@pytest.mark.timeout(240)
+def test_pipeline_end_to_end():
+    """Market monitor -> analyzer -> risk -> executor over the in-process
+    bus: the reference topology's 'forward pass' (SURVEY.md §3.1)."""
+    from ai_crypto_trader_amd.services.analyzer import AnalyzerService
+    from ai_crypto_trader_amd.services.market_monitor import (
+        MarketMonitorService,
+    )
+    from ai_crypto_trader_amd.services.market_regime import (
+        MarketRegimeService,
+    )
+    from ai_crypto_trader_amd.services.portfolio_risk import (
+        PortfolioRiskService,
+    )
+    from ai_crypto_trader_amd.services.trade_executor import (
+        TradeExecutorService,
+    )
+
+    async def go():
+        symbols = ["BTCUSDC", "ETHUSDC"]
+        cfg = AppConfig()
+        cfg.trading.symbols = symbols
+        cfg.trading.ai_analysis_interval = 0.0      # analyze every update
+        cfg.trading.min_confidence = 0.2            # trade eagerly
+        bus = InProcessBus()
+        market = candles_chl_v(generate_ohlcv(1500, 2, seed=23, sigma=2.5))
+        feed = SyntheticFeed(market, symbols)
+        ex = FakeExchange()
+        monitor = MarketMonitorService(bus, feed, cfg)
+        orig_push = monitor._push
+
+        def push_and_tick(c):
+            ex.set_price(c.symbol, c.close)
+            return orig_push(c)
+
+        monitor._push = push_and_tick
+        services = [
+            monitor, AnalyzerService(bus, cfg),
+            PortfolioRiskService(bus, cfg),
+            MarketRegimeService(bus, cfg),
+            TradeExecutorService(bus, ex, cfg),
+        ]
+        for s in services:
+            await s.start()
+        # run until the feed is exhausted
+        while monitor.running:
+            await asyncio.sleep(0.2)
+        await asyncio.sleep(6.0)       # let the periodic loops flush
+        state = {
+            "holdings": await bus.get_json(Keys.HOLDINGS),
+            "regime": await bus.get_json(Keys.CURRENT_MARKET_REGIME),
+            "risk": await bus.get_json(Keys.PORTFOLIO_RISK),
+            "prices": await bus.hgetall(Keys.CURRENT_PRICES),
+            "monitor": monitor.updates_published,
+            "signals": services[1].signals_published,
+            "trades": services[-1].trades_done,
+        }
+        for s in services:
+            assert s.health()["healthy"], s.name
+            await s.stop()
+        return state
+
+    state = asyncio.run(go())
+    assert state["monitor"] > 1000
+    assert state["signals"] > 50
+    assert state["holdings"] is not None
+    assert state["holdings"]["total_value"] > 0
+    assert state["regime"] is not None
+    assert state["regime"]["regime"] in ("bull", "bear", "ranging",
+                                         "volatile")
+    assert state["risk"] is not None and "portfolio_var" in state["risk"]
+    assert len(state["prices"]) == 2
+    assert state["trades"] >= 1        # eager confidence gate must trade
