@@ -1,0 +1,256 @@
+"""Second service batch: social/news/order-book/arbitrage/grid/DCA/
+registry/explainability/feature-importance/selection/patterns."""
+
+import asyncio
+import time
+
+import numpy as np
+import pytest
+
+from ai_crypto_trader_amd.bus.message_bus import InProcessBus
+from ai_crypto_trader_amd.config import AppConfig
+from ai_crypto_trader_amd.services.arbitrage import PairGraph
+from ai_crypto_trader_amd.services.news import (
+    NewsAnalyzer, SyntheticNewsSource,
+)
+from ai_crypto_trader_amd.services.order_book import OrderBookAnalyzer
+from ai_crypto_trader_amd.services.registry import (
+    AIExplainabilityService, FeatureImportanceIntegrator,
+)
+from ai_crypto_trader_amd.services.social import SocialMetricsAnalyzer
+from ai_crypto_trader_amd.services.strategy_selection import (
+    StrategySelectionService,
+)
+from ai_crypto_trader_amd.utils.exchange import FakeExchange
+
+
+# ------------------------------- social ------------------------------------
+
+def test_lead_lag_recovers_known_lead():
+    """Sentiment constructed to lead returns by 3 steps must be detected
+    with a positive-lag peak."""
+    rng = np.random.default_rng(0)
+    an = SocialMetricsAnalyzer(max_lag=10)
+    n = 600
+    ret = rng.standard_normal(n) * 0.001
+    price = np.exp(np.cumsum(ret))
+    lead = 3
+    for t in range(n - lead):
+        sent = 0.5 + 40 * ret[t + lead] + rng.standard_normal() * 0.005
+        an.record("X", float(sent), float(price[t]), ts=float(t))
+    ll = an.lead_lag("X")
+    # analyzer convention: lag k pairs s[t] with the return over candle
+    # (t+k, t+k+1]; a sentiment coupled to the return INTO t+3 peaks at k=2
+    assert ll["lag"] == lead - 1
+    assert ll["pearson"] > 0.5
+    assert abs(ll["spearman"]) > 0.3
+
+
+def test_decayed_sentiment_weighting():
+    an = SocialMetricsAnalyzer(half_life_h=1.0)
+    now = 1_000_000.0
+    an.record("X", 0.9, 1.0, ts=now - 7200)    # 2 half-lives old
+    an.record("X", 0.1, 1.0, ts=now)
+    s = an.decayed_sentiment("X", now=now)
+    assert 0.1 < s < 0.5        # recent bearish dominates
+    assert s < 0.3
+
+
+# ------------------------------- news --------------------------------------
+
+def test_news_sentiment_and_entities():
+    na = NewsAnalyzer()
+    assert na.sentiment("Bitcoin surges to record high on ETF approval") > 0.6
+    assert na.sentiment("Massive hack triggers crash and liquidations") < 0.4
+    assert abs(na.sentiment("Bitcoin trades sideways") - 0.5) < 0.15
+    ents = na.entities("Ethereum and $SOL rally while Bitcoin dips")
+    assert "ETH" in ents and "SOL" in ents and "BTC" in ents
+    assert "regulation" in na.topics("SEC lawsuit threatens approval")
+    assert na.relevance("Bitcoin surges", "BTCUSDC") >= 0.7
+
+
+def test_synthetic_news_deterministic():
+    a = SyntheticNewsSource(seed=1).headlines("BTCUSDC", 3)
+    b = SyntheticNewsSource(seed=1).headlines("BTCUSDC", 3)
+    assert a == b
+
+
+# ----------------------------- order book ----------------------------------
+
+def test_order_book_analyzer():
+    ex = FakeExchange()
+    ex.prices["BTCUSDC"] = 100.0
+    book = ex.get_order_book("BTCUSDC", limit=50)
+    a = OrderBookAnalyzer().analyze(book)
+    assert a["ok"]
+    assert a["spread_bps"] > 0
+    assert -1 <= a["imbalance"] <= 1
+    assert a["price_impact"]["10000"]["buy_impact_bps"] >= 0
+    assert len(a["levels"]["support"]) >= 1
+    assert 0 <= a["microstructure"]["gini"] <= 1
+    assert a["signal"]["direction"] in ("bullish", "bearish", "neutral")
+
+    # skewed book -> buy pressure
+    book["bids"] = [[99.9, 100.0], [99.8, 100.0]]
+    book["asks"] = [[100.1, 1.0], [100.2, 1.0]]
+    a2 = OrderBookAnalyzer().analyze(book)
+    assert a2["pressure"] == "buy" and a2["imbalance"] > 0.5
+
+
+# ----------------------------- arbitrage -----------------------------------
+
+def test_triangle_arbitrage_detection():
+    g = PairGraph(fee=0.0)
+    # consistent prices -> no profit
+    g.add_pair("BTC", "USDC", 100.0)
+    g.add_pair("ETH", "USDC", 10.0)
+    g.add_pair("ETH", "BTC", 0.1)
+    cycles = g.cycles("USDC")
+    assert cycles, "triangle must be found"
+    assert all(abs(g.cycle_profit(c) - 1.0) < 1e-9 for c in cycles)
+    # mispriced ETH/BTC -> profitable cycle exists
+    g.add_pair("ETH", "BTC", 0.095)
+    profits = [g.cycle_profit(c) for c in g.cycles("USDC")]
+    assert max(profits) > 1.02
+    # fees eat it
+    g.fee = 0.02
+    assert max(g.cycle_profit(c) for c in g.cycles("USDC")) < 1.0
+
+
+# ----------------------------- grid / dca ----------------------------------
+
+def test_grid_strategy_fills():
+    async def go():
+        bus = InProcessBus()
+        from ai_crypto_trader_amd.services.grid_dca import (
+            GridTradingStrategy,
+        )
+        g = GridTradingStrategy(bus, FakeExchange(), "BTCUSDC", AppConfig())
+        g.build_grid(1.0)
+        assert len(g.levels) >= 8
+        buys = [lv for lv in g.levels if lv["side"] == "BUY"]
+        sells = [lv for lv in g.levels if lv["side"] == "SELL"]
+        assert buys and sells
+        fills = g.on_price(min(lv["price"] for lv in buys) - 1e-6)
+        assert any(f["side"] == "BUY" for f in fills)
+        fills = g.on_price(max(lv["price"] for lv in sells) + 1e-6)
+        assert any(f["side"] == "SELL" for f in fills)
+        assert g.pnl > 0
+        # breakout rebuilds around the new price
+        g.on_price(2.0)
+        assert abs(g.center - 2.0) < 1e-9
+
+    asyncio.run(go())
+
+
+def test_dca_strategy_schedule_and_dip():
+    async def go():
+        bus = InProcessBus()
+        from ai_crypto_trader_amd.services.grid_dca import DCAStrategy
+        cfg = AppConfig()
+        d = DCAStrategy(bus, FakeExchange(), "BTCUSDC", cfg,
+                        candles_per_period=10)
+        for i in range(1, 10):
+            assert await d.maybe_buy(1.0, 0.5, "ranging") is None or i == 10
+        rec = await d.maybe_buy(1.0, 0.5, "ranging")     # 10th tick
+        assert rec is not None and not rec["dip"]
+        # build history then dip
+        for _ in range(130):
+            d.recent.append(1.0)
+        rec2 = await d.maybe_buy(1.0 * (1 - cfg.dca.dip_threshold_pct - 0.01),
+                                 0.5, "ranging")
+        assert rec2 is not None and rec2["dip"]
+        assert rec2["usd"] > rec["usd"]                  # dip multiplier
+
+    asyncio.run(go())
+
+
+# ------------------------ registry / explainability ------------------------
+
+def test_model_registry_roundtrip(tmp_path):
+    async def go():
+        bus = InProcessBus()
+        from ai_crypto_trader_amd.services.registry import (
+            ModelRegistryService,
+        )
+        reg = ModelRegistryService(bus, AppConfig(),
+                                   store_path=str(tmp_path / "r.json"))
+        a = await reg.register("lstm-btc", "lstm")
+        b = await reg.register("ga-strat", "strategy")
+        await reg.update_performance(a, {"sharpe": 1.5})
+        await reg.update_performance(b, {"sharpe": 0.5})
+        best = reg.best_model(metric="sharpe")
+        assert best["id"] == a
+        cmp_ = reg.compare([a, b])
+        assert cmp_[0]["id"] == a
+        await reg.set_status(a, "retired")
+        assert reg.best_model(metric="sharpe")["id"] == b
+        # persistence
+        reg2 = ModelRegistryService(bus, AppConfig(),
+                                    store_path=str(tmp_path / "r.json"))
+        assert a in reg2.models
+
+    asyncio.run(go())
+
+
+def test_explanation_formatting():
+    sig = {"symbol": "X", "decision": "BUY", "confidence": 0.8,
+           "explanation": {"momentum": 0.5, "trend": -0.1},
+           "factor_weights": {"momentum": 0.6, "trend": 0.4}}
+    ex = AIExplainabilityService.format_explanation(sig)
+    assert ex["factors"][0]["factor"] == "momentum"
+    assert ex["factors"][0]["contribution"] == pytest.approx(0.3)
+
+
+def test_feature_importance_integrator():
+    fi = FeatureImportanceIntegrator()
+    fi.update({"importances": {
+        "rsi": {"permutation": 0.6, "impurity": 0.5},
+        "macd": {"permutation": 0.2, "impurity": 0.3},
+        "price_change_5m": {"permutation": 0.2, "impurity": 0.2},
+    }, "at": time.time()})
+    assert fi.weights["rsi"] == pytest.approx(0.6)
+    p = fi.predict_outcome({"rsi": 40.0, "macd": 1.0})
+    assert 0 <= p <= 1
+    w = fi.adjust_factor_weights({"oscillators": 0.25, "momentum": 0.25,
+                                  "trend": 0.25, "social_sentiment": 0.25})
+    assert abs(sum(w.values()) - 1.0) < 1e-6
+
+
+# -------------------------- strategy selection -----------------------------
+
+def test_strategy_selection_scoring():
+    bus = InProcessBus()
+    svc = StrategySelectionService(bus, AppConfig())
+    best_bull, _, scores = svc.select_optimal("bull", 0.8, 0.2, hour=14)
+    assert best_bull == "momentum"
+    best_vol, _, _ = svc.select_optimal("volatile", 0.5, 0.9, hour=14)
+    assert best_vol in ("conservative", "dca_strategy")
+    # hysteresis: small improvements don't switch
+    svc.current = best_bull
+    assert not svc.should_switch(best_bull, scores)
+    low = {k: v * 0.99 for k, v in scores.items()}
+    low[best_bull] = scores[best_bull]
+    assert not svc.should_switch(max(low, key=low.get), low)
+
+
+# ------------------------------ patterns -----------------------------------
+
+def test_pattern_model_classifies_synthetic():
+    from ai_crypto_trader_amd.models.patterns import (
+        PATTERNS, PatternRecognitionModel, generate_pattern,
+    )
+
+    m = PatternRecognitionModel("cpu", seed=0)
+    acc = m.train(epochs=6, n_per_class=48, seed=1)
+    assert acc > 0.8, acc
+    rng = np.random.default_rng(99)
+    hits = 0
+    trials = 20
+    for i in range(trials):
+        name = PATTERNS[i % (len(PATTERNS) - 1)]
+        det = m.detect(generate_pattern(name, rng) * 100 + 50)
+        hits += det["pattern"] == name
+    assert hits / trials > 0.5
+    det = m.detect(generate_pattern("double_top", rng) * 100 + 50)
+    assert det["signal"] in ("bearish", "neutral", "bullish")
