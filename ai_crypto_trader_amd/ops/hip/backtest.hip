@@ -218,13 +218,18 @@ __global__ void __launch_bounds__(BT_BLOCK) backtest_kernel(
             }
 
             // --- 4. mark to market -----------------------------------
-            float new_eq = cash + units * close;
-            float r = new_eq / equity - 1.0f;
-            sum_ret += r;
-            sum_ret2 += r * r;
-            equity = new_eq;
-            max_eq = fmaxf(max_eq, equity);
-            max_dd = fmaxf(max_dd, (max_eq - equity) / max_eq);
+            // flat lanes: new_eq == cash == equity exactly, so r == 0 and
+            // every accumulator is unchanged — skipping is bit-identical
+            // to engine_cpu.py (which computes r = cash/cash - 1 = 0)
+            if (units != 0.0f || cash != equity) {
+                float new_eq = cash + units * close;
+                float r = new_eq / equity - 1.0f;
+                sum_ret += r;
+                sum_ret2 += r * r;
+                equity = new_eq;
+                max_eq = fmaxf(max_eq, equity);
+                max_dd = fmaxf(max_dd, (max_eq - equity) / max_eq);
+            }
         }
     }
 
