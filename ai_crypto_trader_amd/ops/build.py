@@ -25,6 +25,7 @@ SOURCES = [
     "covar.hip",
     "indicators.hip",
     "lstm.hip",
+    "gru.hip",
     "rl_env.hip",
     "bindings.cpp",
 ]

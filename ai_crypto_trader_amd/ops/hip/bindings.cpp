@@ -35,6 +35,12 @@ extern "C" void launch_mfma_gemm_test(const void*, const void*, float*, int,
                                       int, int, hipStream_t);
 extern "C" void launch_mfma_gemm_test_bf16(const void*, const void*, float*,
                                            int, int, int, hipStream_t);
+extern "C" void launch_gru_seq_fwd(const void*, const void*, const float*,
+                                   void*, void*, float*, int, int, int,
+                                   hipStream_t);
+extern "C" void launch_gru_seq_bwd(const float*, const void*, const float*,
+                                   const void*, const void*, void*, int, int,
+                                   int, hipStream_t);
 extern "C" void launch_env_reset(const float*, float*, float*, int, int, int,
                                  int, uint64_t, uint64_t, hipStream_t);
 extern "C" void launch_env_step(const float*, float*, const int*, float*,
@@ -203,6 +209,34 @@ PYBIND11_MODULE(_hip_ops, m) {
                          reinterpret_cast<float*>(returns), T, E, gamma, lam,
                          as_stream(stream));
               check(hipGetLastError(), "gae launch");
+          });
+
+    m.def("gru_seq_fwd",
+          [](uintptr_t xproj, uintptr_t Wt, uintptr_t bias, uintptr_t h_out,
+             uintptr_t gates_out, uintptr_t hpn_out, int B, int T, int H,
+             uintptr_t stream) {
+              launch_gru_seq_fwd(reinterpret_cast<const void*>(xproj),
+                                 reinterpret_cast<const void*>(Wt),
+                                 reinterpret_cast<const float*>(bias),
+                                 reinterpret_cast<void*>(h_out),
+                                 reinterpret_cast<void*>(gates_out),
+                                 reinterpret_cast<float*>(hpn_out), B, T, H,
+                                 as_stream(stream));
+              check(hipGetLastError(), "gru_seq_fwd launch");
+          });
+
+    m.def("gru_seq_bwd",
+          [](uintptr_t dh_up, uintptr_t gates, uintptr_t hpn,
+             uintptr_t h_out, uintptr_t W, uintptr_t dgates_out, int B,
+             int T, int H, uintptr_t stream) {
+              launch_gru_seq_bwd(reinterpret_cast<const float*>(dh_up),
+                                 reinterpret_cast<const void*>(gates),
+                                 reinterpret_cast<const float*>(hpn),
+                                 reinterpret_cast<const void*>(h_out),
+                                 reinterpret_cast<const void*>(W),
+                                 reinterpret_cast<void*>(dgates_out), B, T,
+                                 H, as_stream(stream));
+              check(hipGetLastError(), "gru_seq_bwd launch");
           });
 
     m.def("device_synchronize", []() { check(hipDeviceSynchronize(), "sync"); });
