@@ -125,21 +125,12 @@ __global__ void __launch_bounds__(BT_BLOCK) backtest_kernel(
         }
         __syncthreads();
         const int tend = min(BT_TILE, T - t0);
-        // Software-pipelined LDS reads: with pop=1024 x 64 symbols the
-        // launch is exactly 1 wave/SIMD, so nothing else hides the
-        // ~100-cycle LDS latency — prefetch candle tt+1's values behind
-        // candle tt's ~130-instruction compute (next_* rotate at the end).
-        float close = chist[BT_HALO];
-        float high = hl[BT_HALO][0];
-        float low = hl[BT_HALO][1];
-        float oldc = chist[BT_HALO - q.bb_w];
         for (int tt = 0; tt < tend; ++tt) {
             const int t = t0 + tt;
-            const int nx = (tt + 1 < tend ? tt + 1 : tt) + BT_HALO;
-            const float n_close = chist[nx];
-            const float n_high = hl[nx][0];
-            const float n_low = hl[nx][1];
-            const float n_oldc = chist[nx - q.bb_w];
+            const float close = chist[tt + BT_HALO];
+            const float high = hl[tt + BT_HALO][0];
+            const float low = hl[tt + BT_HALO][1];
+            const float oldc = chist[tt + BT_HALO - q.bb_w];
 
             // --- 1. indicators ---------------------------------------
             float change;
@@ -240,11 +231,6 @@ __global__ void __launch_bounds__(BT_BLOCK) backtest_kernel(
                 max_eq = fmaxf(max_eq, equity);
                 max_dd = fmaxf(max_dd, (max_eq - equity) / max_eq);
             }
-
-            close = n_close;
-            high = n_high;
-            low = n_low;
-            oldc = n_oldc;
         }
     }
 
