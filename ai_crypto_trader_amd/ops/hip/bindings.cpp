@@ -22,6 +22,9 @@ extern "C" void launch_ga_evolve(const float*, const float*, const int*,
 extern "C" void launch_mc_paths(const float*, const float*, const float*,
                                 const float*, float*, float*, int, int, long,
                                 float, uint64_t, hipStream_t);
+extern "C" void launch_mc_paths_mfma(const float*, const float*,
+                                     const float*, float*, float*, int, int,
+                                     long, float, uint64_t, hipStream_t);
 extern "C" void launch_cov(const float*, float*, int, int, hipStream_t);
 extern "C" void launch_indicators(const float*, float*, int, int, int,
                                   hipStream_t);
@@ -104,6 +107,21 @@ PYBIND11_MODULE(_hip_ops, m) {
                               reinterpret_cast<float*>(max_dd), n_assets,
                               n_steps, n_paths, s0, seed, as_stream(stream));
               check(hipGetLastError(), "mc_paths launch");
+          });
+
+    m.def("mc_paths_mfma",
+          [](uintptr_t chol, uintptr_t drift, uintptr_t vol_sqrt_dt,
+             uintptr_t final_value, uintptr_t max_dd, int n_assets,
+             int n_steps, long n_paths, float s0, uint64_t seed,
+             uintptr_t stream) {
+              launch_mc_paths_mfma(reinterpret_cast<const float*>(chol),
+                                   reinterpret_cast<const float*>(drift),
+                                   reinterpret_cast<const float*>(vol_sqrt_dt),
+                                   reinterpret_cast<float*>(final_value),
+                                   reinterpret_cast<float*>(max_dd),
+                                   n_assets, n_steps, n_paths, s0, seed,
+                                   as_stream(stream));
+              check(hipGetLastError(), "mc_paths_mfma launch");
           });
 
     m.def("cov",

@@ -122,13 +122,23 @@ def mc_paths_cpu(
 
 def mc_paths_gpu(
     chol, mu, sigma, weights, *, n_steps: int, n_paths: int, dt: float,
-    s0: float = 1.0, seed: int = 0, device="cuda",
+    s0: float = 1.0, seed: int = 0, device="cuda", use_mfma: bool | None = None,
 ):
-    """GPU path generation; returns (final_value, max_dd) torch tensors."""
+    """GPU path generation; returns (final_value, max_dd) torch tensors.
+
+    use_mfma=None (auto): the bf16 MFMA pathgen kernel when A==64 and
+    n_paths is a multiple of 256, else the exact f32 VALU kernel. The two
+    draw identical Philox normals; MFMA quantizes Z/CVOL to bf16
+    (statistics agree to ~1e-2, see tests)."""
     import torch
 
     ops = require_hip_ops()
     A = int(chol.shape[0])
+    if use_mfma is None:
+        use_mfma = (A == 64)
+    if use_mfma and A != 64:
+        use_mfma = False
+    pad = ((-n_paths) % 256) if use_mfma else 0   # MFMA tile = 256 paths
     f32 = np.float32
     cvol = (np.asarray(sigma)[:, None] * np.asarray(chol) * np.sqrt(dt))
     cvol_k_major = np.ascontiguousarray(cvol.T, dtype=f32)
@@ -139,13 +149,23 @@ def mc_paths_gpu(
     t_cvol = torch.from_numpy(cvol_k_major).to(device)
     t_drift = torch.from_numpy(drift).to(device)
     t_w = torch.from_numpy(wS0).to(device)
-    fv = torch.empty(n_paths, dtype=torch.float32, device=device)
-    dd = torch.empty(n_paths, dtype=torch.float32, device=device)
+    fv = torch.empty(n_paths + pad, dtype=torch.float32, device=device)
+    dd = torch.empty(n_paths + pad, dtype=torch.float32, device=device)
     stream = torch.cuda.current_stream(fv.device).cuda_stream
-    ops.mc_paths(
-        t_cvol.data_ptr(), t_drift.data_ptr(), t_w.data_ptr(), 0,
-        fv.data_ptr(), dd.data_ptr(), A, n_steps, n_paths, v0, seed, stream,
-    )
+    if use_mfma:
+        ops.mc_paths_mfma(
+            t_cvol.data_ptr(), t_drift.data_ptr(), t_w.data_ptr(),
+            fv.data_ptr(), dd.data_ptr(), A, n_steps, n_paths + pad, v0,
+            seed, stream,
+        )
+        if pad:
+            fv, dd = fv[:n_paths], dd[:n_paths]
+    else:
+        ops.mc_paths(
+            t_cvol.data_ptr(), t_drift.data_ptr(), t_w.data_ptr(), 0,
+            fv.data_ptr(), dd.data_ptr(), A, n_steps, n_paths, v0, seed,
+            stream,
+        )
     return fv, dd
 
 

@@ -60,8 +60,19 @@ def bench_mc(reps):
     sig = np.full(A, 0.5)
     w = np.full(A, 1.0 / A)
     dt = timed(lambda: mc_paths_gpu(chol, mu, sig, w, n_steps=steps,
-                                    n_paths=paths, dt=1 / 252, seed=2),
+                                    n_paths=paths, dt=1 / 252, seed=2,
+                                    use_mfma=False),
                reps)
+    dt_m = timed(lambda: mc_paths_gpu(chol, mu, sig, w, n_steps=steps,
+                                      n_paths=paths, dt=1 / 252, seed=2,
+                                      use_mfma=True),
+                 reps)
+    print(json.dumps({"kernel": "mc_paths_mfma", "ms": dt_m * 1e3,
+                      "paths_per_sec": paths / dt_m,
+                      "asset_steps_per_sec": paths * A * steps / dt_m,
+                      "mfma_tflops_bf16":
+                          paths * steps * A * A * 2 / dt_m / 1e12}),
+          flush=True)
     return {"kernel": "mc_paths", "ms": dt * 1e3,
             "paths_per_sec": paths / dt,
             "asset_steps_per_sec": paths * A * steps / dt,
