@@ -54,6 +54,11 @@ def parse_args():
     ap.add_argument("--gpu", action="store_true",
                     help="use cuda:0 for the numeric services")
     ap.add_argument("--status-interval", type=float, default=5.0)
+    ap.add_argument("--core-only", action="store_true",
+                    help="run only the 8 core services (default: the full "
+                         "17+-service topology)")
+    ap.add_argument("--dashboard", action="store_true",
+                    help="serve the dashboard API on :8050")
     return ap.parse_args()
 
 
@@ -133,8 +138,62 @@ async def main():
         StrategyEvolutionService(bus, cfg, candles=market, device=device),
         TradeExecutorService(bus, exchange, cfg),
     ]
+    if not args.core_only:
+        # full reference topology
+        from ai_crypto_trader_amd.services.arbitrage import (
+            ArbitrageDetectionService,
+        )
+        from ai_crypto_trader_amd.services.grid_dca import (
+            DCAStrategy, GridTradingStrategy,
+        )
+        from ai_crypto_trader_amd.services.news import NewsAnalysisService
+        from ai_crypto_trader_amd.services.order_book import (
+            OrderBookAnalysisService,
+        )
+        from ai_crypto_trader_amd.services.pattern_recognition import (
+            PatternRecognitionService,
+        )
+        from ai_crypto_trader_amd.services.registry import (
+            AIExplainabilityService, FeatureImportanceAnalyzer,
+            ModelRegistryService,
+        )
+        from ai_crypto_trader_amd.services.social import (
+            EnhancedSocialMonitorService, SocialMonitorService,
+            SocialRiskAdjuster,
+        )
+        from ai_crypto_trader_amd.services.strategy_selection import (
+            StrategySelectionService,
+        )
+        quote = cfg.trading.quote_asset
+        pairs = [(s[: -len(quote)], quote) for s in symbols]
+        services += [
+            SocialMonitorService(bus, cfg),
+            EnhancedSocialMonitorService(bus, cfg),
+            SocialRiskAdjuster(bus, cfg),
+            NewsAnalysisService(bus, cfg),
+            OrderBookAnalysisService(bus, exchange, cfg),
+            PatternRecognitionService(bus, cfg, device=device),
+            StrategySelectionService(bus, cfg),
+            ModelRegistryService(bus, cfg),
+            AIExplainabilityService(bus, cfg),
+            FeatureImportanceAnalyzer(bus, cfg),
+            GridTradingStrategy(bus, exchange, symbols[0], cfg),
+            DCAStrategy(bus, exchange, symbols[0], cfg),
+            ArbitrageDetectionService(bus, exchange, cfg, pairs=pairs),
+        ]
     for s in services:
         await s.start()
+
+    dash_server = None
+    if args.dashboard:
+        import uvicorn
+
+        from dashboard import DataStore, build_app
+        store = DataStore(bus)
+        app = build_app(bus, store)
+        dash_server = uvicorn.Server(uvicorn.Config(
+            app, host="127.0.0.1", port=8050, log_level="warning"))
+        asyncio.create_task(dash_server.serve())
     status = asyncio.create_task(
         print_status(bus, services, args.status_interval))
 
