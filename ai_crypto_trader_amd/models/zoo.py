@@ -121,19 +121,26 @@ class PositionalEncoding(nn.Module):
 
 
 class TransformerPredictor(nn.Module):
+    """2-block encoder on the fused gfx950 attention kernel
+    (models/attention.py; torch SDPA on CPU)."""
+
     def __init__(self, n_features: int = 9, d_model: int = 64,
                  n_heads: int = 4, n_layers: int = 2):
         super().__init__()
+        from .attention import FusedTransformerEncoderLayer
+
         self.proj = nn.Linear(n_features, d_model)
         self.pos = PositionalEncoding(d_model)
-        layer = nn.TransformerEncoderLayer(
-            d_model, n_heads, dim_feedforward=128, batch_first=True,
-            dropout=0.0)
-        self.enc = nn.TransformerEncoder(layer, n_layers)
+        self.layers = nn.ModuleList([
+            FusedTransformerEncoderLayer(d_model, n_heads, 128)
+            for _ in range(n_layers)
+        ])
         self.head = _Head(d_model)
 
     def forward(self, x):
-        h = self.enc(self.pos(self.proj(x)))
+        h = self.pos(self.proj(x))
+        for layer in self.layers:
+            h = layer(h)
         return self.head(h[:, -1]).squeeze(-1)
 
 

@@ -26,6 +26,7 @@ SOURCES = [
     "indicators.hip",
     "lstm.hip",
     "gru.hip",
+    "attention.hip",
     "rl_env.hip",
     "bindings.cpp",
 ]

@@ -44,6 +44,8 @@ extern "C" void launch_gru_seq_fwd(const void*, const void*, const float*,
 extern "C" void launch_gru_seq_bwd(const float*, const void*, const float*,
                                    const void*, const void*, void*, int, int,
                                    int, hipStream_t);
+extern "C" void launch_attn_fwd(const void*, const void*, const void*,
+                                void*, long, int, int, float, hipStream_t);
 extern "C" void launch_env_reset(const float*, float*, float*, int, int, int,
                                  int, uint64_t, uint64_t, hipStream_t);
 extern "C" void launch_env_step(const float*, float*, const int*, float*,
@@ -255,6 +257,18 @@ PYBIND11_MODULE(_hip_ops, m) {
                                  reinterpret_cast<void*>(dgates_out), B, T,
                                  H, as_stream(stream));
               check(hipGetLastError(), "gru_seq_bwd launch");
+          });
+
+    m.def("attn_fwd",
+          [](uintptr_t Q, uintptr_t K, uintptr_t V, uintptr_t O,
+             long bh_count, int s_len, int d_head, float scale,
+             uintptr_t stream) {
+              launch_attn_fwd(reinterpret_cast<const void*>(Q),
+                              reinterpret_cast<const void*>(K),
+                              reinterpret_cast<const void*>(V),
+                              reinterpret_cast<void*>(O), bh_count, s_len,
+                              d_head, scale, as_stream(stream));
+              check(hipGetLastError(), "attn_fwd launch");
           });
 
     m.def("device_synchronize", []() { check(hipDeviceSynchronize(), "sync"); });

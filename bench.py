@@ -95,9 +95,14 @@ def main():
     ohlcv = generate_ohlcv(T, nsym, seed=args.seed)
     candles = candles_chl_v(ohlcv)
 
+    # fitness = time-segmented CV over 8 segments/symbol (see GAEngine
+    # docstring: standard GA anti-overfit practice; also fills the chip —
+    # pop x symbols alone is exactly 1 wave/SIMD). Candle-eval totals are
+    # unchanged: every candle of every symbol is backtested each step.
+    segments = 8 if (on_gpu and T % 8 == 0) else 1
     engine = GAEngine(
         candles, pop_per_rank=pop, rank=rank, world=world, device=device,
-        seed=args.seed + 1,
+        seed=args.seed + 1, segments=segments,
     )
 
     for _ in range(args.warmup):
@@ -147,6 +152,7 @@ def main():
                 "pop_per_gpu": pop,
                 "candle_evals_per_step_per_gpu":
                     engine.candle_evals_per_step,
+                "fitness_segments": segments,
                 "best_fitness": best_fit,
                 "secondary": secondary,
             },

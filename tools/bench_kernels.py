@@ -44,6 +44,12 @@ def bench_backtest(reps):
         candles_chl_v(generate_ohlcv(T, nsym, seed=0))).cuda()
     pop = torch.from_numpy(random_population(P, seed=1)).cuda()
     dt = timed(lambda: run_backtest_gpu(candles, pop), reps)
+    for seg in (4, 8, 16):
+        cseg = candles.reshape(nsym * seg, T // seg, 4).contiguous()
+        dts = timed(lambda: run_backtest_gpu(cseg, pop), reps)
+        print(json.dumps({"kernel": f"backtest_seg{seg}", "ms": dts * 1e3,
+                          "candles_per_sec": P * nsym * T / dts}),
+              flush=True)
     return {"kernel": "backtest", "ms": dt * 1e3,
             "candles_per_sec": P * nsym * T / dt,
             "config": {"nsym": nsym, "T": T, "P": P}}
