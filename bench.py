@@ -99,7 +99,7 @@ def main():
     # docstring: standard GA anti-overfit practice; also fills the chip —
     # pop x symbols alone is exactly 1 wave/SIMD). Candle-eval totals are
     # unchanged: every candle of every symbol is backtested each step.
-    segments = 8 if (on_gpu and T % 8 == 0) else 1
+    segments = 16 if (on_gpu and T % 16 == 0) else 1
     engine = GAEngine(
         candles, pop_per_rank=pop, rank=rank, world=world, device=device,
         seed=args.seed + 1, segments=segments,

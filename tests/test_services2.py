@@ -254,3 +254,34 @@ def test_pattern_model_classifies_synthetic():
     assert hits / trials > 0.5
     det = m.detect(generate_pattern("double_top", rng) * 100 + 50)
     assert det["signal"] in ("bearish", "neutral", "bullish")
+
+
+def test_generic_ga_optimizes():
+    from ai_crypto_trader_amd.services.genetic_algorithm import (
+        GeneticAlgorithm,
+    )
+
+    ranges = {"x": (-5.0, 5.0), "n": (1, 10, "int")}
+
+    def fitness(ind):
+        return -(ind["x"] - 2.0) ** 2 - (ind["n"] - 7) ** 2
+
+    ga = GeneticAlgorithm(ranges, fitness, population_size=48,
+                          generations=15, seed=3)
+    best, fit = ga.run()
+    assert abs(best["x"] - 2.0) < 0.5
+    assert best["n"] == 7
+    assert isinstance(best["n"], int)
+    assert ga.diversity() >= 0
+    assert len(ga.history) == 15
+
+    # batched fitness path
+    def fitness_batch(pop):
+        import numpy as np
+        return np.asarray([-(p["x"] - 1.0) ** 2 for p in pop])
+
+    fitness_batch.batch = True
+    ga2 = GeneticAlgorithm({"x": (-5.0, 5.0)}, fitness_batch,
+                           population_size=32, generations=10, seed=1)
+    best2, _ = ga2.run()
+    assert abs(best2["x"] - 1.0) < 0.5
