@@ -123,6 +123,49 @@ class MonteCarloService(Service):
         })
         return stats
 
+    def simulate_scenarios_gpu(self, syms: list[str]) -> dict:
+        """All scenarios concurrently on separate HIP streams: kernels
+        launch back-to-back with no host sync between scenarios, then one
+        join before the (device-side) stats. Overlaps the path generation
+        of the 5 scenarios across the GPU (SURVEY.md build-plan item:
+        'overlap copies and collectives with compute on separate HIP
+        streams')."""
+        import torch
+
+        from ..ops.montecarlo import mc_paths_gpu, risk_stats
+
+        mu0, sigma0, chol = self._estimate_params(syms)
+        w = np.full(len(syms), 1.0 / len(syms))
+        days = self.config.monte_carlo.time_horizon_days
+        n_paths = self.config.monte_carlo.num_simulations
+        streams = {}
+        results = {}
+        t0 = time.perf_counter()
+        for scen, (m_mu, m_sig) in self.config.monte_carlo.scenarios.items():
+            st = torch.cuda.Stream()
+            streams[scen] = st
+            with torch.cuda.stream(st):
+                fv, dd = mc_paths_gpu(
+                    chol, mu0 * m_mu, np.maximum(sigma0 * m_sig, 1e-4), w,
+                    n_steps=days, n_paths=n_paths, dt=1.0 / 365.0,
+                    seed=(self.runs * 977 + 13) & 0xFFFFFFFF)
+                results[scen] = (fv, dd)
+            self.runs += 1
+        for st in streams.values():
+            st.synchronize()
+        el = time.perf_counter() - t0
+        self.metrics.record_kernel_time("mc_paths_scenarios", el)
+        report = {}
+        for scen, (fv, dd) in results.items():
+            stats = risk_stats(fv, v0=1.0)
+            stats["max_drawdown_mean"] = float(dd.mean())
+            stats["max_drawdown_p95"] = float(torch.quantile(dd, 0.95))
+            stats.update({"scenario": scen, "symbols": syms,
+                          "n_paths": int(n_paths), "horizon_days": days,
+                          "timestamp": time.time()})
+            report[scen] = stats
+        return report
+
     async def run_portfolio_mc(self) -> dict | None:
         syms = sorted(s for s, h in self.prices.items() if len(h) >= 64)
         if not syms:
@@ -132,9 +175,12 @@ class MonteCarloService(Service):
             if len(syms) >= k:
                 syms = syms[:k]
                 break
-        report = {}
-        for scen in self.config.monte_carlo.scenarios:
-            report[scen] = self.simulate(syms, scen)
+        if gpu_available() and len(syms) in (4, 8, 16, 32, 64):
+            report = self.simulate_scenarios_gpu(syms)
+        else:
+            report = {}
+            for scen in self.config.monte_carlo.scenarios:
+                report[scen] = self.simulate(syms, scen)
         await self.bus.set(Keys.MONTE_CARLO_RESULTS, report)
         await self.bus.set(Keys.MONTE_CARLO_LATEST_REPORT, {
             "generated_at": time.time(),

@@ -183,6 +183,10 @@ async def main():
         ]
     for s in services:
         await s.start()
+    from ai_crypto_trader_amd.services.supervisor import ServiceSupervisor
+    supervisor = ServiceSupervisor(bus, services, cfg)
+    await supervisor.start()
+    services = services + [supervisor]
 
     dash_server = None
     if args.dashboard:

@@ -203,3 +203,59 @@ def test_pipeline_end_to_end():
     assert state["risk"] is not None and "portfolio_var" in state["risk"]
     assert len(state["prices"]) == 2
     assert state["trades"] >= 1        # eager confidence gate must trade
+
+
+def test_supervisor_restarts_crashed_service():
+    """Fault injection: a crashed service gets restarted (elastic
+    recovery the reference delegated to Docker)."""
+    from ai_crypto_trader_amd.services.supervisor import ServiceSupervisor
+
+    class Flaky(Service := __import__(
+            "ai_crypto_trader_amd.services.base",
+            fromlist=["Service"]).Service):
+        name = "flaky"
+        started = 0
+
+        async def run(self):
+            type(self).started += 1
+            while self.running:
+                await asyncio.sleep(0.05)
+
+    async def go():
+        bus = InProcessBus()
+        svc = Flaky(bus, AppConfig())
+        await svc.start()
+        sup = ServiceSupervisor(bus, [svc], AppConfig(),
+                                check_interval=0.05, backoff_base=0.01)
+        await sup.start()
+        await asyncio.sleep(0.1)          # let tasks begin
+        assert Flaky.started == 1
+        await ServiceSupervisor.inject_failure(svc)
+        assert not svc.healthy
+        await asyncio.sleep(0.5)
+        assert svc.healthy                      # restarted
+        assert Flaky.started >= 2
+        assert sup.restarts["flaky"] >= 1
+        await sup.stop()
+        await svc.stop()
+
+    asyncio.run(go())
+
+
+def test_tracer_spans():
+    from ai_crypto_trader_amd.utils.tracing import Tracer
+
+    tr = Tracer("t")
+    with tr.span("outer", x=1):
+        with tr.span("inner"):
+            pass
+    tr.instant("marker")
+    tr.counter("queue", depth=3)
+    assert len(tr.events) == 4
+    import json as _json
+    import tempfile
+    with tempfile.TemporaryDirectory() as d:
+        p = tr.dump(d + "/trace.json")
+        data = _json.loads(open(p).read())
+        names = [e["name"] for e in data["traceEvents"]]
+        assert "outer" in names and "inner" in names
