@@ -316,7 +316,8 @@ class PPOAgent:
 
     def __init__(self, device, n_obs=N_OBS, n_act=N_ACT, lr=3e-4,
                  gamma=0.99, lam=0.95, clip=0.2, epochs=4,
-                 minibatches=4, ent_coef=0.01, vf_coef=0.5, seed=0):
+                 minibatches=4, ent_coef=0.01, vf_coef=0.5, seed=0,
+                 use_graph: bool = False):
         torch.manual_seed(seed)
         self.device = torch.device(device)
         self.net = ActorCritic(n_obs, n_act).to(device)
@@ -324,6 +325,14 @@ class PPOAgent:
         self.gamma, self.lam, self.clip = gamma, lam, clip
         self.epochs, self.minibatches = epochs, minibatches
         self.ent_coef, self.vf_coef = ent_coef, vf_coef
+        # hipGraph-captured rollout: the rollout is launch-bound (T
+        # alternating tiny policy/env kernels), so one graph replay per
+        # rollout removes the per-launch overhead. Capture is lazy and
+        # per (env, T); falls back to eager on capture failure.
+        self.use_graph = use_graph
+        self._graph = None
+        self._graph_key = None
+        self._gbuf = None
 
     @torch.no_grad()
     def policy(self, obs):
@@ -331,6 +340,60 @@ class PPOAgent:
         dist_ = torch.distributions.Categorical(logits=logits)
         a = dist_.sample()
         return a, dist_.log_prob(a), v
+
+    def _rollout_body(self, env, T, bufs):
+        """The rollout loop against preallocated buffers — runs eagerly
+        AND under hip-graph stream capture (no host reads, no allocs that
+        escape; sampling via gumbel-argmax on capture-safe torch.rand)."""
+        obs_buf, act_buf, logp_buf, rew_buf, done_buf, val_buf = bufs
+        obs = env.obs
+        for t in range(T):
+            logits, v = self.net(obs)
+            logp_all = torch.log_softmax(logits, dim=-1)
+            u = torch.rand_like(logits)
+            g = -torch.log(-torch.log(u + 1e-20) + 1e-20)
+            a = (logp_all + g).argmax(dim=-1)
+            obs_buf[t].copy_(obs)
+            act_buf[t].copy_(a)
+            logp_buf[t].copy_(logp_all.gather(1, a[:, None]).squeeze(1))
+            val_buf[t].copy_(v)
+            o, r, d = env.step(a)
+            rew_buf[t].copy_(r)
+            done_buf[t].copy_(d)
+            obs = env.obs
+        _, v_last = self.net(obs)
+        val_buf[T].copy_(v_last)
+
+    def _make_bufs(self, env, T):
+        E, dev = env.n_envs, self.device
+        return (torch.zeros(T, E, N_OBS, device=dev),
+                torch.zeros(T, E, dtype=torch.long, device=dev),
+                torch.zeros(T, E, device=dev),
+                torch.zeros(T, E, device=dev),
+                torch.zeros(T, E, device=dev),
+                torch.zeros(T + 1, E, device=dev))
+
+    @torch.no_grad()
+    def rollout_graphed(self, env: TradingVecEnv, T: int):
+        key = (id(env), T)
+        if self._graph_key != key:
+            self._gbuf = self._make_bufs(env, T)
+            torch.cuda.synchronize()
+            # warmup on a side stream (required before capture)
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                self._rollout_body(env, T, self._gbuf)
+            torch.cuda.current_stream().wait_stream(s)
+            torch.cuda.synchronize()
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                self._rollout_body(env, T, self._gbuf)
+            self._graph = g
+            self._graph_key = key
+        else:
+            self._graph.replay()
+        return self._gbuf
 
     def rollout(self, env: TradingVecEnv, T: int):
         E = env.n_envs
@@ -410,5 +473,12 @@ class PPOAgent:
             off += g.numel()
 
     def train_step(self, env: TradingVecEnv, horizon: int = 128) -> dict:
-        out = self.rollout(env, horizon)
+        if self.use_graph and self.device.type == "cuda":
+            try:
+                out = self.rollout_graphed(env, horizon)
+            except Exception:
+                self.use_graph = False        # fall back to eager forever
+                out = self.rollout(env, horizon)
+        else:
+            out = self.rollout(env, horizon)
         return self.update(*out)

@@ -102,7 +102,8 @@ def bench_lstm_train(nsym=32, T=1_000_000, seq=60, batch=16_384,
     }
 
 
-def bench_ppo(n_envs=256, horizon=128, steps=6, warmup=2, seed=0):
+def bench_ppo(n_envs=256, horizon=128, steps=6, warmup=2, seed=0,
+              use_graph=False):
     from ai_crypto_trader_amd.data.synthetic import (
         candles_chl_v, generate_ohlcv,
     )
@@ -112,7 +113,7 @@ def bench_ppo(n_envs=256, horizon=128, steps=6, warmup=2, seed=0):
         candles_chl_v(generate_ohlcv(500_000, 16, seed=seed))).cuda()
     env = TradingVecEnv(market, n_envs=n_envs, ep_len=1024, seed=seed)
     env.reset()
-    agent = PPOAgent("cuda", seed=seed)
+    agent = PPOAgent("cuda", seed=seed, use_graph=use_graph)
 
     for _ in range(warmup):
         agent.train_step(env, horizon)
@@ -125,6 +126,7 @@ def bench_ppo(n_envs=256, horizon=128, steps=6, warmup=2, seed=0):
     dt = (time.perf_counter() - t0) / steps
     return {
         "workload": "ppo_train (BASELINE config #4, 1-GPU shard)",
+        "rollout_graph": agent.use_graph,
         "ms_per_step": dt * 1e3,
         "env_steps_per_sec": n_envs * horizon / dt,
         "stats": stats,
@@ -143,6 +145,7 @@ def main():
         print(json.dumps(bench_lstm_train()), flush=True)
     if args.workload in ("all", "ppo"):
         print(json.dumps(bench_ppo()), flush=True)
+        print(json.dumps(bench_ppo(use_graph=True)), flush=True)
 
 
 if __name__ == "__main__":
