@@ -30,6 +30,40 @@ ANNUALIZE = np.float32(np.sqrt(525_600.0))   # 1m candles per year
 EPS = np.float32(1e-9)
 
 
+def shared_series(candles: np.ndarray):
+    """Param-independent per-symbol series shared by every lane (the GPU
+    kernel computes these cooperatively during tile staging — see
+    strategy.py step-1 spec): rolling hmax14/lmin14 and sma20/sma50 with
+    min(t+1, W) windows; f64 window sums -> f64 divide -> f32.
+
+    candles: (nsym, T, 4) -> dict of (nsym, T) f32 arrays."""
+    from numpy.lib.stride_tricks import sliding_window_view
+
+    f32 = np.float32
+    high = candles[:, :, 1]
+    low = candles[:, :, 2]
+    close64 = candles[:, :, 0].astype(np.float64)
+    nsym, T = high.shape
+
+    def roll_extreme(x, W, fn, pad):
+        xp = np.concatenate(
+            [np.full((nsym, W - 1), pad, x.dtype), x], axis=1)
+        return fn(sliding_window_view(xp, W, axis=1), axis=-1)
+
+    hmax = roll_extreme(high, 14, np.max, -np.inf).astype(f32)
+    lmin = roll_extreme(low, 14, np.min, np.inf).astype(f32)
+
+    def roll_mean(W):
+        cp = np.concatenate(
+            [np.zeros((nsym, W - 1), np.float64), close64], axis=1)
+        s = sliding_window_view(cp, W, axis=1).sum(axis=-1)
+        L = np.minimum(np.arange(T) + 1, W).astype(np.float64)
+        return (s / L).astype(f32)
+
+    return {"hmax": hmax, "lmin": lmin,
+            "sma20": roll_mean(20), "sma50": roll_mean(50)}
+
+
 def run_backtest_cpu(
     candles: np.ndarray,       # (nsym, T, 4) f32 [close, high, low, volume]
     population: np.ndarray,    # (P, NPARAM) f32
@@ -65,6 +99,11 @@ def run_backtest_cpu(
     exit_v = par[:, 11].astype(np.int32)
     size_pct, sl_pct, tp_pct = par[:, 12], par[:, 13], par[:, 14]
     trail_pct, trail_act = par[:, 15], par[:, 16]
+    stoch_os, stoch_ob = par[:, 17], par[:, 18]
+    will_os = stoch_os - f32(100.0)
+    will_ob = stoch_ob - f32(100.0)
+
+    shared = shared_series(candles)                    # (nsym, T) each
 
     # Indicator state.
     ema_f = np.empty(L, f32)
@@ -154,17 +193,31 @@ def run_backtest_cpu(
 
         prev_close = close
 
-        # --- 2. votes ------------------------------------------------------
+        # --- 2. votes (6-indicator TradingSignal voting) -------------------
         if t >= WARMUP:
+            hmax = shared["hmax"][sym_idx, t]
+            lmin = shared["lmin"][sym_idx, t]
+            sma20 = shared["sma20"][sym_idx, t]
+            sma50 = shared["sma50"][sym_idx, t]
+            # stoch/will division-free: num/den vs thr <=> num vs thr*den
+            srange = np.maximum(hmax - lmin, EPS)
+            st_num = f32(100.0) * (close - lmin)
+            wl_num = f32(-100.0) * (hmax - close)
             buy = (
                 (rsi_num < rsi_os * rsi_den).astype(np.int32)
                 + (macd_hist > 0).astype(np.int32)
                 + (bb_num < bb_bth * bb_den).astype(np.int32)
+                + (st_num < stoch_os * srange).astype(np.int32)
+                + (wl_num < will_os * srange).astype(np.int32)
+                + ((close > sma20) & (sma20 > sma50)).astype(np.int32)
             )
             sell = (
                 (rsi_num > rsi_ob * rsi_den).astype(np.int32)
                 + (macd_hist < 0).astype(np.int32)
                 + (bb_num > bb_sth * bb_den).astype(np.int32)
+                + (st_num > stoch_ob * srange).astype(np.int32)
+                + (wl_num > will_ob * srange).astype(np.int32)
+                + ((close < sma20) & (sma20 < sma50)).astype(np.int32)
             )
             net = buy - sell
         else:

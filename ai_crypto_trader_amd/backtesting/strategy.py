@@ -31,14 +31,29 @@ Param vector layout (NPARAM float32 slots):
   14 take_profit_pct   take-profit distance         [0.004, 0.4]
   15 trailing_stop_pct trailing distance (0 = off)  [0.0, 0.1]
   16 trailing_act_pct  gain before trailing arms    [0.0, 0.1]
+  17 stoch_os          stochastic oversold          [5, 45]
+  18 stoch_ob          stochastic overbought        [55, 95]
+                       (Williams %R thresholds are tied: stoch_os-100 /
+                        stoch_ob-100, the reference's -80/-20 at 20/80)
 
 Per-candle semantics (both engines implement EXACTLY this order):
   1. update indicators with candle t (EMA fast/slow, MACD + signal EMA,
-     Wilder RSI with zero-init recurrence, Bollinger rolling sum/sumsq ring)
-  2. votes (only when t >= WARMUP):
-       buy  += rsi < rsi_oversold;  sell += rsi > rsi_overbought
-       buy  += macd_hist > 0;       sell += macd_hist < 0
-       buy  += bb_pos < bb_buy_th;  sell += bb_pos > bb_sell_th
+     Wilder RSI with zero-init recurrence, Bollinger rolling sum/sumsq
+     ring). Param-INDEPENDENT series are shared by every lane of a symbol
+     and precomputed once (GPU: cooperatively during tile staging):
+       hmax14/lmin14 = rolling max(high)/min(low) over min(t+1, 14)
+       sma20/sma50   = rolling close means over min(t+1, 20/50)
+                       (f64 window sums, f64 divide, cast f32)
+       stoch = 100*(close-lmin14)/max(hmax14-lmin14, eps)
+       will  = -100*(hmax14-close)/max(hmax14-lmin14, eps)
+  2. votes (only when t >= WARMUP) — the reference's 6-indicator
+     TradingSignal voting (binance_ml_strategy.py:489-543):
+       buy  += rsi < rsi_oversold;     sell += rsi > rsi_overbought
+       buy  += macd_hist > 0;          sell += macd_hist < 0
+       buy  += bb_pos < bb_buy_th;     sell += bb_pos > bb_sell_th
+       buy  += stoch < stoch_os;       sell += stoch > stoch_ob
+       buy  += will < stoch_os - 100;  sell += will > stoch_ob - 100
+       buy  += close > sma20 > sma50;  sell += close < sma20 < sma50
        net = buy - sell
   3. if in position:
        peak = max(peak, high)
@@ -61,8 +76,9 @@ from __future__ import annotations
 
 import numpy as np
 
-NPARAM = 17
+NPARAM = 19
 MAX_WIN = 32           # max Bollinger window — sized to the kernel's LDS ring
+SHARED_HALO = 64       # kernel halo: covers bb<=32 and sma50 windows
 WARMUP = 128           # candles before the first vote (covers 3x max period)
 FEE = 0.001            # taker fee per side (reference: strategy_tester.py 0.1%)
 
@@ -73,6 +89,7 @@ PARAM_NAMES = [
     "entry_votes", "exit_votes",
     "position_size_pct", "stop_loss_pct", "take_profit_pct",
     "trailing_stop_pct", "trailing_act_pct",
+    "stoch_os", "stoch_ob",
 ]
 
 # (low, high, is_int) bounds per slot — used by the GA and random init.
@@ -80,13 +97,15 @@ PARAM_BOUNDS = np.array([
     (2, 64, 1), (5, 50, 0), (50, 95, 0),
     (2, 32, 1), (4, 64, 1), (2, 32, 1),
     (4, 32, 1), (0.5, 4.0, 0), (0.0, 0.5, 0), (0.5, 1.0, 0),
-    (1, 3, 1), (1, 3, 1),
+    (1, 4, 1), (1, 4, 1),
     (0.05, 1.0, 0), (0.002, 0.2, 0), (0.004, 0.4, 0),
     (0.0, 0.1, 0), (0.0, 0.1, 0),
+    (5, 45, 0), (55, 95, 0),
 ], dtype=np.float32)
 
 # Named defaults ≈ the reference's dca/threshold strategy parameter block
-# (config.json trading defaults: RSI 14/30/70, MACD 12/26/9, BB 20/2).
+# (config.json trading defaults: RSI 14/30/70, MACD 12/26/9, BB 20/2,
+# Stoch 20/80).
 DEFAULT_PARAMS = np.array([
     14, 30, 70,
     12, 26, 9,
@@ -94,6 +113,7 @@ DEFAULT_PARAMS = np.array([
     2, 2,
     0.5, 0.02, 0.04,
     0.0, 0.01,
+    20, 80,
 ], dtype=np.float32)
 
 

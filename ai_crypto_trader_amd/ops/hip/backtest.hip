@@ -23,9 +23,9 @@
 #define BT_BLOCK 256
 #define BT_TILE 256
 #define BT_MAXWIN 32           // == strategy.py MAX_WIN
-#define BT_HALO BT_MAXWIN
+#define BT_HALO 64             // == strategy.py SHARED_HALO (covers sma50)
 #define BT_SPAN (BT_TILE + BT_HALO)
-#define BT_NPARAM 17
+#define BT_NPARAM 19
 #define BT_NMETRIC 10
 #define BT_WARMUP 128          // == strategy.py WARMUP
 #define BT_FEE 0.001f
@@ -41,6 +41,7 @@ struct LaneParams {
     float bb_k, bb_bth, bb_sth;
     int entry_v, exit_v;
     float size_pct, sl_pct, tp_pct, trail_pct, trail_act;
+    float stoch_os, stoch_ob, will_os, will_ob;
 };
 
 __global__ void __launch_bounds__(BT_BLOCK) backtest_kernel(
@@ -52,6 +53,12 @@ __global__ void __launch_bounds__(BT_BLOCK) backtest_kernel(
 #pragma clang fp contract(off)           // match the numpy f32 reference
     __shared__ float chist[BT_SPAN];     // close history incl. halo
     __shared__ float hl[BT_SPAN][2];     // high, low
+    // param-independent shared series, computed cooperatively per tile
+    // (strategy.py step-1 spec: every lane of a symbol shares them)
+    __shared__ float sh_hmax[BT_TILE];   // rolling max(high, 14)
+    __shared__ float sh_lmin[BT_TILE];   // rolling min(low, 14)
+    __shared__ float sh_sma20[BT_TILE];
+    __shared__ float sh_sma50[BT_TILE];
 
     // block -> (symbol, param chunk); same-symbol blocks share an XCD when
     // the shape allows (dispatcher places block b on XCD b%8).
@@ -86,6 +93,9 @@ __global__ void __launch_bounds__(BT_BLOCK) backtest_kernel(
         q.entry_v = (int)pr[10]; q.exit_v = (int)pr[11];
         q.size_pct = pr[12]; q.sl_pct = pr[13]; q.tp_pct = pr[14];
         q.trail_pct = pr[15]; q.trail_act = pr[16];
+        q.stoch_os = pr[17]; q.stoch_ob = pr[18];
+        q.will_os = q.stoch_os - 100.0f;
+        q.will_ob = q.stoch_ob - 100.0f;
     }
 
     // ---- state ----------------------------------------------------------
@@ -125,6 +135,28 @@ __global__ void __launch_bounds__(BT_BLOCK) backtest_kernel(
         }
         __syncthreads();
         const int tend = min(BT_TILE, T - t0);
+        // cooperative shared-series precompute: one thread per tile candle
+        if (tid < tend) {
+            const int t = t0 + tid;
+            const int base = tid + BT_HALO;
+            const int L14 = min(t + 1, 14);
+            float hmax = -1e30f, lmin = 1e30f;
+            for (int j = 0; j < L14; ++j) {
+                hmax = fmaxf(hmax, hl[base - j][0]);
+                lmin = fminf(lmin, hl[base - j][1]);
+            }
+            sh_hmax[tid] = hmax;
+            sh_lmin[tid] = lmin;
+            const int L20 = min(t + 1, 20);
+            double s20 = 0.0;
+            for (int j = 0; j < L20; ++j) s20 += (double)chist[base - j];
+            sh_sma20[tid] = (float)(s20 / (double)L20);
+            const int L50 = min(t + 1, 50);
+            double s50 = 0.0;
+            for (int j = 0; j < L50; ++j) s50 += (double)chist[base - j];
+            sh_sma50[tid] = (float)(s50 / (double)L50);
+        }
+        __syncthreads();
         for (int tt = 0; tt < tend; ++tt) {
             const int t = t0 + tt;
             const float close = chist[tt + BT_HALO];
@@ -174,13 +206,28 @@ __global__ void __launch_bounds__(BT_BLOCK) backtest_kernel(
 
             prev_close = close;
 
-            // --- 2. votes --------------------------------------------
+            // --- 2. votes (6-indicator TradingSignal voting) ---------
             int net = 0;
             if (t >= BT_WARMUP) {
+                const float hmax = sh_hmax[tt];
+                const float lmin = sh_lmin[tt];
+                const float sma20 = sh_sma20[tt];
+                const float sma50 = sh_sma50[tt];
+                const float srange = fmaxf(hmax - lmin, BT_EPS);
+                const float st_num = 100.0f * (close - lmin);
+                const float wl_num = -100.0f * (hmax - close);
                 int buy = (rsi_num < q.rsi_os * rsi_den) +
-                          (macd_hist > 0.0f) + (bb_num < q.bb_bth * bb_den);
+                          (macd_hist > 0.0f) +
+                          (bb_num < q.bb_bth * bb_den) +
+                          (st_num < q.stoch_os * srange) +
+                          (wl_num < q.will_os * srange) +
+                          (close > sma20 && sma20 > sma50);
                 int sell = (rsi_num > q.rsi_ob * rsi_den) +
-                           (macd_hist < 0.0f) + (bb_num > q.bb_sth * bb_den);
+                           (macd_hist < 0.0f) +
+                           (bb_num > q.bb_sth * bb_den) +
+                           (st_num > q.stoch_ob * srange) +
+                           (wl_num > q.will_ob * srange) +
+                           (close < sma20 && sma20 < sma50);
                 net = buy - sell;
             }
 

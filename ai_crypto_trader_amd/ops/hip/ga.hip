@@ -11,7 +11,7 @@
 
 #include "common.hpp"
 
-#define GA_NPARAM 17
+#define GA_NPARAM 19
 
 namespace {
 
