@@ -18,7 +18,8 @@ typedef __bf16 bf16;
 typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
 typedef float f32x4 __attribute__((ext_vector_type(4)));
 
-#define GRU_BM 64
+#define GRU_MT 2            // M-tiles per wave (batch tile = 16*MT rows)
+#define GRU_BM (16 * GRU_MT)
 
 namespace {
 
@@ -63,17 +64,17 @@ __global__ void __launch_bounds__((H / 16) * 64) gru_seq_fwd_kernel(
     for (int g = 0; g < 3; ++g) bias_g[g] = bias[g * H + ncol];
 
     // previous h at this lane's fragment positions (for h' = ... + z*h)
-    float hprev[4][4];
+    float hprev[GRU_MT][4];
 #pragma unroll
-    for (int mt = 0; mt < 4; ++mt)
+    for (int mt = 0; mt < GRU_MT; ++mt)
 #pragma unroll
         for (int r = 0; r < 4; ++r) hprev[mt][r] = 0.0f;
 
     for (int t = 0; t < T; ++t) {
         const long base_tb = ((long)t * B + b0);
-        f32x4 acc[4][3];                 // hp = h @ W_hh + b_hh
+        f32x4 acc[GRU_MT][3];            // hp = h @ W_hh + b_hh
 #pragma unroll
-        for (int mt = 0; mt < 4; ++mt)
+        for (int mt = 0; mt < GRU_MT; ++mt)
 #pragma unroll
             for (int g = 0; g < 3; ++g) {
 #pragma unroll
@@ -83,7 +84,7 @@ __global__ void __launch_bounds__((H / 16) * 64) gru_seq_fwd_kernel(
         for (int ks = 0; ks < H / 32; ++ks) {
             const int k0 = ks * 32 + fq * 8;
 #pragma unroll
-            for (int mt = 0; mt < 4; ++mt) {
+            for (int mt = 0; mt < GRU_MT; ++mt) {
                 const int arow = mt * 16 + fr;
                 bf16x8 a = *reinterpret_cast<const bf16x8*>(
                     &lds_h[arow * HP + k0]);
@@ -100,7 +101,7 @@ __global__ void __launch_bounds__((H / 16) * 64) gru_seq_fwd_kernel(
         __syncthreads();
 
 #pragma unroll
-        for (int mt = 0; mt < 4; ++mt) {
+        for (int mt = 0; mt < GRU_MT; ++mt) {
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 const int row = mt * 16 + fq * 4 + r;
@@ -159,16 +160,16 @@ __global__ void __launch_bounds__((H / 16) * 64) gru_seq_bwd_kernel(
 
     const int ncol = 16 * w + fr;
 
-    float dhrec[4][4];
+    float dhrec[GRU_MT][4];
 #pragma unroll
-    for (int mt = 0; mt < 4; ++mt)
+    for (int mt = 0; mt < GRU_MT; ++mt)
 #pragma unroll
         for (int r = 0; r < 4; ++r) dhrec[mt][r] = 0.0f;
 
     for (int t = T - 1; t >= 0; --t) {
         const long base_tb = ((long)t * B + b0);
 #pragma unroll
-        for (int mt = 0; mt < 4; ++mt) {
+        for (int mt = 0; mt < GRU_MT; ++mt) {
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 const int row = mt * 16 + fq * 4 + r;
@@ -204,14 +205,14 @@ __global__ void __launch_bounds__((H / 16) * 64) gru_seq_bwd_kernel(
         __syncthreads();
 
         // dh_prev += dgates_h @ W_hh^T : (64 x 3H) @ (3H x H)
-        f32x4 acc[4];
+        f32x4 acc[GRU_MT];
 #pragma unroll
-        for (int mt = 0; mt < 4; ++mt) acc[mt] = {0.f, 0.f, 0.f, 0.f};
+        for (int mt = 0; mt < GRU_MT; ++mt) acc[mt] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll 3
         for (int ks = 0; ks < THREEH / 32; ++ks) {
             const int k0 = ks * 32 + fq * 8;
 #pragma unroll
-            for (int mt = 0; mt < 4; ++mt) {
+            for (int mt = 0; mt < GRU_MT; ++mt) {
                 const int arow = mt * 16 + fr;
                 bf16x8 a = *reinterpret_cast<const bf16x8*>(
                     &lds_dg[arow * GP + k0]);
@@ -222,7 +223,7 @@ __global__ void __launch_bounds__((H / 16) * 64) gru_seq_bwd_kernel(
             }
         }
 #pragma unroll
-        for (int mt = 0; mt < 4; ++mt)
+        for (int mt = 0; mt < GRU_MT; ++mt)
 #pragma unroll
             for (int r = 0; r < 4; ++r) dhrec[mt][r] += acc[mt][r];
         __syncthreads();

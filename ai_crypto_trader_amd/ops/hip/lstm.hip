@@ -33,7 +33,10 @@ typedef __bf16 bf16;
 typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
 typedef float f32x4 __attribute__((ext_vector_type(4)));
 
-#define LSTM_BM 64          // batch rows per block
+#define LSTM_MT 2           // M-tiles per wave (batch tile = 16*MT rows)
+#define LSTM_BM (16 * LSTM_MT)   // 32-row batch tiles: 2x the blocks of a
+                                 // 64-row tile -> 2 blocks/CU at B=16384
+                                 // (B/64 blocks was exactly 1/CU: latency-bound)
 
 namespace {
 
@@ -81,18 +84,18 @@ __global__ void __launch_bounds__((H / 16) * 64) lstm_seq_fwd_kernel(
 #pragma unroll
     for (int g = 0; g < 4; ++g) bias_g[g] = bias[g * H + ncol];
 
-    float c_reg[4][4];                  // [m_tile][reg] cell state
+    float c_reg[LSTM_MT][4];            // [m_tile][reg] cell state
 #pragma unroll
-    for (int mt = 0; mt < 4; ++mt)
+    for (int mt = 0; mt < LSTM_MT; ++mt)
 #pragma unroll
         for (int r = 0; r < 4; ++r) c_reg[mt][r] = 0.0f;
 
     for (int t = 0; t < T; ++t) {
         // ---- gates = h @ W_hh^T(+) xproj + bias --------------------------
-        f32x4 acc[4][4];                // [m_tile][gate]
+        f32x4 acc[LSTM_MT][4];          // [m_tile][gate]
         const long base_tb = ((long)t * B + b0);
 #pragma unroll
-        for (int mt = 0; mt < 4; ++mt) {
+        for (int mt = 0; mt < LSTM_MT; ++mt) {
 #pragma unroll
             for (int g = 0; g < 4; ++g) {
                 const int col = g * H + ncol;
@@ -109,7 +112,7 @@ __global__ void __launch_bounds__((H / 16) * 64) lstm_seq_fwd_kernel(
         for (int ks = 0; ks < KST; ++ks) {
             const int k0 = ks * 32 + fq * 8;
 #pragma unroll
-            for (int mt = 0; mt < 4; ++mt) {
+            for (int mt = 0; mt < LSTM_MT; ++mt) {
                 const int arow = mt * 16 + fr;
                 bf16x8 a = *reinterpret_cast<const bf16x8*>(
                     &lds_h[arow * HP + k0]);
@@ -127,7 +130,7 @@ __global__ void __launch_bounds__((H / 16) * 64) lstm_seq_fwd_kernel(
 
         // ---- cell update + write h ---------------------------------------
 #pragma unroll
-        for (int mt = 0; mt < 4; ++mt) {
+        for (int mt = 0; mt < LSTM_MT; ++mt) {
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 const int row = mt * 16 + fq * 4 + r;
@@ -183,17 +186,17 @@ __global__ void __launch_bounds__((H / 16) * 64) lstm_seq_bwd_kernel(
 
     const int ncol = 16 * w + fr;
 
-    float dc[4][4];
-    float dhrec[4][4];
+    float dc[LSTM_MT][4];
+    float dhrec[LSTM_MT][4];
 #pragma unroll
-    for (int mt = 0; mt < 4; ++mt)
+    for (int mt = 0; mt < LSTM_MT; ++mt)
 #pragma unroll
         for (int r = 0; r < 4; ++r) { dc[mt][r] = 0.0f; dhrec[mt][r] = 0.0f; }
 
     for (int t = T - 1; t >= 0; --t) {
         const long base_tb = ((long)t * B + b0);
 #pragma unroll
-        for (int mt = 0; mt < 4; ++mt) {
+        for (int mt = 0; mt < LSTM_MT; ++mt) {
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 const int row = mt * 16 + fq * 4 + r;
@@ -232,15 +235,15 @@ __global__ void __launch_bounds__((H / 16) * 64) lstm_seq_bwd_kernel(
         }
         __syncthreads();
 
-        // dh_rec = dgates @ W_hh^T : (64 x 4H) @ (4H x H) -> (64 x H)
-        f32x4 acc[4];
+        // dh_rec = dgates @ W_hh^T : (BM x 4H) @ (4H x H) -> (BM x H)
+        f32x4 acc[LSTM_MT];
 #pragma unroll
-        for (int mt = 0; mt < 4; ++mt) acc[mt] = {0.f, 0.f, 0.f, 0.f};
+        for (int mt = 0; mt < LSTM_MT; ++mt) acc[mt] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll 4
         for (int ks = 0; ks < FOURH / 32; ++ks) {
             const int k0 = ks * 32 + fq * 8;
 #pragma unroll
-            for (int mt = 0; mt < 4; ++mt) {
+            for (int mt = 0; mt < LSTM_MT; ++mt) {
                 const int arow = mt * 16 + fr;
                 bf16x8 a = *reinterpret_cast<const bf16x8*>(
                     &lds_dg[arow * GP + k0]);
@@ -252,7 +255,7 @@ __global__ void __launch_bounds__((H / 16) * 64) lstm_seq_bwd_kernel(
             }
         }
 #pragma unroll
-        for (int mt = 0; mt < 4; ++mt)
+        for (int mt = 0; mt < LSTM_MT; ++mt)
 #pragma unroll
             for (int r = 0; r < 4; ++r) dhrec[mt][r] = acc[mt][r];
         __syncthreads();
