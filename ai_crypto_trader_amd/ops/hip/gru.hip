@@ -18,7 +18,7 @@ typedef __bf16 bf16;
 typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
 typedef float f32x4 __attribute__((ext_vector_type(4)));
 
-#define GRU_MT 2            // M-tiles per wave (batch tile = 16*MT rows)
+#define GRU_MT 4            // M-tiles per wave (batch tile = 16*MT rows)
 #define GRU_BM (16 * GRU_MT)
 
 namespace {

@@ -33,7 +33,7 @@ typedef __bf16 bf16;
 typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
 typedef float f32x4 __attribute__((ext_vector_type(4)));
 
-#define LSTM_MT 2           // M-tiles per wave (batch tile = 16*MT rows)
+#define LSTM_MT 4           // M-tiles per wave (batch tile = 16*MT rows)
 #define LSTM_BM (16 * LSTM_MT)   // 32-row batch tiles: 2x the blocks of a
                                  // 64-row tile -> 2 blocks/CU at B=16384
                                  // (B/64 blocks was exactly 1/CU: latency-bound)
