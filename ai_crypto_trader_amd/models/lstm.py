@@ -45,11 +45,30 @@ class _FusedLSTMSeq(torch.autograd.Function):
         ops.lstm_seq_fwd(
             xproj.contiguous().data_ptr(), wt_bf.data_ptr(), bias.data_ptr(),
             h_out.data_ptr(), gates.data_ptr(), c_sav.data_ptr(), B, T, H,
-            stream,
+            1, stream,
         )
         ctx.save_for_backward(gates, c_sav, h_out, w_bf)
         ctx.dims = (T, B, H)
         return h_out
+
+
+def lstm_seq_infer(xproj: torch.Tensor, w_hh: torch.Tensor,
+                   b_hh: torch.Tensor) -> torch.Tensor:
+    """Inference-only fused forward: skips the backward-save stores
+    (gates/c), ~5x fewer global writes per cell — the serving path."""
+    ops = require_hip_ops()
+    T, B, four_h = xproj.shape
+    H = four_h // 4
+    dev = xproj.device
+    w_bf = w_hh.detach().to(torch.bfloat16).t().contiguous()
+    bias = b_hh.detach().float().contiguous()
+    h_out = torch.empty((T, B, H), dtype=torch.bfloat16, device=dev)
+    stream = torch.cuda.current_stream(dev).cuda_stream
+    ops.lstm_seq_fwd(
+        xproj.contiguous().data_ptr(), w_bf.data_ptr(), bias.data_ptr(),
+        h_out.data_ptr(), 0, 0, B, T, H, 0, stream,
+    )
+    return h_out
 
     @staticmethod
     def backward(ctx, grad_h: torch.Tensor):
@@ -101,9 +120,13 @@ class FusedLSTMLayer(nn.Module):
             if pad:
                 x = torch.cat([x, x.new_zeros(T, pad, x.shape[2])], dim=1)
             xproj = (x.to(self.w_ih.dtype) @ self.w_ih + self.b_ih)
-            h = _FusedLSTMSeq.apply(
-                xproj.to(torch.bfloat16), self.w_hh, self.b_hh
-            )
+            if torch.is_grad_enabled():
+                h = _FusedLSTMSeq.apply(
+                    xproj.to(torch.bfloat16), self.w_hh, self.b_hh
+                )
+            else:   # serving path: no backward saves
+                h = lstm_seq_infer(
+                    xproj.to(torch.bfloat16), self.w_hh, self.b_hh)
             return h[:, :B] if pad else h
         return self._forward_reference(x)
 

@@ -27,6 +27,7 @@ SOURCES = [
     "lstm.hip",
     "gru.hip",
     "attention.hip",
+    "laggedcorr.hip",
     "rl_env.hip",
     "bindings.cpp",
 ]

@@ -29,7 +29,7 @@ extern "C" void launch_cov(const float*, float*, int, int, hipStream_t);
 extern "C" void launch_indicators(const float*, float*, int, int, int,
                                   hipStream_t);
 extern "C" void launch_lstm_seq_fwd(const void*, const void*, const float*,
-                                    void*, void*, float*, int, int, int,
+                                    void*, void*, float*, int, int, int, int,
                                     hipStream_t);
 extern "C" void launch_lstm_seq_bwd(const float*, const void*, const float*,
                                     const void*, void*, int, int, int,
@@ -46,6 +46,8 @@ extern "C" void launch_gru_seq_bwd(const float*, const void*, const float*,
                                    int, hipStream_t);
 extern "C" void launch_attn_fwd(const void*, const void*, const void*,
                                 void*, long, int, int, float, hipStream_t);
+extern "C" void launch_lagged_corr(const float*, const float*, float*, int,
+                                   int, hipStream_t);
 extern "C" void launch_env_reset(const float*, float*, float*, int, int, int,
                                  int, uint64_t, uint64_t, hipStream_t);
 extern "C" void launch_env_step(const float*, float*, const int*, float*,
@@ -167,14 +169,14 @@ PYBIND11_MODULE(_hip_ops, m) {
     m.def("lstm_seq_fwd",
           [](uintptr_t xproj, uintptr_t Wt, uintptr_t bias, uintptr_t h_out,
              uintptr_t gates_out, uintptr_t c_out, int B, int T, int H,
-             uintptr_t stream) {
+             int save_mode, uintptr_t stream) {
               launch_lstm_seq_fwd(reinterpret_cast<const void*>(xproj),
                                   reinterpret_cast<const void*>(Wt),
                                   reinterpret_cast<const float*>(bias),
                                   reinterpret_cast<void*>(h_out),
                                   reinterpret_cast<void*>(gates_out),
                                   reinterpret_cast<float*>(c_out), B, T, H,
-                                  as_stream(stream));
+                                  save_mode, as_stream(stream));
               check(hipGetLastError(), "lstm_seq_fwd launch");
           });
 
@@ -269,6 +271,16 @@ PYBIND11_MODULE(_hip_ops, m) {
                               reinterpret_cast<void*>(O), bh_count, s_len,
                               d_head, scale, as_stream(stream));
               check(hipGetLastError(), "attn_fwd launch");
+          });
+
+    m.def("lagged_corr",
+          [](uintptr_t a, uintptr_t b, uintptr_t out, int n, int max_lag,
+             uintptr_t stream) {
+              launch_lagged_corr(reinterpret_cast<const float*>(a),
+                                 reinterpret_cast<const float*>(b),
+                                 reinterpret_cast<float*>(out), n, max_lag,
+                                 as_stream(stream));
+              check(hipGetLastError(), "lagged_corr launch");
           });
 
     m.def("device_synchronize", []() { check(hipDeviceSynchronize(), "sync"); });

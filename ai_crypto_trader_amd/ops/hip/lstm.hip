@@ -54,7 +54,7 @@ __global__ void __launch_bounds__((H / 16) * 64) lstm_seq_fwd_kernel(
     bf16* __restrict__ h_out,          // (T, B, H)
     bf16* __restrict__ gates_out,      // (T, B, 4H)  post-activation i,f,g,o
     float* __restrict__ c_out,         // (T, B, H)
-    int B, int T)
+    int B, int T, int save_mode)       // 0: inference (h only), 1: training
 {
     constexpr int NW = H / 16;          // waves per block
     constexpr int FOURH = 4 * H;
@@ -143,12 +143,15 @@ __global__ void __launch_bounds__((H / 16) * 64) lstm_seq_fwd_kernel(
                 const float h = go * tanhf_(c);
                 const bf16 hb = (bf16)h;
                 lds_h[row * HP + ncol] = hb;
-                const long gb = (base_tb + row) * FOURH;
-                gates_out[gb + 0 * H + ncol] = (bf16)gi;
-                gates_out[gb + 1 * H + ncol] = (bf16)gf;
-                gates_out[gb + 2 * H + ncol] = (bf16)gg;
-                gates_out[gb + 3 * H + ncol] = (bf16)go;
-                c_out[(base_tb + row) * H + ncol] = c;
+                if (save_mode) {       // backward saves: ~5x the stores of
+                                       // the inference path, skip them there
+                    const long gb = (base_tb + row) * FOURH;
+                    gates_out[gb + 0 * H + ncol] = (bf16)gi;
+                    gates_out[gb + 1 * H + ncol] = (bf16)gf;
+                    gates_out[gb + 2 * H + ncol] = (bf16)gg;
+                    gates_out[gb + 3 * H + ncol] = (bf16)go;
+                    c_out[(base_tb + row) * H + ncol] = c;
+                }
                 h_out[(base_tb + row) * H + ncol] = hb;
             }
         }
@@ -289,18 +292,21 @@ __global__ void mfma_gemm_test_bf16_kernel(const bf16* __restrict__ A,
 extern "C" void launch_lstm_seq_fwd(const void* xproj, const void* Wt,
                                     const float* bias, void* h_out,
                                     void* gates_out, float* c_out, int B,
-                                    int T, int H, hipStream_t stream) {
+                                    int T, int H, int save_mode,
+                                    hipStream_t stream) {
     if (B % LSTM_BM != 0)
         throw std::runtime_error("lstm_fwd: B must be a multiple of 64");
     dim3 grid(B / LSTM_BM);
     if (H == 64) {
         hipLaunchKernelGGL(lstm_seq_fwd_kernel<64>, grid, dim3(256), 0,
                            stream, (const bf16*)xproj, (const bf16*)Wt, bias,
-                           (bf16*)h_out, (bf16*)gates_out, c_out, B, T);
+                           (bf16*)h_out, (bf16*)gates_out, c_out, B, T,
+                           save_mode);
     } else if (H == 32) {
         hipLaunchKernelGGL(lstm_seq_fwd_kernel<32>, grid, dim3(128), 0,
                            stream, (const bf16*)xproj, (const bf16*)Wt, bias,
-                           (bf16*)h_out, (bf16*)gates_out, c_out, B, T);
+                           (bf16*)h_out, (bf16*)gates_out, c_out, B, T,
+                           save_mode);
     } else {
         throw std::runtime_error("lstm_fwd: H must be 32 or 64");
     }

@@ -127,6 +127,7 @@ def main():
     secondary = {}
     if on_gpu and rank == 0 and not args.skip_mc:
         secondary = measure_mc(device)
+    pdist.barrier()      # all ranks leave together (rank 0 runs MC above)
 
     if rank == 0:
         best_fit, _ = engine.best()

@@ -497,3 +497,29 @@ def test_ppo_graph_rollout(dev):
         assert np.isfinite(s["pi_loss"]) and np.isfinite(s["v_loss"])
     # buffers actually advance between replays (env state moves)
     assert float(env.state[:, 1].min()) > 0
+
+
+def test_lagged_corr_gpu(dev):
+    from ai_crypto_trader_amd.ops.social_corr import (
+        lagged_corr_gpu, lagged_pearson_cpu, rank_transform_gpu,
+    )
+
+    rng = np.random.default_rng(51)
+    n, L, lead = 4000, 24, 5
+    b = rng.standard_normal(n).astype(np.float32)
+    a = np.roll(b, lead) * 0.8 + rng.standard_normal(n).astype(
+        np.float32) * 0.3
+    # a[i] ~ b[i-lead]: pairing a[i]~b[i+k] peaks at k = -lead
+    ref = lagged_pearson_cpu(a, b, L)
+    got = lagged_corr_gpu(torch.from_numpy(a).to(dev),
+                          torch.from_numpy(b).to(dev), L)
+    torch.cuda.synchronize()
+    np.testing.assert_allclose(got.cpu().numpy(), ref, rtol=1e-3,
+                               atol=1e-4)
+    assert int(got.argmax()) == L - lead       # peak at the known lag
+    # spearman path via rank transform
+    ra = rank_transform_gpu(torch.from_numpy(a).to(dev))
+    rb = rank_transform_gpu(torch.from_numpy(b).to(dev))
+    sp = lagged_corr_gpu(ra, rb, L)
+    torch.cuda.synchronize()
+    assert int(sp.argmax()) == L - lead
