@@ -52,24 +52,6 @@ class _FusedLSTMSeq(torch.autograd.Function):
         return h_out
 
 
-def lstm_seq_infer(xproj: torch.Tensor, w_hh: torch.Tensor,
-                   b_hh: torch.Tensor) -> torch.Tensor:
-    """Inference-only fused forward: skips the backward-save stores
-    (gates/c), ~5x fewer global writes per cell — the serving path."""
-    ops = require_hip_ops()
-    T, B, four_h = xproj.shape
-    H = four_h // 4
-    dev = xproj.device
-    w_bf = w_hh.detach().to(torch.bfloat16).t().contiguous()
-    bias = b_hh.detach().float().contiguous()
-    h_out = torch.empty((T, B, H), dtype=torch.bfloat16, device=dev)
-    stream = torch.cuda.current_stream(dev).cuda_stream
-    ops.lstm_seq_fwd(
-        xproj.contiguous().data_ptr(), w_bf.data_ptr(), bias.data_ptr(),
-        h_out.data_ptr(), 0, 0, B, T, H, 0, stream,
-    )
-    return h_out
-
     @staticmethod
     def backward(ctx, grad_h: torch.Tensor):
         ops = require_hip_ops()
@@ -93,6 +75,25 @@ def lstm_seq_infer(xproj: torch.Tensor, w_hh: torch.Tensor,
         db_hh = dg_f.sum(dim=0)
         return dgates, dw_hh, db_hh
 
+
+
+def lstm_seq_infer(xproj: torch.Tensor, w_hh: torch.Tensor,
+                   b_hh: torch.Tensor) -> torch.Tensor:
+    """Inference-only fused forward: skips the backward-save stores
+    (gates/c), ~5x fewer global writes per cell — the serving path."""
+    ops = require_hip_ops()
+    T, B, four_h = xproj.shape
+    H = four_h // 4
+    dev = xproj.device
+    w_bf = w_hh.detach().to(torch.bfloat16).t().contiguous()
+    bias = b_hh.detach().float().contiguous()
+    h_out = torch.empty((T, B, H), dtype=torch.bfloat16, device=dev)
+    stream = torch.cuda.current_stream(dev).cuda_stream
+    ops.lstm_seq_fwd(
+        xproj.contiguous().data_ptr(), w_bf.data_ptr(), bias.data_ptr(),
+        h_out.data_ptr(), 0, 0, B, T, H, 0, stream,
+    )
+    return h_out
 
 class FusedLSTMLayer(nn.Module):
     """One LSTM layer: hipBLASLt input projection + fused HIP recurrence.
