@@ -55,10 +55,11 @@ __global__ void __launch_bounds__(BT_BLOCK) backtest_kernel(
     __shared__ float hl[BT_SPAN][2];     // high, low
     // param-independent shared series, computed cooperatively per tile
     // (strategy.py step-1 spec: every lane of a symbol shares them)
-    __shared__ float sh_hmax[BT_TILE];   // rolling max(high, 14)
-    __shared__ float sh_lmin[BT_TILE];   // rolling min(low, 14)
-    __shared__ float sh_sma20[BT_TILE];
-    __shared__ float sh_sma50[BT_TILE];
+    // packed per-candle shared vote inputs: {st_num, wl_num, srange,
+    // trend} — st_num = 100*(close-lmin14), wl_num = -100*(hmax14-close),
+    // srange = max(hmax-lmin, eps), trend = +-1/0 from close vs sma20/50.
+    // All lane-independent, so one b128 broadcast read per candle.
+    __shared__ float4 sh_vote[BT_TILE];
 
     // block -> (symbol, param chunk); same-symbol blocks share an XCD when
     // the shape allows (dispatcher places block b on XCD b%8).
@@ -139,22 +140,27 @@ __global__ void __launch_bounds__(BT_BLOCK) backtest_kernel(
         if (tid < tend) {
             const int t = t0 + tid;
             const int base = tid + BT_HALO;
+            const float cl = chist[base];
             const int L14 = min(t + 1, 14);
             float hmax = -1e30f, lmin = 1e30f;
             for (int j = 0; j < L14; ++j) {
                 hmax = fmaxf(hmax, hl[base - j][0]);
                 lmin = fminf(lmin, hl[base - j][1]);
             }
-            sh_hmax[tid] = hmax;
-            sh_lmin[tid] = lmin;
             const int L20 = min(t + 1, 20);
             double s20 = 0.0;
             for (int j = 0; j < L20; ++j) s20 += (double)chist[base - j];
-            sh_sma20[tid] = (float)(s20 / (double)L20);
+            const float sma20 = (float)(s20 / (double)L20);
             const int L50 = min(t + 1, 50);
             double s50 = 0.0;
             for (int j = 0; j < L50; ++j) s50 += (double)chist[base - j];
-            sh_sma50[tid] = (float)(s50 / (double)L50);
+            const float sma50 = (float)(s50 / (double)L50);
+            const float trend = (cl > sma20 && sma20 > sma50) ? 1.0f
+                                : ((cl < sma20 && sma20 < sma50) ? -1.0f
+                                                                 : 0.0f);
+            sh_vote[tid] = make_float4(
+                100.0f * (cl - lmin), -100.0f * (hmax - cl),
+                fmaxf(hmax - lmin, BT_EPS), trend);
         }
         __syncthreads();
         for (int tt = 0; tt < tend; ++tt) {
@@ -209,25 +215,20 @@ __global__ void __launch_bounds__(BT_BLOCK) backtest_kernel(
             // --- 2. votes (6-indicator TradingSignal voting) ---------
             int net = 0;
             if (t >= BT_WARMUP) {
-                const float hmax = sh_hmax[tt];
-                const float lmin = sh_lmin[tt];
-                const float sma20 = sh_sma20[tt];
-                const float sma50 = sh_sma50[tt];
-                const float srange = fmaxf(hmax - lmin, BT_EPS);
-                const float st_num = 100.0f * (close - lmin);
-                const float wl_num = -100.0f * (hmax - close);
+                const float4 sv = sh_vote[tt];   // st_num, wl_num, srange,
+                                                 // trend (b128 broadcast)
                 int buy = (rsi_num < q.rsi_os * rsi_den) +
                           (macd_hist > 0.0f) +
                           (bb_num < q.bb_bth * bb_den) +
-                          (st_num < q.stoch_os * srange) +
-                          (wl_num < q.will_os * srange) +
-                          (close > sma20 && sma20 > sma50);
+                          (sv.x < q.stoch_os * sv.z) +
+                          (sv.y < q.will_os * sv.z) +
+                          (sv.w > 0.0f);
                 int sell = (rsi_num > q.rsi_ob * rsi_den) +
                            (macd_hist < 0.0f) +
                            (bb_num > q.bb_sth * bb_den) +
-                           (st_num > q.stoch_ob * srange) +
-                           (wl_num > q.will_ob * srange) +
-                           (close < sma20 && sma20 < sma50);
+                           (sv.x > q.stoch_ob * sv.z) +
+                           (sv.y > q.will_ob * sv.z) +
+                           (sv.w < 0.0f);
                 net = buy - sell;
             }
 
