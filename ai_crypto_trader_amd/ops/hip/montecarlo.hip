@@ -32,7 +32,7 @@ __global__ void __launch_bounds__(256) mc_paths_kernel(
     const float* __restrict__ wS0,      // (A,)   weight_a * S0_a
     float* __restrict__ final_value,    // (n_paths,)
     float* __restrict__ max_dd,         // (n_paths,)
-    int n_steps, long n_paths, float v0, uint64_t seed)
+    int n_steps, long n_paths, float v0, uint64_t seed, long path_base)
 {
     __shared__ float lds_cvol[A * A];
     __shared__ float lds_drift[A];
@@ -56,7 +56,7 @@ __global__ void __launch_bounds__(256) mc_paths_kernel(
         for (int step = 0; step < n_steps; ++step) {
             for (int k4 = 0; k4 < A / 4; ++k4) {
                 float4 z4 = philox_normal4(
-                    seed, (uint64_t)path,
+                    seed, (uint64_t)(path + path_base),
                     ((uint64_t)step << 32) | (uint64_t)k4);
 #pragma unroll
                 for (int dz = 0; dz < 4; ++dz) {
@@ -121,7 +121,7 @@ __global__ void __launch_bounds__(512) mc_paths_mfma_kernel(
     const float* __restrict__ wS0,      // (A,)
     float* __restrict__ final_value,    // (n_paths,)
     float* __restrict__ max_dd,         // (n_paths,)
-    int n_steps, long n_paths, float v0, uint64_t seed)
+    int n_steps, long n_paths, float v0, uint64_t seed, long path_base)
 {
     // 8 waves: wave w owns path-columns [32w, 32w+32) -> 8 accumulator
     // fragments per lane (32 f32) instead of 16, so 2+ waves/SIMD fit;
@@ -150,7 +150,7 @@ __global__ void __launch_bounds__(512) mc_paths_mfma_kernel(
     if (path0 >= n_paths) return;
     const int zrow = tid & (MCM_PB - 1);      // this thread's Z^T row
     const int zhalf = tid >> 8;               // which 32-normal half
-    const long zpath = path0 + zrow;
+    const long zpath = path0 + zrow + path_base;
 
     mc_f32x4 logS[4][2];                      // [m_tile][col_tile]
 #pragma unroll
@@ -238,7 +238,8 @@ extern "C" void launch_mc_paths_mfma(const float* cvol, const float* drift,
                                      const float* wS0, float* final_value,
                                      float* max_dd, int n_assets,
                                      int n_steps, long n_paths, float v0,
-                                     uint64_t seed, hipStream_t stream) {
+                                     uint64_t seed, long path_base,
+                                     hipStream_t stream) {
     if (n_assets != MCM_A)
         throw std::runtime_error("mc_paths_mfma: n_assets must be 64");
     if (n_paths % MCM_PB != 0)
@@ -247,14 +248,15 @@ extern "C" void launch_mc_paths_mfma(const float* cvol, const float* drift,
     long blocks = n_paths / MCM_PB;
     hipLaunchKernelGGL(mc_paths_mfma_kernel, dim3((unsigned)blocks),
                        dim3(512), 0, stream, cvol, drift, wS0, final_value,
-                       max_dd, n_steps, n_paths, v0, seed);
+                       max_dd, n_steps, n_paths, v0, seed, path_base);
 }
 
 extern "C" void launch_mc_paths(const float* cvol, const float* drift,
                                 const float* wS0, const float* weights_unused,
                                 float* final_value, float* max_dd,
                                 int n_assets, int n_steps, long n_paths,
-                                float v0, uint64_t seed, hipStream_t stream) {
+                                float v0, uint64_t seed, long path_base,
+                                hipStream_t stream) {
     (void)weights_unused;
     long want = (n_paths + 255) / 256;
     int blocks = (int)(want < 8192 ? want : 8192);
@@ -263,7 +265,7 @@ extern "C" void launch_mc_paths(const float* cvol, const float* drift,
     case AA:                                                                  \
         hipLaunchKernelGGL(mc_paths_kernel<AA>, grid, block, 0, stream, cvol, \
                            drift, wS0, final_value, max_dd, n_steps, n_paths, \
-                           v0, seed);                                         \
+                           v0, seed, path_base);                              \
         break;
     switch (n_assets) {
         MC_CASE(4)

@@ -123,6 +123,7 @@ def mc_paths_cpu(
 def mc_paths_gpu(
     chol, mu, sigma, weights, *, n_steps: int, n_paths: int, dt: float,
     s0: float = 1.0, seed: int = 0, device="cuda", use_mfma: bool | None = None,
+    path_base: int = 0,
 ):
     """GPU path generation; returns (final_value, max_dd) torch tensors.
 
@@ -156,7 +157,7 @@ def mc_paths_gpu(
         ops.mc_paths_mfma(
             t_cvol.data_ptr(), t_drift.data_ptr(), t_w.data_ptr(),
             fv.data_ptr(), dd.data_ptr(), A, n_steps, n_paths + pad, v0,
-            seed, stream,
+            seed, path_base, stream,
         )
         if pad:
             fv, dd = fv[:n_paths], dd[:n_paths]
@@ -164,7 +165,7 @@ def mc_paths_gpu(
         ops.mc_paths(
             t_cvol.data_ptr(), t_drift.data_ptr(), t_w.data_ptr(), 0,
             fv.data_ptr(), dd.data_ptr(), A, n_steps, n_paths, v0, seed,
-            stream,
+            path_base, stream,
         )
     return fv, dd
 
@@ -193,3 +194,80 @@ def risk_stats(final_values, v0: float, confidences=(0.95, 0.99)):
     out["mean"] = float(fv.mean())
     out["prob_profit"] = float((fv > v0).sum() / n)
     return out
+
+
+def mc_paths_sharded(
+    chol, mu, sigma, weights, *, n_paths_total: int, n_steps: int,
+    dt: float, rank: int = 0, world: int = 1, seed: int = 0, s0: float = 1.0,
+    device="cuda", n_bins: int = 2048,
+):
+    """Monte-Carlo sharded across ranks (path batches DP
+    across GPUs, stats via RCCL all-reduce).
+
+    Each rank generates its contiguous PATH-INDEX slice — the Philox
+    counter is the global path id, so the union over ranks is bit-identical
+    to a single-GPU run of n_paths_total. Global statistics come from
+    collectives only (no gather of the 10M-path vectors):
+      - mean/std: all-reduced sums
+      - VaR/CVaR/percentiles: an all-reduced fixed-range histogram
+        (n_bins bins => quantile resolution ~ range/n_bins)
+    Returns (local_fv, local_dd, global_stats dict).
+    """
+    import torch
+
+    from ..parallel import dist as pdist
+
+    per = (n_paths_total + world - 1) // world
+    lo = rank * per
+    hi = min(lo + per, n_paths_total)
+    n_local = hi - lo
+    # the Philox counter is the GLOBAL path id: each rank passes its slice
+    # start as path_base, so the union over ranks is bit-identical to a
+    # single-GPU run and no rank generates another rank's paths.
+    if device == "cpu":
+        fv_np, dd_np = mc_paths_cpu(
+            chol, mu, sigma, weights, n_steps=n_steps,
+            n_paths=n_paths_total, dt=dt, s0=s0, seed=seed)
+        fv = torch.from_numpy(fv_np[lo:hi])
+        dd = torch.from_numpy(dd_np[lo:hi])
+    else:
+        fv, dd = mc_paths_gpu(
+            chol, mu, sigma, weights, n_steps=n_steps, n_paths=n_local,
+            dt=dt, s0=s0, seed=seed, device=device, path_base=lo)
+
+    v0 = float(np.sum(np.asarray(weights) * s0))
+    # all-reduced moments
+    sums = torch.stack([
+        fv.sum(), (fv.double() ** 2).sum().float(), dd.sum(),
+        torch.tensor(float(n_local), device=fv.device),
+    ])
+    pdist.all_reduce_sum_(sums)
+    n = float(sums[3])
+    mean = float(sums[0]) / n
+    var = max(float(sums[1]) / n - mean * mean, 0.0)
+    # all-reduced histogram for quantiles
+    fmin = torch.tensor(float(fv.min()), device=fv.device)
+    fmax = torch.tensor(float(fv.max()), device=fv.device)
+    if pdist.is_dist():
+        import torch.distributed as tdist
+        tdist.all_reduce(fmin, op=tdist.ReduceOp.MIN)
+        tdist.all_reduce(fmax, op=tdist.ReduceOp.MAX)
+    lo_v, hi_v = float(fmin), float(fmax) + 1e-9
+    hist = torch.histc(fv, bins=n_bins, min=lo_v, max=hi_v)
+    pdist.all_reduce_sum_(hist)
+    cum = torch.cumsum(hist, 0)
+    edges = torch.linspace(lo_v, hi_v, n_bins + 1, device=fv.device)
+
+    def quantile(q):
+        idx = int(torch.searchsorted(cum, torch.tensor(
+            q * n, device=fv.device)))
+        return float(edges[min(idx + 1, n_bins)])
+
+    stats = {
+        "mean": mean, "std": var ** 0.5,
+        "var_95": v0 - quantile(0.05), "var_99": v0 - quantile(0.01),
+        "p5": quantile(0.05), "p50": quantile(0.5), "p95": quantile(0.95),
+        "max_drawdown_mean": float(sums[2]) / n,
+        "n_paths": int(n), "world": world,
+    }
+    return fv, dd, stats

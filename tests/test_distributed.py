@@ -89,3 +89,57 @@ def test_world2_matches_single_process():
     np.testing.assert_allclose(
         np.concatenate([fa, fb]), f_single, rtol=1e-6
     )
+
+
+def _mc_worker(rank, world, port, out_q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from ai_crypto_trader_amd.ops.montecarlo import mc_paths_sharded
+
+        A = 4
+        corr = np.full((A, A), 0.3) + 0.7 * np.eye(A)
+        chol = np.linalg.cholesky(corr)
+        _, _, stats = mc_paths_sharded(
+            chol, np.full(A, 0.1), np.full(A, 0.4), np.full(A, 0.25),
+            n_paths_total=8000, n_steps=8, dt=1 / 252, rank=rank,
+            world=world, seed=3, device="cpu")
+        out_q.put((rank, stats))
+    finally:
+        dist.destroy_process_group()
+
+
+def test_mc_sharded_matches_single():
+    """2-rank sharded MC stats == single-process stats (same Philox ids)."""
+    from ai_crypto_trader_amd.ops.montecarlo import mc_paths_sharded
+
+    A = 4
+    corr = np.full((A, A), 0.3) + 0.7 * np.eye(A)
+    chol = np.linalg.cholesky(corr)
+    _, _, ref = mc_paths_sharded(
+        chol, np.full(A, 0.1), np.full(A, 0.4), np.full(A, 0.25),
+        n_paths_total=8000, n_steps=8, dt=1 / 252, rank=0, world=1,
+        seed=3, device="cpu")
+
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_mc_worker, args=(r, 2, 29881, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        r, stats = q.get()
+        results[r] = stats
+    for p in procs:
+        p.join(timeout=180)
+        assert p.exitcode == 0
+
+    for r in (0, 1):
+        s = results[r]
+        assert s["n_paths"] == 8000
+        assert abs(s["mean"] - ref["mean"]) < 1e-5
+        assert abs(s["std"] - ref["std"]) < 1e-5
+        assert abs(s["var_95"] - ref["var_95"]) < 2e-3   # histogram binning
+    assert results[0] == results[1]
