@@ -63,6 +63,8 @@ def parse_args():
 
 
 async def print_status(bus: InProcessBus, services: list, interval: float):
+    """Status report (reference print_status run_trader.py:39-1247 reads
+    ~25 Redis keys; this reads the same keys off the bus)."""
     while True:
         await asyncio.sleep(interval)
         holdings = await bus.get_json(Keys.HOLDINGS) or {}
@@ -70,6 +72,14 @@ async def print_status(bus: InProcessBus, services: list, interval: float):
         regime = await bus.get_json(Keys.CURRENT_MARKET_REGIME) or {}
         prices = await bus.hgetall(Keys.CURRENT_PRICES)
         active = await bus.get_json(Keys.ACTIVE_TRADES) or {}
+        mc = await bus.get_json(Keys.MONTE_CARLO_LATEST_REPORT) or {}
+        patt = await bus.get_json(Keys.PATTERN_ANALYSIS_REPORT) or {}
+        news = await bus.get_json(Keys.NEWS_SUMMARY_REPORT) or {}
+        grid = await bus.get_json(Keys.GRID_PERFORMANCE) or {}
+        dca = await bus.get_json(Keys.DCA_PERFORMANCE) or {}
+        params = await bus.get_json(Keys.STRATEGY_PARAMS) or {}
+        social = await bus.hgetall(Keys.SOCIAL_METRICS)
+        diver = await bus.get_json(Keys.PORTFOLIO_DIVERSIFICATION) or {}
         print("=" * 72)
         print(f"status @ {time.strftime('%H:%M:%S')}  | "
               f"portfolio ${holdings.get('total_value', 0):,.2f}  | "
@@ -83,6 +93,32 @@ async def print_status(bus: InProcessBus, services: list, interval: float):
             ps = "  ".join(f"{s}={float(p):.4f}"
                            for s, p in sorted(prices.items())[:6])
             print(f"prices: {ps}")
+        extras = []
+        if mc.get("base_var_95") is not None:
+            extras.append(f"MC var95={mc['base_var_95']:.4f}")
+        if news.get("market_sentiment") is not None:
+            extras.append(f"news={news['market_sentiment']:.2f}")
+        if social:
+            import json as _j
+            vals = [_j.loads(v).get("sentiment", 0.5)
+                    for v in social.values()]
+            extras.append(f"social={sum(vals) / len(vals):.2f}")
+        if patt.get("patterns"):
+            found = [f"{s}:{d['pattern']}"
+                     for s, d in patt["patterns"].items()
+                     if d["pattern"] != "none"][:3]
+            if found:
+                extras.append("patterns=" + ",".join(found))
+        if grid.get("fills"):
+            extras.append(f"grid fills={grid['fills']}")
+        if dca.get("n_purchases"):
+            extras.append(f"dca buys={dca['n_purchases']}")
+        if params:
+            extras.append("params=evolved")
+        if diver.get("avg_correlation") is not None:
+            extras.append(f"divers={1 - diver['avg_correlation']:.2f}")
+        if extras:
+            print(" | ".join(extras))
         for s in services:
             h = s.health()
             extra = ""
