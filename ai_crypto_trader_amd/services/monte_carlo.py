@@ -54,8 +54,12 @@ class MonteCarloService(Service):
                        for s in syms], axis=1)
         rets = np.diff(np.log(px), axis=0)
         per_year = 525_600.0
-        mu = rets.mean(axis=0) * per_year
-        sigma = np.maximum(rets.std(axis=0) * np.sqrt(per_year), 1e-4)
+        # clamp the annualized drift: a 1m-candle sample mean annualizes
+        # x525600, so short histories produce absurd mu estimates (the
+        # reference has the same naive annualization bug at x252 from 30d;
+        # monte_carlo_service.py:238-247 — we implement the intent)
+        mu = np.clip(rets.mean(axis=0) * per_year, -2.0, 2.0)
+        sigma = np.clip(rets.std(axis=0) * np.sqrt(per_year), 1e-4, 4.0)
         if len(syms) > 1:
             corr = np.corrcoef(rets.T)
             corr = np.nan_to_num(corr, nan=0.0)
