@@ -24,15 +24,20 @@ def test_monte_carlo_service_gpu_streams():
         svc = MonteCarloService(bus, cfg)
         rng = np.random.default_rng(0)
         for s in [f"S{i}USDC" for i in range(4)]:
-            svc.prices[s] = list(
-                1.0 + 0.001 * np.cumsum(rng.standard_normal(256)))
+            # clear POSITIVE drift: the bear scenario multiplies the
+            # estimated mu by -1 (reference config.json:97-103), so
+            # bear-vs-bull ordering is only defined for mu_est > 0
+            steps = 0.0005 + 0.001 * rng.standard_normal(256)
+            svc.prices[s] = list(np.exp(np.cumsum(steps)))
         report = await svc.run_portfolio_mc()
         assert report is not None
         assert set(report) == set(cfg.monte_carlo.scenarios)
         base = report["base"]
         assert base["var_95"] > 0 and base["n_paths"] == 200_000
-        # bear scenario must be riskier than bull at same vol multiplier
         assert report["bear"]["mean"] < report["bull"]["mean"]
+        # 2x vol multiplier widens the outcome distribution
+        spread = lambda r: r["p95"] - r["p5"]
+        assert spread(report["volatile"]) > spread(report["crab"])
 
     asyncio.run(go())
 
