@@ -28,6 +28,7 @@ SOURCES = [
     "gru.hip",
     "attention.hip",
     "laggedcorr.hip",
+    "histogram.hip",
     "rl_env.hip",
     "bindings.cpp",
 ]

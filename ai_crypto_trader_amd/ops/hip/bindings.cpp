@@ -49,6 +49,8 @@ extern "C" void launch_attn_fwd(const void*, const void*, const void*,
                                 void*, long, int, int, float, hipStream_t);
 extern "C" void launch_lagged_corr(const float*, const float*, float*, int,
                                    int, hipStream_t);
+extern "C" void launch_vp_hist(const float*, const float*, const float*,
+                               float*, float*, int, int, int, hipStream_t);
 extern "C" void launch_env_reset(const float*, float*, float*, int, int, int,
                                  int, uint64_t, uint64_t, hipStream_t);
 extern "C" void launch_env_step(const float*, float*, const int*, float*,
@@ -283,6 +285,19 @@ PYBIND11_MODULE(_hip_ops, m) {
                                  reinterpret_cast<float*>(out), n, max_lag,
                                  as_stream(stream));
               check(hipGetLastError(), "lagged_corr launch");
+          });
+
+    m.def("vp_hist",
+          [](uintptr_t candles, uintptr_t lo, uintptr_t hi, uintptr_t hist,
+             uintptr_t updown, int nsym, int T, int n_bins,
+             uintptr_t stream) {
+              launch_vp_hist(reinterpret_cast<const float*>(candles),
+                             reinterpret_cast<const float*>(lo),
+                             reinterpret_cast<const float*>(hi),
+                             reinterpret_cast<float*>(hist),
+                             reinterpret_cast<float*>(updown), nsym, T,
+                             n_bins, as_stream(stream));
+              check(hipGetLastError(), "vp_hist launch");
           });
 
     m.def("device_synchronize", []() { check(hipDeviceSynchronize(), "sync"); });

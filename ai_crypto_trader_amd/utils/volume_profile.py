@@ -72,3 +72,28 @@ class VolumeProfileAnalyzer:
             "bins": [float(x) for x in hist],
             "bin_edges": [float(x) for x in edges],
         }
+
+
+def vp_hist_gpu(candles, n_bins: int = 24):
+    """Batched GPU volume histograms: (nsym, T, 4) f32 cuda ->
+    (hist (nsym, n_bins), updown (nsym, 2), lo, hi). Kernel:
+    ops/hip/histogram.hip (LDS-atomic privatized histograms)."""
+    import torch
+
+    from ..ops import require_hip_ops
+
+    ops = require_hip_ops()
+    assert candles.is_cuda and candles.dtype == torch.float32
+    candles = candles.contiguous()
+    nsym, T, _ = candles.shape
+    lo = candles[:, :, 2].amin(dim=1).contiguous()
+    hi = (candles[:, :, 1].amax(dim=1) + 1e-9).contiguous()
+    hist = torch.empty(nsym, n_bins, dtype=torch.float32,
+                       device=candles.device)
+    updown = torch.empty(nsym, 2, dtype=torch.float32,
+                         device=candles.device)
+    stream = torch.cuda.current_stream(candles.device).cuda_stream
+    ops.vp_hist(candles.data_ptr(), lo.data_ptr(), hi.data_ptr(),
+                hist.data_ptr(), updown.data_ptr(), nsym, T, n_bins,
+                stream)
+    return hist, updown, lo, hi

@@ -523,3 +523,29 @@ def test_lagged_corr_gpu(dev):
     sp = lagged_corr_gpu(ra, rb, L)
     torch.cuda.synchronize()
     assert int(sp.argmax()) == L - lead
+
+
+def test_vp_hist_gpu(dev):
+    """Volume-profile histogram kernel vs numpy histogram."""
+    from ai_crypto_trader_amd.data.synthetic import (
+        candles_chl_v, generate_ohlcv,
+    )
+    from ai_crypto_trader_amd.utils.volume_profile import vp_hist_gpu
+
+    market = candles_chl_v(generate_ohlcv(50_000, 4, seed=61))
+    c = torch.from_numpy(market).to(dev)
+    hist, updown, lo, hi = vp_hist_gpu(c, n_bins=32)
+    torch.cuda.synchronize()
+    for s in range(4):
+        ref, _ = np.histogram(
+            market[s, :, 0], bins=32,
+            range=(float(lo[s]), float(hi[s])),
+            weights=market[s, :, 3])
+        np.testing.assert_allclose(hist[s].cpu().numpy(), ref, rtol=1e-4,
+                                   atol=1e-2)
+        close = market[s, :, 0]
+        vol = market[s, :, 3]
+        up = vol[1:][close[1:] >= close[:-1]].sum()
+        dn = vol[1:][close[1:] < close[:-1]].sum()
+        np.testing.assert_allclose(updown[s].cpu().numpy(), [up, dn],
+                                   rtol=1e-4)
