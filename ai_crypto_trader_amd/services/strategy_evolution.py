@@ -180,9 +180,18 @@ class StrategyEvolutionService(Service):
 
     async def run(self):
         while self.running:
-            perf_raw = await self.bus.get_json(
-                Keys.strategy_performance(self.current_strategy_id))
-            perf = perf_raw or {}
+            # monitor the live strategy: evaluate the current params on the
+            # evaluation market and publish strategy_performance_{id}
+            # (reference monitor_strategy :513 reads the Redis perf key the
+            # executor fills; here the evolution service measures directly)
+            try:
+                perf = self.evaluate_params(self.current_params)
+                await self.bus.set(
+                    Keys.strategy_performance(self.current_strategy_id),
+                    perf)
+            except Exception as e:
+                self.log.warning("perf evaluation failed: %r", e)
+                perf = {}
             if not perf or self.needs_improvement(perf):
                 try:
                     out = await self.evolve_once()
@@ -196,10 +205,23 @@ class StrategyEvolutionService(Service):
     def evaluate_params(self, params: np.ndarray) -> dict:
         """Metrics of a param set on the evaluation market (used for the
         strategy_performance key; strategy_evaluation.py:32-228 formulas
-        via the engines)."""
-        from ..backtesting.engine_cpu import run_backtest_cpu
+        via the engines). GPU kernel when available."""
+        from ..backtesting.strategy import clip_params
 
-        m = run_backtest_cpu(self.candles, params[None]).mean(axis=(0, 1))
+        vec = clip_params(np.asarray(params, np.float32)[None])
+        if str(self.device).startswith("cuda"):
+            import torch
+
+            from ..ops.backtest import run_backtest_gpu
+            mt = run_backtest_gpu(
+                torch.from_numpy(self.candles).to(self.device),
+                torch.from_numpy(vec).to(self.device))
+            torch.cuda.synchronize()
+            m = mt.float().mean(dim=(0, 1)).cpu().numpy()
+        else:
+            from ..backtesting.engine_cpu import run_backtest_cpu
+
+            m = run_backtest_cpu(self.candles, vec).mean(axis=(0, 1))
         d = dict(zip(METRIC_NAMES, (float(x) for x in m)))
         gp, gl = d["gross_profit"], d["gross_loss"]
         d["profit_factor"] = gp / gl if gl > 0 else float("inf")
