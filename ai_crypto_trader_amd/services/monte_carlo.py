@@ -142,9 +142,11 @@ class MonteCarloService(Service):
         w = np.full(len(syms), 1.0 / len(syms))
         days = self.config.monte_carlo.time_horizon_days
         n_paths = self.config.monte_carlo.num_simulations
+        from ..utils.metrics import GpuTimer
+
         streams = {}
         results = {}
-        t0 = time.perf_counter()
+        timer = GpuTimer(self.metrics, "mc_paths_scenarios").__enter__()
         for scen, (m_mu, m_sig) in self.config.monte_carlo.scenarios.items():
             st = torch.cuda.Stream()
             streams[scen] = st
@@ -157,8 +159,7 @@ class MonteCarloService(Service):
             self.runs += 1
         for st in streams.values():
             st.synchronize()
-        el = time.perf_counter() - t0
-        self.metrics.record_kernel_time("mc_paths_scenarios", el)
+        timer.__exit__()
         report = {}
         for scen, (fv, dd) in results.items():
             stats = risk_stats(fv, v0=1.0)

@@ -71,10 +71,11 @@ class PortfolioRiskService(Service):
         if gpu_available() and rets.shape[1] % 16 == 0 and \
                 rets.shape[1] <= 64:
             import torch
+
             from ..ops.covar import cov_gpu
-            t0 = time.perf_counter()
-            cov = cov_gpu(torch.from_numpy(rets).cuda()).cpu().numpy()
-            self.metrics.record_kernel_time("cov", time.perf_counter() - t0)
+            from ..utils.metrics import GpuTimer
+            with GpuTimer(self.metrics, "cov"):
+                cov = cov_gpu(torch.from_numpy(rets).cuda()).cpu().numpy()
             return cov
         return cov_cpu(rets)
 

@@ -97,3 +97,50 @@ def get_metrics(service: str = "app") -> Metrics:
     if service not in _metrics:
         _metrics[service] = Metrics(service)
     return _metrics[service]
+
+
+class GpuTimer:
+    """hipEvent-based kernel timing surfaced as Prometheus histograms
+    ('per-kernel timing via hipEvents surfaced as Prometheus
+    metrics'). torch.cuda.Event IS a hipEvent on ROCm.
+
+        with GpuTimer(metrics, "mc_paths"):
+            ...launch kernels...
+    Falls back to wall time on CPU."""
+
+    def __init__(self, metrics: Metrics, kernel: str):
+        self.metrics = metrics
+        self.kernel = kernel
+        self._ev = None
+        self._t0 = None
+
+    def __enter__(self):
+        try:
+            import torch
+
+            if torch.cuda.is_available():
+                self._ev = (torch.cuda.Event(enable_timing=True),
+                            torch.cuda.Event(enable_timing=True))
+                self._ev[0].record()
+                return self
+        except Exception:
+            pass
+        import time
+
+        self._t0 = time.perf_counter()
+        return self
+
+    def __exit__(self, *exc):
+        if self._ev is not None:
+            import torch
+
+            self._ev[1].record()
+            self._ev[1].synchronize()
+            secs = self._ev[0].elapsed_time(self._ev[1]) / 1e3
+        else:
+            import time
+
+            secs = time.perf_counter() - self._t0
+        self.metrics.record_kernel_time(self.kernel, secs)
+        self.seconds = secs
+        return False

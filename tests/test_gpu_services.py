@@ -104,3 +104,27 @@ def test_scanner_gpu_batch():
     out = CryptoScanner().scan_market(market, top_k=5)
     assert len(out) == 5
     assert all(0 <= r["score"] <= 100 for r in out)
+
+
+def test_dqn_on_gpu_envs():
+    """DQN agent over the HIP vectorized envs, fully device-resident."""
+    from ai_crypto_trader_amd.data.synthetic import (
+        candles_chl_v, generate_ohlcv,
+    )
+    from ai_crypto_trader_amd.models.rl import DQNAgent, TradingVecEnv
+
+    market = candles_chl_v(generate_ohlcv(20_000, 4, seed=77))
+    env = TradingVecEnv(torch.from_numpy(market).cuda(), n_envs=128,
+                        ep_len=512, seed=3)
+    obs = env.reset()
+    agent = DQNAgent("cuda", seed=1)
+    losses = []
+    for _ in range(50):
+        a = agent.act(obs)
+        nxt, r, d = env.step(a)
+        agent.remember(obs.clone(), a, r.clone(), nxt, d.clone())
+        losses.append(agent.replay())
+        obs = env.obs
+    torch.cuda.synchronize()
+    assert np.isfinite(losses[-1])
+    assert agent.updates > 0 and agent.eps < 1.0
