@@ -285,3 +285,41 @@ def test_generic_ga_optimizes():
                            population_size=32, generations=10, seed=1)
     best2, _ = ga2.run()
     assert abs(best2["x"] - 1.0) < 0.5
+
+
+def test_serving_api():
+    """serve.py ASGI surface: analyze/scan/backtest/predict on CPU."""
+    import asyncio as _a
+
+    import httpx
+
+    from serve import build_server
+    from ai_crypto_trader_amd.data.synthetic import (
+        candles_chl_v, generate_ohlcv,
+    )
+
+    async def go():
+        app, nn = build_server("models_store_test_nonexistent", "cpu")
+        candles = candles_chl_v(generate_ohlcv(400, 1, seed=9))[0]
+        transport = httpx.ASGITransport(app=app)
+        async with httpx.AsyncClient(transport=transport,
+                                     base_url="http://t") as c:
+            r = await c.get("/healthz")
+            assert r.json()["ok"]
+            r = await c.post("/analyze", json={
+                "symbol": "X", "current_price": 1.0, "avg_volume": 1.0,
+                "rsi": 25.0, "price_change_5m": 0.5})
+            assert r.json()["decision"] in ("BUY", "SELL", "HOLD")
+            r = await c.post("/scan", json={
+                "AUSDC": candles.tolist(), "BUSDC": candles.tolist()})
+            assert len(r.json()) == 2
+            r = await c.post("/backtest", json={
+                "candles": candles.tolist(),
+                "params": {"entry_votes": 1}})
+            assert "sharpe" in r.json()
+            r = await c.post("/predict", json={
+                "symbol": "AUSDC", "candles": candles.tolist()})
+            out = r.json()
+            assert "predicted_price" in out or "error" in out
+
+    _a.run(go())
