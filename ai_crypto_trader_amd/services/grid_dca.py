@@ -23,14 +23,21 @@ from .base import Service
 class GridTradingStrategy(Service):
     name = "grid_trading"
 
-    def __init__(self, bus, exchange, symbol: str, config=None):
+    def __init__(self, bus, exchange, symbol: str, config=None,
+                 live: bool = False, order_qty: float = 1.0):
+        """live=False: internal simulation fill engine (:679-780).
+        live=True: place resting LIMIT orders on the exchange seam and
+        track their fills (:418-509)."""
         super().__init__(bus, config)
         self.exchange = exchange
         self.symbol = symbol
+        self.live = live
+        self.order_qty = order_qty
         self.levels: list[dict] = []
         self.center = 0.0
         self.fills = 0
         self.pnl = 0.0
+        self._orders: dict[str, dict] = {}    # order_id -> level
 
     def build_grid(self, center: float, regime: str = "ranging"):
         """(:347-386, :840-906) level construction."""
@@ -42,20 +49,60 @@ class GridTradingStrategy(Service):
         else:
             ratios = np.linspace(1 - rng, 1 + rng, n)
         self.center = center
+        if self.live:
+            for oid in list(self._orders):
+                self.exchange.cancel_order(self.symbol, oid)
+            self._orders.clear()
         self.levels = [
             {"price": float(center * r),
              "side": "BUY" if r < 1.0 else "SELL",
              "filled": False}
             for r in ratios if abs(r - 1.0) > 1e-9
         ]
+        if self.live:
+            for lv in self.levels:
+                o = self.exchange.create_order(
+                    self.symbol, lv["side"], "LIMIT", self.order_qty,
+                    price=lv["price"])
+                if o.status == "NEW":
+                    self._orders[o.order_id] = lv
+                elif o.status == "FILLED":
+                    lv["filled"] = True
         return self.levels
 
+    def poll_live_fills(self) -> list[dict]:
+        """Live mode: reconcile resting orders with exchange fills and
+        re-arm the grid (:418-509)."""
+        fills = []
+        for oid, lv in list(self._orders.items()):
+            o = self.exchange.orders.get(oid)
+            if o is None:
+                self._orders.pop(oid, None)
+                continue
+            if o.status == "FILLED":
+                lv["filled"] = True
+                self._orders.pop(oid)
+                fills.append({**lv, "order_id": oid, "at": time.time()})
+                if lv["side"] == "SELL":
+                    self.pnl += (lv["price"] - self.center) / self.center
+            elif o.status == "CANCELED":
+                self._orders.pop(oid)
+        self.fills += len(fills)
+        return fills
+
     def on_price(self, price: float) -> list[dict]:
-        """Simulation fill engine (:679-780): crossing a level fills it
-        and re-arms the opposite side one notch over."""
+        """Per-tick driver: simulation fill engine (:679-780), or live
+        order reconciliation when live=True."""
         fills = []
         if not self.levels:
             self.build_grid(price)
+            return fills
+        if self.live:
+            fills = self.poll_live_fills()
+            lo = min(lv["price"] for lv in self.levels)
+            hi = max(lv["price"] for lv in self.levels)
+            if price < lo * 0.995 or price > hi * 1.005:
+                self.build_grid(price)
             return fills
         for lv in self.levels:
             if lv["filled"]:

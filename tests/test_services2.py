@@ -323,3 +323,36 @@ def test_serving_api():
             assert "predicted_price" in out or "error" in out
 
     _a.run(go())
+
+
+def test_grid_live_orders():
+    """Live grid mode: resting LIMIT orders on the exchange, fills
+    reconciled on tick."""
+    async def go():
+        from ai_crypto_trader_amd.services.grid_dca import (
+            GridTradingStrategy,
+        )
+        ex = FakeExchange(initial_balance=100_000.0)
+        ex.prices["BTCUSDC"] = 1.0
+        # seed base-asset inventory so SELL limits can rest
+        ex.balances["BTC"] = 100.0
+        g = GridTradingStrategy(ex.__class__ and InProcessBus(), ex,
+                                "BTCUSDC", AppConfig(), live=True,
+                                order_qty=1.0)
+        g.build_grid(1.0)
+        assert len(g._orders) >= 8          # resting LIMIT orders
+        buy_prices = [lv["price"] for lv in g.levels
+                      if lv["side"] == "BUY"]
+        # price drops through the highest buy level -> exchange fills it
+        ex.set_price("BTCUSDC", max(buy_prices) - 1e-6)
+        fills = g.on_price(max(buy_prices) - 1e-6)
+        assert any(f["side"] == "BUY" for f in fills)
+        # price spikes through a sell level
+        sell_prices = [lv["price"] for lv in g.levels
+                       if lv["side"] == "SELL"]
+        ex.set_price("BTCUSDC", min(sell_prices) + 1e-6)
+        fills2 = g.on_price(min(sell_prices) + 1e-6)
+        assert any(f["side"] == "SELL" for f in fills2)
+        assert g.pnl > 0
+
+    asyncio.run(go())
