@@ -13,6 +13,7 @@ egress exist (ai_trader.py parity seam).
 
 from __future__ import annotations
 
+import json
 import time
 
 from ..bus.schema import Channels, Keys, TradingSignal
@@ -144,8 +145,7 @@ class AnalyzerService(Service):
             self.last_analysis[sym] = now
             news = await self.bus.hget(Keys.NEWS_ANALYSIS, sym)
             nn = await self.bus.get_json(Keys.nn_prediction(sym, "1m"))
-            import json as _json
-            news_d = _json.loads(news) if news else None
+            news_d = json.loads(news) if news else None
             analysis = self.analyst.analyze(
                 market, self.social_cache.get(sym), news_d, nn)
             sig = TradingSignal(

@@ -336,9 +336,8 @@ def test_grid_live_orders():
         ex.prices["BTCUSDC"] = 1.0
         # seed base-asset inventory so SELL limits can rest
         ex.balances["BTC"] = 100.0
-        g = GridTradingStrategy(ex.__class__ and InProcessBus(), ex,
-                                "BTCUSDC", AppConfig(), live=True,
-                                order_qty=1.0)
+        g = GridTradingStrategy(InProcessBus(), ex, "BTCUSDC",
+                                AppConfig(), live=True, order_qty=1.0)
         g.build_grid(1.0)
         assert len(g._orders) >= 8          # resting LIMIT orders
         buy_prices = [lv["price"] for lv in g.levels
