@@ -321,7 +321,11 @@ class PPOAgent:
         torch.manual_seed(seed)
         self.device = torch.device(device)
         self.net = ActorCritic(n_obs, n_act).to(device)
-        self.opt = torch.optim.Adam(self.net.parameters(), lr=lr)
+        # capturable Adam: the optimizer step must be hip-graph-replayable
+        # when use_graph captures the whole update phase
+        self.opt = torch.optim.Adam(self.net.parameters(), lr=lr,
+                                    capturable=bool(use_graph) and
+                                    self.device.type == "cuda")
         self.gamma, self.lam, self.clip = gamma, lam, clip
         self.epochs, self.minibatches = epochs, minibatches
         self.ent_coef, self.vf_coef = ent_coef, vf_coef
@@ -333,6 +337,9 @@ class PPOAgent:
         self._graph = None
         self._graph_key = None
         self._gbuf = None
+        self._ugraph = None
+        self._ugraph_key = None
+        self._uloss = None
 
     @torch.no_grad()
     def policy(self, obs):
@@ -472,6 +479,71 @@ class PPOAgent:
             g.copy_(flat[off:off + g.numel()].view_as(g))
             off += g.numel()
 
+    def _update_body(self, bufs, loss_out):
+        """The PPO update against the static rollout buffers — runs
+        eagerly AND under hip-graph capture (no host reads; losses
+        accumulate into a device tensor)."""
+        obs, act, logp_old, rew, done, val = bufs
+        adv, ret = gae_gpu(rew, val, done, self.gamma, self.lam)
+        T, E = rew.shape
+        n = T * E
+        obs_f = obs.reshape(n, -1)
+        act_f = act.reshape(n)
+        logp_f = logp_old.reshape(n)
+        adv_f = adv.reshape(n)
+        ret_f = ret.reshape(n)
+        adv_f = (adv_f - adv_f.mean()) / (adv_f.std() + 1e-8)
+        loss_out.zero_()
+        mb = n // self.minibatches
+        for _ in range(self.epochs):
+            perm = torch.randperm(n, device=obs.device)
+            for i in range(self.minibatches):
+                j = perm[i * mb:(i + 1) * mb]
+                logits, v = self.net(obs_f[j])
+                dist_ = torch.distributions.Categorical(logits=logits)
+                logp = dist_.log_prob(act_f[j])
+                ratio = torch.exp(logp - logp_f[j])
+                s1 = ratio * adv_f[j]
+                s2 = torch.clamp(ratio, 1 - self.clip,
+                                 1 + self.clip) * adv_f[j]
+                pi_loss = -torch.min(s1, s2).mean()
+                v_loss = ((v - ret_f[j]) ** 2).mean()
+                ent = dist_.entropy().mean()
+                loss = pi_loss + self.vf_coef * v_loss \
+                    - self.ent_coef * ent
+                self.opt.zero_grad(set_to_none=False)
+                loss.backward()
+                self._allreduce_grads()
+                self.opt.step()
+                loss_out[0] += pi_loss.detach()
+                loss_out[1] += v_loss.detach()
+                loss_out[2] += ent.detach()
+
+    def update_graphed(self, bufs) -> dict:
+        """Whole update phase as one hipGraph (forward+backward+capturable
+        Adam, the torch full-iteration-capture pattern)."""
+        key = tuple(id(b) for b in bufs)
+        if self._ugraph_key != key:
+            self._uloss = torch.zeros(3, device=self.device)
+            # eager warmup initializes Adam state + grad buffers
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                self._update_body(bufs, self._uloss)
+            torch.cuda.current_stream().wait_stream(s)
+            torch.cuda.synchronize()
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                self._update_body(bufs, self._uloss)
+            self._ugraph = g
+            self._ugraph_key = key
+        else:
+            self._ugraph.replay()
+        k = self.epochs * self.minibatches
+        vals = (self._uloss / k).cpu()
+        return {"pi_loss": float(vals[0]), "v_loss": float(vals[1]),
+                "entropy": float(vals[2])}
+
     def train_step(self, env: TradingVecEnv, horizon: int = 128) -> dict:
         if self.use_graph and self.device.type == "cuda":
             try:
@@ -479,6 +551,11 @@ class PPOAgent:
             except Exception:
                 self.use_graph = False        # fall back to eager forever
                 out = self.rollout(env, horizon)
+            if self.use_graph:
+                try:
+                    return self.update_graphed(out)
+                except Exception:
+                    self.use_graph = False
         else:
             out = self.rollout(env, horizon)
         return self.update(*out)
