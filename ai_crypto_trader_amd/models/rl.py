@@ -500,15 +500,17 @@ class PPOAgent:
             for i in range(self.minibatches):
                 j = perm[i * mb:(i + 1) * mb]
                 logits, v = self.net(obs_f[j])
-                dist_ = torch.distributions.Categorical(logits=logits)
-                logp = dist_.log_prob(act_f[j])
+                # manual categorical math: torch.distributions validates
+                # with a host sync, which hip-graph capture forbids
+                logp_all = torch.log_softmax(logits, dim=-1)
+                logp = logp_all.gather(1, act_f[j][:, None]).squeeze(1)
                 ratio = torch.exp(logp - logp_f[j])
                 s1 = ratio * adv_f[j]
                 s2 = torch.clamp(ratio, 1 - self.clip,
                                  1 + self.clip) * adv_f[j]
                 pi_loss = -torch.min(s1, s2).mean()
                 v_loss = ((v - ret_f[j]) ** 2).mean()
-                ent = dist_.entropy().mean()
+                ent = -(logp_all.exp() * logp_all).sum(-1).mean()
                 loss = pi_loss + self.vf_coef * v_loss \
                     - self.ent_coef * ent
                 self.opt.zero_grad(set_to_none=False)
