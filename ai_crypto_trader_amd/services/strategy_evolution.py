@@ -72,7 +72,9 @@ class StrategyEvolutionService(Service):
         if regime == "volatile" or volatility > 0.8:
             return "ga"           # broad search when conditions shift
         if self.evolutions % 3 == 2:
-            return "rl"
+            # every 3rd evolution: RL nudging — PPO on the hipGraph env
+            # when a GPU is present, DQN self-play otherwise
+            return "ppo" if str(self.device).startswith("cuda") else "rl"
         return "ga"
 
     # --- optimizers ------------------------------------------------------
@@ -131,6 +133,44 @@ class StrategyEvolutionService(Service):
             p[14] *= 1.2
         return clip_params(p[None])[0], {"rl_action": a}
 
+    def optimize_with_ppo(self, train_steps: int = 3):
+        """PPO parameter guidance (the BASELINE's PPO counterpart of the
+        reference's DQN nudging :696-975): train PPO briefly on the HIP
+        envs over the evaluation market (hipGraph-captured rollout+update
+        on GPU) and map the learned action preference — mean entry
+        propensity vs exit propensity — to risk-appetite nudges."""
+        import torch
+
+        from ..models.rl import PPOAgent, TradingVecEnv
+
+        if not str(self.device).startswith("cuda"):
+            return self.optimize_with_rl()    # CPU fallback: DQN nudging
+        market = torch.from_numpy(self.candles).to(self.device)
+        env = TradingVecEnv(market, n_envs=256,
+                            ep_len=min(1024, self.candles.shape[1] - 64),
+                            seed=self.evolutions)
+        env.reset()
+        agent = PPOAgent(self.device, seed=self.evolutions, use_graph=True)
+        for _ in range(train_steps):
+            stats = agent.train_step(env, horizon=64)
+        with torch.no_grad():
+            logits, _ = agent.net(env.obs)
+            probs = torch.softmax(logits, dim=-1).mean(dim=0)
+        buy_pref = float(probs[1])
+        sell_pref = float(probs[2])
+        p = self.current_params.copy()
+        if buy_pref > sell_pref + 0.1:        # risk-on: looser entries
+            p[10] = max(p[10] - 1, 1)         # entry_votes
+            p[14] *= 1.2                      # take_profit
+        elif sell_pref > buy_pref + 0.1:      # risk-off
+            p[10] = min(p[10] + 1, 4)
+            p[13] *= 0.8                      # tighter stop
+            p[12] *= 0.8                      # smaller size
+        return clip_params(p[None])[0], {
+            "ppo_buy_pref": buy_pref, "ppo_sell_pref": sell_pref,
+            "ppo_entropy": stats.get("entropy", 0.0),
+        }
+
     def adjust_for_regime(self, params: np.ndarray, regime: str):
         """per-regime multiplier tables (reference :302, :145-174)."""
         adj = REGIME_ADJUSTMENTS.get(regime, {})
@@ -169,6 +209,8 @@ class StrategyEvolutionService(Service):
         t0 = time.perf_counter()
         if method == "rl":
             params, perf = self.optimize_with_rl()
+        elif method == "ppo":
+            params, perf = self.optimize_with_ppo()
         else:
             params, perf = self.optimize_with_ga()
         params = self.adjust_for_regime(params, regime)

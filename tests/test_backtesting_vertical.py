@@ -99,3 +99,43 @@ def test_cli_smoke(tmp_path):
         [sys.executable, "run_backtest.py", "--data-dir", d, "list"],
         capture_output=True, text=True, timeout=120)
     assert r.returncode == 0 and "SOLUSDC" in r.stdout
+
+
+def test_custom_strategy_tester():
+    from ai_crypto_trader_amd.backtesting.strategy_tester import (
+        StrategyTester, rsi_threshold_strategy,
+    )
+    from ai_crypto_trader_amd.data.synthetic import (
+        candles_chl_v, generate_ohlcv,
+    )
+
+    candles = candles_chl_v(generate_ohlcv(4000, 1, seed=31, sigma=1.5))[0]
+    tester = StrategyTester()
+    res = tester.backtest_strategy(candles, rsi_threshold_strategy(35, 65))
+    assert res.stats["n_trades"] > 0
+    assert len(res.trades) == res.stats["n_trades"]
+    assert res.equity_curve is not None and len(res.equity_curve) == 4000
+    # every closed trade honors the stop/take mechanics
+    for tr in res.trades:
+        assert tr.exit_t > tr.entry_t
+        assert tr.reason in ("stop_loss", "take_profit", "signal")
+        if tr.reason == "stop_loss":
+            assert tr.exit_price < tr.entry_price
+        if tr.reason == "take_profit":
+            assert tr.exit_price > tr.entry_price
+
+    # arbitrary callable: never trades -> flat equity
+    res2 = tester.backtest_strategy(candles, lambda ctx: "HOLD")
+    assert res2.stats["n_trades"] == 0
+    assert res2.stats["final_equity"] == 1.0
+
+    # context carries indicators + social seam
+    seen = {}
+
+    def probe(ctx):
+        seen.update(ctx)
+        return "HOLD"
+
+    tester.backtest_strategy(candles, probe)
+    for k in ("rsi14", "macd", "bb_up", "close", "in_position"):
+        assert k in seen

@@ -128,3 +128,31 @@ def test_dqn_on_gpu_envs():
     torch.cuda.synchronize()
     assert np.isfinite(losses[-1])
     assert agent.updates > 0 and agent.eps < 1.0
+
+
+def test_ppo_evolution_optimizer_gpu():
+    """optimize_with_ppo: PPO trains on the hipGraph-captured envs over the
+    evaluation market and returns a valid nudged parameter vector."""
+    from ai_crypto_trader_amd.backtesting.strategy import (
+        PARAM_BOUNDS, clip_params,
+    )
+    from ai_crypto_trader_amd.bus.message_bus import InProcessBus
+    from ai_crypto_trader_amd.config import AppConfig
+    from ai_crypto_trader_amd.data.synthetic import (
+        candles_chl_v, generate_ohlcv,
+    )
+    from ai_crypto_trader_amd.services.strategy_evolution import (
+        StrategyEvolutionService,
+    )
+
+    candles = candles_chl_v(generate_ohlcv(3000, 2, seed=5))
+    svc = StrategyEvolutionService(InProcessBus(), AppConfig(),
+                                   candles=candles, device="cuda:0")
+    params, perf = svc.optimize_with_ppo(train_steps=2)
+    assert "ppo_buy_pref" in perf and "ppo_sell_pref" in perf
+    assert 0.0 <= perf["ppo_buy_pref"] <= 1.0
+    # returned params respect the bounds (clip_params applied)
+    np.testing.assert_array_equal(params, clip_params(params[None])[0])
+    lo = np.array([b[0] for b in PARAM_BOUNDS], np.float32)
+    hi = np.array([b[1] for b in PARAM_BOUNDS], np.float32)
+    assert np.all(params >= lo - 1e-6) and np.all(params <= hi + 1e-6)

@@ -355,3 +355,50 @@ def test_grid_live_orders():
         assert g.pnl > 0
 
     asyncio.run(go())
+
+
+def test_strategy_evolution_service_cpu():
+    """evolve_once on CPU: GA path runs, hot-swaps params, publishes the
+    update + registry event; hybrid method selection prefers PPO only on
+    GPU devices."""
+    from ai_crypto_trader_amd.bus.schema import Channels, Keys
+    from ai_crypto_trader_amd.data.synthetic import (
+        candles_chl_v, generate_ohlcv,
+    )
+    from ai_crypto_trader_amd.services.strategy_evolution import (
+        StrategyEvolutionService,
+    )
+
+    candles = candles_chl_v(generate_ohlcv(1500, 2, seed=9))
+    cfg = AppConfig()
+    cfg.evolution.population_size = 16
+    cfg.evolution.generations = 2
+
+    async def go():
+        bus = InProcessBus()
+        svc = StrategyEvolutionService(bus, cfg, candles=candles,
+                                       device="cpu")
+        sub = bus.subscribe(Channels.STRATEGY_EVOLUTION_UPDATES)
+        perf = await svc.evolve_once()
+        assert perf["method"] in ("ga", "rl")
+        assert perf["seconds"] > 0
+        # hot swap happened: key set + channel published + version logged
+        params = await bus.get_json(Keys.STRATEGY_PARAMS)
+        assert params is not None and "stop_loss_pct" in params
+        ch, msg = await asyncio.wait_for(sub.get(), timeout=2)
+        assert msg["strategy_id"].startswith("evolved-")
+        assert svc.model_versions and svc.evolutions == 1
+        # live perf evaluation (the strategy_performance key content)
+        d = svc.evaluate_params(svc.current_params)
+        assert "win_rate" in d and "profit_factor" in d
+
+    asyncio.run(go())
+
+    # hybrid method selection: ppo only on cuda
+    bus = InProcessBus()
+    svc = StrategyEvolutionService(bus, cfg, candles=candles, device="cpu")
+    svc.evolutions = 2
+    assert svc.select_method("ranging", 0.2) == "rl"
+    svc.device = "cuda:0"
+    assert svc.select_method("ranging", 0.2) == "ppo"
+    assert svc.select_method("volatile", 0.9) == "ga"
