@@ -3,7 +3,8 @@ services/market_regime_service.py + services/utils/market_regime_detector.py).
 
 Features per window: return, volatility, trend strength (rolling slope),
 RSI, MACD, BB width (market_regime_detector.py:64-110). Detectors: KMeans
-/ GMM (torch, GPU-capable) and the reference's heuristic cluster->label
+/ GMM / Gaussian HMM (torch, GPU-capable) and the reference's heuristic
+cluster->label
 mapping by mean return & volatility (:226-296), plus the rule-based
 fallback (market_regime_service.py:503-604). Publishes `strategy_switch`
 on regime changes; maintains `current_market_regime` /
@@ -113,6 +114,82 @@ class GMMTorch:
         return (lp - torch.logsumexp(lp, 1, keepdim=True)).exp()
 
 
+class GaussianHMMTorch:
+    """Diagonal-covariance Gaussian HMM via Baum-Welch EM, log-space
+    forward-backward (market_regime_detector.py:150 — hmmlearn GaussianHMM
+    equivalent; data is tiny so a per-timestep torch loop is fine)."""
+
+    def __init__(self, k: int = 4, iters: int = 30, seed: int = 0):
+        self.k = k
+        self.iters = iters
+        self.seed = seed
+
+    def _emission_logp(self, X):
+        d = X[:, None, :] - self.mu[None]                    # (T, k, F)
+        return (-0.5 * ((d ** 2) / self.var[None]).sum(-1)
+                - 0.5 * torch.log(self.var[None]).sum(-1))
+
+    def fit(self, X: torch.Tensor):
+        T = X.shape[0]
+        k = self.k
+        km = KMeansTorch(k, 20, self.seed).fit(X)
+        self.mu = km.centers.clone()
+        self.var = torch.ones_like(self.mu) * X.var(dim=0, keepdim=True)
+        self.var = torch.clamp(self.var, min=1e-6)
+        dev = X.device
+        # sticky prior: regimes persist (diag-heavy transition init)
+        self.logA = torch.log(torch.full((k, k), 0.05 / (k - 1), device=dev)
+                              .fill_diagonal_(0.95))
+        self.logpi = torch.full((k,), -np.log(k), device=dev)
+        for _ in range(self.iters):
+            logb = self._emission_logp(X)                    # (T, k)
+            # forward / backward in log space
+            la = torch.empty(T, k, device=dev)
+            lb = torch.zeros(T, k, device=dev)
+            la[0] = self.logpi + logb[0]
+            for t in range(1, T):
+                la[t] = logb[t] + torch.logsumexp(
+                    la[t - 1][:, None] + self.logA, dim=0)
+            for t in range(T - 2, -1, -1):
+                lb[t] = torch.logsumexp(
+                    self.logA + (logb[t + 1] + lb[t + 1])[None], dim=1)
+            lg = la + lb
+            lg = lg - torch.logsumexp(lg, dim=1, keepdim=True)
+            g = lg.exp()                                     # (T, k) gammas
+            # xi sums for the transition update
+            xi = torch.zeros(k, k, device=dev)
+            ll = torch.logsumexp(la[-1], dim=0)
+            for t in range(T - 1):
+                lx = (la[t][:, None] + self.logA
+                      + (logb[t + 1] + lb[t + 1])[None] - ll)
+                xi += lx.exp()
+            self.logA = torch.log(xi / (g[:-1].sum(0)[:, None] + 1e-9)
+                                  + 1e-12)
+            self.logpi = torch.log(g[0] + 1e-12)
+            nk = g.sum(0) + 1e-9
+            self.mu = (g.T @ X) / nk[:, None]
+            self.var = torch.clamp(
+                (g.T @ (X ** 2)) / nk[:, None] - self.mu ** 2, min=1e-6)
+        return self
+
+    def predict(self, X: torch.Tensor) -> torch.Tensor:
+        """Viterbi path."""
+        T = X.shape[0]
+        logb = self._emission_logp(X)
+        delta = self.logpi + logb[0]
+        back = torch.zeros(T, self.k, dtype=torch.long, device=X.device)
+        for t in range(1, T):
+            s = delta[:, None] + self.logA
+            best = s.max(dim=0)
+            delta = best.values + logb[t]
+            back[t] = best.indices
+        path = torch.empty(T, dtype=torch.long, device=X.device)
+        path[-1] = delta.argmax()
+        for t in range(T - 2, -1, -1):
+            path[t] = back[t + 1, path[t + 1]]
+        return path
+
+
 def label_clusters(X: np.ndarray, assign: np.ndarray, k: int) -> dict:
     """Heuristic cluster -> regime mapping by mean return & volatility
     (market_regime_detector.py:226-296)."""
@@ -201,7 +278,8 @@ class MarketRegimeService(Service):
         if self.device != "cpu":
             X = X.to(self.device)
         Xn = (X - X.mean(0)) / (X.std(0) + 1e-9)
-        cls = GMMTorch if method == "gmm" else KMeansTorch
+        cls = {"gmm": GMMTorch, "hmm": GaussianHMMTorch}.get(
+            method, KMeansTorch)
         self.model = cls(self.config.regime.n_regimes,
                          seed=self.config.seed).fit(Xn)
         assign = self.model.predict(Xn).cpu().numpy()
