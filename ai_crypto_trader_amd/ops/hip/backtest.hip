@@ -2,8 +2,14 @@
 //
 // Replaces the reference's per-candle Python loop
 // (backtesting/strategy_tester.py:190-300, services/strategy_evaluation.py:777-878)
-// with one CDNA4 lane per (param-set x symbol) marching candles:
-//   - a block = 256 lanes = 256 param-sets of one symbol
+// with CDNA4 lanes marching candles:
+//   - a block = 256 lanes; each lane simulates ILP param-sets of one
+//     symbol. Measured: ILP=2 (two independent dependency chains/lane,
+//     148 VGPR, 3 waves/SIMD) is 10% SLOWER than ILP=1 (96 VGPR, 5
+//     waves/SIMD) — 323 vs 290 G candles/s at pop=1024 x 64 sym x 16
+//     segments: wave-level latency hiding beats in-lane ILP here because
+//     both chains stall on the same LDS broadcast + f64 sequence. ILP=1
+//     is the default; BT_FORCE_ILP2=1 re-runs the experiment.
 //   - candle tiles staged in LDS with a MAX_WIN-candle HALO: close history
 //     is shared by every lane of the block, so the Bollinger "ring buffer"
 //     is just a broadcast read of close[t-W] from the halo tile — no
@@ -15,8 +21,9 @@
 //     matches backtesting/engine_cpu.py bit-for-bit (fp-contract off, same
 //     operation order, f64 Bollinger sums both sides).
 //
-// Grid: nsym * ceil(P/256) blocks, XCD-affine map keeps the blocks of one
-// symbol on one XCD so their shared candle stream stays in that XCD's L2.
+// Grid: nsym * ceil(P/(256*ILP)) blocks, XCD-affine map keeps the blocks
+// of one symbol on one XCD so their shared candle stream stays in that
+// XCD's L2.
 
 #include "common.hpp"
 
@@ -44,6 +51,176 @@ struct LaneParams {
     float stoch_os, stoch_ob, will_os, will_ob;
 };
 
+// one strategy's full simulation state (indicators + position + metrics)
+struct BtState {
+    LaneParams q;
+    float ema_f, ema_s, sig;
+    float avg_gain, avg_loss;
+    double bb_sum, bb_sum2, inv_w;    // f64 sums: f32 drifts + cancels
+    float cash, units;
+    bool in_pos;
+    float entry_cost, entry_price;
+    float stop, tp, peak;
+    float equity, max_eq, max_dd;
+    float n_trades, wins, gross_p, gross_l;
+    float sum_ret, sum_ret2;
+
+    __device__ void load(const float* __restrict__ pr, float initial_equity)
+    {
+        q.inv_rsi_p = 1.0f / fmaxf(floorf(pr[0]), 1.0f);
+        q.rsi_os = pr[1]; q.rsi_ob = pr[2];
+        q.a_f = 2.0f / (pr[3] + 1.0f);
+        q.a_s = 2.0f / (pr[4] + 1.0f);
+        q.a_sig = 2.0f / (pr[5] + 1.0f);
+        q.bb_w = min(max((int)pr[6], 2), BT_MAXWIN);
+        q.bb_k = pr[7]; q.bb_bth = pr[8]; q.bb_sth = pr[9];
+        q.entry_v = (int)pr[10]; q.exit_v = (int)pr[11];
+        q.size_pct = pr[12]; q.sl_pct = pr[13]; q.tp_pct = pr[14];
+        q.trail_pct = pr[15]; q.trail_act = pr[16];
+        q.stoch_os = pr[17]; q.stoch_ob = pr[18];
+        q.will_os = q.stoch_os - 100.0f;
+        q.will_ob = q.stoch_ob - 100.0f;
+        ema_f = ema_s = sig = 0.f;
+        avg_gain = avg_loss = 0.f;
+        bb_sum = bb_sum2 = 0.0;
+        inv_w = 1.0 / (double)q.bb_w;
+        cash = initial_equity; units = 0.f;
+        in_pos = false;
+        entry_cost = entry_price = stop = tp = peak = 0.f;
+        equity = initial_equity; max_eq = initial_equity; max_dd = 0.f;
+        n_trades = wins = gross_p = gross_l = 0.f;
+        sum_ret = sum_ret2 = 0.f;
+    }
+
+    __device__ void step(int t, float close, float high, float low,
+                         float oldc, float change, float4 sv)
+    {
+#pragma clang fp contract(off)           // match the numpy f32 reference
+        // --- 1. indicators -------------------------------------------
+        if (t == 0) { ema_f = close; ema_s = close; }
+        else {
+            ema_f += q.a_f * (close - ema_f);
+            ema_s += q.a_s * (close - ema_s);
+        }
+        float macd = ema_f - ema_s;
+        sig += q.a_sig * (macd - sig);
+        float macd_hist = macd - sig;
+
+        float gain = fmaxf(change, 0.0f);
+        float loss = fmaxf(-change, 0.0f);
+        avg_gain += (gain - avg_gain) * q.inv_rsi_p;
+        avg_loss += (loss - avg_loss) * q.inv_rsi_p;
+        // division-free RSI votes (engine_cpu.py): rsi<thr <=>
+        // 100*ag < thr*(ag+al')
+        float rsi_num = 100.0f * avg_gain;
+        float rsi_den = avg_gain + fmaxf(avg_loss, BT_EPS);
+
+        // Bollinger: close[t - W] comes from the shared halo tile
+        // (== the zero-initialized per-lane ring of engine_cpu.py)
+        double old = (double)oldc;
+        double c64 = (double)close;
+        bb_sum += c64 - old;
+        bb_sum2 += c64 * c64 - old * old;
+        double inv_cnt = inv_w;
+        if (t < BT_MAXWIN && t + 1 < q.bb_w)   // uniformly skipped t>=32
+            inv_cnt = 1.0 / (t + 1.0);
+        double mean64 = bb_sum * inv_cnt;
+        double var64 = fmax(bb_sum2 * inv_cnt - mean64 * mean64, 0.0);
+        float mean = (float)mean64;
+        float std_ = sqrtf((float)var64);
+        float band = q.bb_k * std_;
+        // division-free BB votes (engine_cpu.py): pos<thr <=> num<thr*den
+        float bb_num = close - (mean - band);
+        float bb_den = fmaxf(2.0f * band, BT_EPS);
+
+        // --- 2. votes (6-indicator TradingSignal voting) -------------
+        int net = 0;
+        if (t >= BT_WARMUP) {
+            int buy = (rsi_num < q.rsi_os * rsi_den) +
+                      (macd_hist > 0.0f) +
+                      (bb_num < q.bb_bth * bb_den) +
+                      (sv.x < q.stoch_os * sv.z) +
+                      (sv.y < q.will_os * sv.z) +
+                      (sv.w > 0.0f);
+            int sell = (rsi_num > q.rsi_ob * rsi_den) +
+                       (macd_hist < 0.0f) +
+                       (bb_num > q.bb_sth * bb_den) +
+                       (sv.x > q.stoch_ob * sv.z) +
+                       (sv.y > q.will_ob * sv.z) +
+                       (sv.w < 0.0f);
+            net = buy - sell;
+        }
+
+        // --- 3. position management ----------------------------------
+        if (in_pos) {
+            peak = fmaxf(peak, high);
+            bool trail_on = (q.trail_pct > 0.0f) &&
+                            (peak >= entry_price * (1.0f + q.trail_act));
+            if (trail_on)
+                stop = fmaxf(stop, peak * (1.0f - q.trail_pct));
+            bool hit_sl = low <= stop;
+            bool hit_tp = !hit_sl && high >= tp;
+            bool hit_sig = !hit_sl && !hit_tp && net <= -q.exit_v;
+            if (hit_sl || hit_tp || hit_sig) {
+                float exit_price = hit_sl ? stop : (hit_tp ? tp : close);
+                float proceeds = units * exit_price * (1.0f - BT_FEE);
+                float pnl = proceeds - entry_cost;
+                cash += proceeds;
+                n_trades += 1.0f;
+                wins += (pnl > 0.0f) ? 1.0f : 0.0f;
+                gross_p += fmaxf(pnl, 0.0f);
+                gross_l += fmaxf(-pnl, 0.0f);
+                units = 0.0f;
+                in_pos = false;
+            }
+        } else if (t >= BT_WARMUP && net >= q.entry_v) {
+            float cost = fminf(q.size_pct * equity, cash);
+            units = cost * (1.0f - BT_FEE) / close;
+            cash -= cost;
+            entry_cost = cost;
+            entry_price = close;
+            stop = close * (1.0f - q.sl_pct);
+            tp = close * (1.0f + q.tp_pct);
+            peak = close;
+            in_pos = true;
+        }
+
+        // --- 4. mark to market ---------------------------------------
+        // flat lanes: new_eq == cash == equity exactly, so r == 0 and
+        // every accumulator is unchanged — skipping is bit-identical
+        // to engine_cpu.py (which computes r = cash/cash - 1 = 0)
+        if (units != 0.0f || cash != equity) {
+            float new_eq = cash + units * close;
+            float r = new_eq / equity - 1.0f;
+            sum_ret += r;
+            sum_ret2 += r * r;
+            equity = new_eq;
+            max_eq = fmaxf(max_eq, equity);
+            max_dd = fmaxf(max_dd, (max_eq - equity) / max_eq);
+        }
+    }
+
+    __device__ void finalize(float* __restrict__ out, int T) const
+    {
+#pragma clang fp contract(off)
+        // (engine_cpu.finalize_metrics semantics)
+        float n = (float)max(T, 1);
+        float mean_r = sum_ret / n;
+        float var_r = fmaxf(sum_ret2 / n - mean_r * mean_r, 0.0f);
+        float sharpe = mean_r / fmaxf(sqrtf(var_r), BT_EPS) * BT_ANNUALIZE;
+        if (!(n_trades > 0.0f)) sharpe = 0.0f;
+        float win_rate = wins / fmaxf(n_trades, 1.0f);
+        float fitness = (n_trades > 0.0f)
+                            ? sharpe + win_rate - 2.0f * max_dd
+                            : -1.0f;
+        out[0] = equity; out[1] = n_trades; out[2] = wins;
+        out[3] = gross_p; out[4] = gross_l; out[5] = max_dd;
+        out[6] = sum_ret; out[7] = sum_ret2;
+        out[8] = sharpe; out[9] = fitness;
+    }
+};
+
+template <int ILP>
 __global__ void __launch_bounds__(BT_BLOCK) backtest_kernel(
     const float* __restrict__ candles,   // (nsym, T, 4)
     const float* __restrict__ pop,       // (P, NPARAM)
@@ -76,45 +253,20 @@ __global__ void __launch_bounds__(BT_BLOCK) backtest_kernel(
     }
 
     const int tid = threadIdx.x;
-    const int p = chunk * BT_BLOCK + tid;
-    const bool active = p < P;
-    const int pl = active ? p : 0;       // inactive lanes shadow lane 0
+    // lane tid simulates params {base + tid, base + tid + 256, ...}
+    const int pbase = chunk * BT_BLOCK * ILP + tid;
 
-    // ---- load + derive params ------------------------------------------
-    LaneParams q;
-    {
-        const float* pr = pop + (long)pl * BT_NPARAM;
-        q.inv_rsi_p = 1.0f / fmaxf(floorf(pr[0]), 1.0f);
-        q.rsi_os = pr[1]; q.rsi_ob = pr[2];
-        q.a_f = 2.0f / (pr[3] + 1.0f);
-        q.a_s = 2.0f / (pr[4] + 1.0f);
-        q.a_sig = 2.0f / (pr[5] + 1.0f);
-        q.bb_w = min(max((int)pr[6], 2), BT_MAXWIN);
-        q.bb_k = pr[7]; q.bb_bth = pr[8]; q.bb_sth = pr[9];
-        q.entry_v = (int)pr[10]; q.exit_v = (int)pr[11];
-        q.size_pct = pr[12]; q.sl_pct = pr[13]; q.tp_pct = pr[14];
-        q.trail_pct = pr[15]; q.trail_act = pr[16];
-        q.stoch_os = pr[17]; q.stoch_ob = pr[18];
-        q.will_os = q.stoch_os - 100.0f;
-        q.will_ob = q.stoch_ob - 100.0f;
+    BtState st[ILP];
+    bool act[ILP];
+#pragma unroll
+    for (int i = 0; i < ILP; ++i) {
+        const int p = pbase + i * BT_BLOCK;
+        act[i] = p < P;
+        st[i].load(pop + (long)(act[i] ? p : 0) * BT_NPARAM,
+                   initial_equity);
     }
 
-    // ---- state ----------------------------------------------------------
-    float ema_f = 0.f, ema_s = 0.f, sig = 0.f;
-    float avg_gain = 0.f, avg_loss = 0.f;
-    // f64 rolling sums (engine_cpu.py rationale: f32 drifts + cancels)
-    double bb_sum = 0.0, bb_sum2 = 0.0;
-    const double inv_w = 1.0 / (double)q.bb_w;
     float prev_close = 0.f;
-
-    float cash = initial_equity, units = 0.f;
-    bool in_pos = false;
-    float entry_cost = 0.f, entry_price = 0.f;
-    float stop = 0.f, tp = 0.f, peak = 0.f;
-    float equity = initial_equity, max_eq = initial_equity, max_dd = 0.f;
-    float n_trades = 0.f, wins = 0.f, gross_p = 0.f, gross_l = 0.f;
-    float sum_ret = 0.f, sum_ret2 = 0.f;
-
     const float4* sym_candles =
         reinterpret_cast<const float4*>(candles + (long)sym * T * 4);
 
@@ -168,137 +320,23 @@ __global__ void __launch_bounds__(BT_BLOCK) backtest_kernel(
             const float close = chist[tt + BT_HALO];
             const float high = hl[tt + BT_HALO][0];
             const float low = hl[tt + BT_HALO][1];
-            const float oldc = chist[tt + BT_HALO - q.bb_w];
-
-            // --- 1. indicators ---------------------------------------
-            float change;
-            if (t == 0) {
-                ema_f = close; ema_s = close; change = 0.0f;
-            } else {
-                ema_f += q.a_f * (close - ema_f);
-                ema_s += q.a_s * (close - ema_s);
-                change = close - prev_close;
-            }
-            float macd = ema_f - ema_s;
-            sig += q.a_sig * (macd - sig);
-            float macd_hist = macd - sig;
-
-            float gain = fmaxf(change, 0.0f);
-            float loss = fmaxf(-change, 0.0f);
-            avg_gain += (gain - avg_gain) * q.inv_rsi_p;
-            avg_loss += (loss - avg_loss) * q.inv_rsi_p;
-            // division-free RSI votes (engine_cpu.py): rsi<thr <=>
-            // 100*ag < thr*(ag+al')
-            float rsi_num = 100.0f * avg_gain;
-            float rsi_den = avg_gain + fmaxf(avg_loss, BT_EPS);
-
-            // Bollinger: close[t - W] comes from the shared halo tile
-            // (== the zero-initialized per-lane ring of engine_cpu.py)
-            double old = (double)oldc;
-            double c64 = (double)close;
-            bb_sum += c64 - old;
-            bb_sum2 += c64 * c64 - old * old;
-            double inv_cnt = inv_w;
-            if (t < BT_MAXWIN && t + 1 < q.bb_w)   // uniformly skipped t>=32
-                inv_cnt = 1.0 / (t + 1.0);
-            double mean64 = bb_sum * inv_cnt;
-            double var64 = fmax(bb_sum2 * inv_cnt - mean64 * mean64, 0.0);
-            float mean = (float)mean64;
-            float std_ = sqrtf((float)var64);
-            float band = q.bb_k * std_;
-            // division-free BB votes (engine_cpu.py): pos<thr <=> num<thr*den
-            float bb_num = close - (mean - band);
-            float bb_den = fmaxf(2.0f * band, BT_EPS);
-
+            const float change = (t == 0) ? 0.0f : close - prev_close;
+            const float4 sv = sh_vote[tt];   // b128 broadcast
+#pragma unroll
+            for (int i = 0; i < ILP; ++i)
+                st[i].step(t, close, high, low,
+                           chist[tt + BT_HALO - st[i].q.bb_w], change, sv);
             prev_close = close;
-
-            // --- 2. votes (6-indicator TradingSignal voting) ---------
-            int net = 0;
-            if (t >= BT_WARMUP) {
-                const float4 sv = sh_vote[tt];   // st_num, wl_num, srange,
-                                                 // trend (b128 broadcast)
-                int buy = (rsi_num < q.rsi_os * rsi_den) +
-                          (macd_hist > 0.0f) +
-                          (bb_num < q.bb_bth * bb_den) +
-                          (sv.x < q.stoch_os * sv.z) +
-                          (sv.y < q.will_os * sv.z) +
-                          (sv.w > 0.0f);
-                int sell = (rsi_num > q.rsi_ob * rsi_den) +
-                           (macd_hist < 0.0f) +
-                           (bb_num > q.bb_sth * bb_den) +
-                           (sv.x > q.stoch_ob * sv.z) +
-                           (sv.y > q.will_ob * sv.z) +
-                           (sv.w < 0.0f);
-                net = buy - sell;
-            }
-
-            // --- 3. position management ------------------------------
-            if (in_pos) {
-                peak = fmaxf(peak, high);
-                bool trail_on = (q.trail_pct > 0.0f) &&
-                                (peak >= entry_price * (1.0f + q.trail_act));
-                if (trail_on)
-                    stop = fmaxf(stop, peak * (1.0f - q.trail_pct));
-                bool hit_sl = low <= stop;
-                bool hit_tp = !hit_sl && high >= tp;
-                bool hit_sig = !hit_sl && !hit_tp && net <= -q.exit_v;
-                if (hit_sl || hit_tp || hit_sig) {
-                    float exit_price = hit_sl ? stop : (hit_tp ? tp : close);
-                    float proceeds = units * exit_price * (1.0f - BT_FEE);
-                    float pnl = proceeds - entry_cost;
-                    cash += proceeds;
-                    n_trades += 1.0f;
-                    wins += (pnl > 0.0f) ? 1.0f : 0.0f;
-                    gross_p += fmaxf(pnl, 0.0f);
-                    gross_l += fmaxf(-pnl, 0.0f);
-                    units = 0.0f;
-                    in_pos = false;
-                }
-            } else if (t >= BT_WARMUP && net >= q.entry_v) {
-                float cost = fminf(q.size_pct * equity, cash);
-                units = cost * (1.0f - BT_FEE) / close;
-                cash -= cost;
-                entry_cost = cost;
-                entry_price = close;
-                stop = close * (1.0f - q.sl_pct);
-                tp = close * (1.0f + q.tp_pct);
-                peak = close;
-                in_pos = true;
-            }
-
-            // --- 4. mark to market -----------------------------------
-            // flat lanes: new_eq == cash == equity exactly, so r == 0 and
-            // every accumulator is unchanged — skipping is bit-identical
-            // to engine_cpu.py (which computes r = cash/cash - 1 = 0)
-            if (units != 0.0f || cash != equity) {
-                float new_eq = cash + units * close;
-                float r = new_eq / equity - 1.0f;
-                sum_ret += r;
-                sum_ret2 += r * r;
-                equity = new_eq;
-                max_eq = fmaxf(max_eq, equity);
-                max_dd = fmaxf(max_dd, (max_eq - equity) / max_eq);
-            }
         }
     }
 
-    if (!active) return;
-
-    // ---- finalize (engine_cpu.finalize_metrics semantics) ---------------
-    float n = (float)max(T, 1);
-    float mean_r = sum_ret / n;
-    float var_r = fmaxf(sum_ret2 / n - mean_r * mean_r, 0.0f);
-    float sharpe = mean_r / fmaxf(sqrtf(var_r), BT_EPS) * BT_ANNUALIZE;
-    if (!(n_trades > 0.0f)) sharpe = 0.0f;
-    float win_rate = wins / fmaxf(n_trades, 1.0f);
-    float fitness = (n_trades > 0.0f)
-                        ? sharpe + win_rate - 2.0f * max_dd
-                        : -1.0f;
-
-    float* out = metrics + ((long)p * nsym + sym) * BT_NMETRIC;
-    out[0] = equity; out[1] = n_trades; out[2] = wins;
-    out[3] = gross_p; out[4] = gross_l; out[5] = max_dd;
-    out[6] = sum_ret; out[7] = sum_ret2; out[8] = sharpe; out[9] = fitness;
+#pragma unroll
+    for (int i = 0; i < ILP; ++i) {
+        const int p = pbase + i * BT_BLOCK;
+        if (act[i])
+            st[i].finalize(metrics + ((long)p * nsym + sym) * BT_NMETRIC,
+                           T);
+    }
 }
 
 }  // namespace
@@ -306,9 +344,23 @@ __global__ void __launch_bounds__(BT_BLOCK) backtest_kernel(
 extern "C" void launch_backtest(const float* candles, const float* pop,
                                 float* metrics, int nsym, int T, int P,
                                 float initial_equity, hipStream_t stream) {
-    int chunks = (P + BT_BLOCK - 1) / BT_BLOCK;
-    dim3 grid(nsym * chunks);
-    hipLaunchKernelGGL(backtest_kernel, grid, dim3(BT_BLOCK), 0, stream,
-                       candles, pop, metrics, nsym, T, P, chunks,
-                       initial_equity);
+    // ILP=1 is the measured winner (see header); BT_FORCE_ILP2=1 opts
+    // into the 2-chain variant for perf A/Bs on future hardware.
+    static const bool force2 = [] {
+        const char* e = getenv("BT_FORCE_ILP2");
+        return e && e[0] == '1';
+    }();
+    if (force2 && P % (BT_BLOCK * 2) == 0) {
+        int chunks = P / (BT_BLOCK * 2);
+        hipLaunchKernelGGL(backtest_kernel<2>, dim3(nsym * chunks),
+                           dim3(BT_BLOCK), 0, stream,
+                           candles, pop, metrics, nsym, T, P, chunks,
+                           initial_equity);
+    } else {
+        int chunks = (P + BT_BLOCK - 1) / BT_BLOCK;
+        hipLaunchKernelGGL(backtest_kernel<1>, dim3(nsym * chunks),
+                           dim3(BT_BLOCK), 0, stream,
+                           candles, pop, metrics, nsym, T, P, chunks,
+                           initial_equity);
+    }
 }
