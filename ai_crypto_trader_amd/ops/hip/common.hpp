@@ -9,13 +9,19 @@
 constexpr int WAVE = 64;   // CDNA wavefront width (gfx950)
 
 // ---------------------------------------------------------------------------
-// Philox4x32-10 counter-based RNG (Salmon et al. 2011). Stateless: the
+// Philox4x32-7 counter-based RNG (Salmon et al. 2011). Stateless: the
 // (seed, counter) pair fully determines the output, so every lane of every
 // kernel draws independent, reproducible streams — used by the Monte-Carlo
 // path generator (replacing monte_carlo_service.py:264-273's
 // np.random.standard_normal) and the GA mutation ops
 // (genetic_algorithm.py:191-223 semantics).
+// 7 rounds: the Random123 paper's BigCrush-passing round count (10 is the
+// library default safety margin). The MC pathgen is RNG-issue-bound on
+// MI355X (PMC: 75 VALU instrs per MFMA instr), so the 3 saved rounds are
+// ~10% of the whole kernel. ops/montecarlo.py's numpy twin uses the same
+// count — change them together.
 // ---------------------------------------------------------------------------
+constexpr int PHILOX_ROUNDS = 7;
 struct Philox4 {
     uint32_t x, y, z, w;
 };
@@ -33,7 +39,7 @@ DEV_INLINE Philox4 philox4x32(uint64_t seed, uint64_t ctr_lo, uint64_t ctr_hi) {
     uint32_t c0 = (uint32_t)ctr_lo, c1 = (uint32_t)(ctr_lo >> 32);
     uint32_t c2 = (uint32_t)ctr_hi, c3 = (uint32_t)(ctr_hi >> 32);
 #pragma unroll
-    for (int r = 0; r < 10; ++r) {
+    for (int r = 0; r < PHILOX_ROUNDS; ++r) {
         uint32_t hi0, hi1;
         uint32_t lo0 = mulhilo(M0, c0, &hi0);
         uint32_t lo1 = mulhilo(M1, c2, &hi1);
@@ -53,12 +59,19 @@ DEV_INLINE float u32_to_unit(uint32_t v) {
 }
 
 // Box-Muller: two N(0,1) from two U(0,1].
+// Instruction-minimal for the RNG-issue-bound MC pathgen:
+//  - V_SIN/V_COS_F32 natively compute sin/cos(2*pi*x) ("turns" input), so
+//    the uniform u2 in (0,1] is EXACTLY the hardware domain: no 2*pi
+//    multiply, no range reduction (__sincosf emits a 1/2pi mul + guards).
+//  - -2*ln(u) = -2ln2 * log2(u): one fused constant on the raw v_log_f32.
+// The numpy twin (ops/montecarlo.py) keeps sin(2*pi*u)/cos/log — identical
+// math, transcendental rounding differs by ulps (tests use tolerances).
 DEV_INLINE float2 box_muller(uint32_t a, uint32_t b) {
     float u1 = u32_to_unit(a);
     float u2 = u32_to_unit(b);
-    float r = sqrtf(-2.0f * __logf(u1));
-    float s, c;
-    __sincosf(6.2831853071795864f * u2, &s, &c);
+    float r = sqrtf(-1.3862943611f * __builtin_amdgcn_logf(u1));
+    float s = __builtin_amdgcn_sinf(u2);
+    float c = __builtin_amdgcn_cosf(u2);
     return make_float2(r * c, r * s);
 }
 

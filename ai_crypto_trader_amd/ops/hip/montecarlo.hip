@@ -6,7 +6,12 @@
 //   z ~ Philox->Box-Muller (counter-based: reproducible, no state)
 //   logS[a] += drift[a] + sum_k cvol[k][a] * z[k]   (cvol = vol*L*sqrt(dt),
 //                                                    L = Cholesky factor)
-//   V_t = sum_a w[a] * exp(logS[a])  -> running max -> max drawdown
+//   V_t = sum_a w[a] * exp2(logS[a])  -> running max -> max drawdown
+// BASE-2 log space: the wrapper pre-scales drift and cvol by log2(e)
+// (ops/montecarlo.py _gbm_terms, shared with the CPU reference), so the
+// per-asset-per-step exponential is native v_exp_f32 with no log2e
+// multiply — the kernel is VALU-issue-bound and exp runs at 1/4 rate, so
+// every instruction on that path counts.
 // cvol is staged in LDS ([k][a] row-major, float4-broadcast reads: every
 // lane reads the same address -> conflict-free broadcast), so the inner
 // correlation loop is pure FMA on registers.
@@ -82,10 +87,10 @@ __global__ void __launch_bounds__(256) mc_paths_kernel(
 #pragma unroll
             for (int a = 0; a < A; ++a) {
                 logS[a] += lds_drift[a];
-                V += lds_wS0[a] * __expf(logS[a]);
+                V += lds_wS0[a] * __builtin_amdgcn_exp2f(logS[a]);
             }
             vmax = fmaxf(vmax, V);
-            mdd = fmaxf(mdd, (vmax - V) / vmax);
+            mdd = fmaxf(mdd, (vmax - V) * __builtin_amdgcn_rcpf(vmax));  // 1-ulp rcp: dd is a statistic
         }
         final_value[path] = V;
         max_dd[path] = mdd;
@@ -219,7 +224,7 @@ __global__ void __launch_bounds__(512) mc_paths_mfma_kernel(
 #pragma unroll
                 for (int r = 0; r < 4; ++r) {
                     logS[mt][ct][r] += r_drift[mt][r];
-                    part += r_w[mt][r] * __expf(logS[mt][ct][r]);
+                    part += r_w[mt][r] * __builtin_amdgcn_exp2f(logS[mt][ct][r]);
                 }
             }
             // sum over the 4 lanes sharing this column (l ^ 16, l ^ 32)
@@ -227,7 +232,7 @@ __global__ void __launch_bounds__(512) mc_paths_mfma_kernel(
             part += __shfl_xor(part, 32, 64);
             V[ct] = part;
             vmax[ct] = fmaxf(vmax[ct], part);
-            mdd[ct] = fmaxf(mdd[ct], (vmax[ct] - part) / vmax[ct]);
+            mdd[ct] = fmaxf(mdd[ct], (vmax[ct] - part) * __builtin_amdgcn_rcpf(vmax[ct]));
         }
     }
 

@@ -2,7 +2,7 @@
 
 Replaces monte_carlo_service.py:197-336: GBM paths, percentiles, VaR/CVaR,
 probability of profit, per-path max drawdown. The CPU reference
-reimplements the exact Philox4x32-10 + Box-Muller stream of the kernel in
+reimplements the exact Philox4x32-7 + Box-Muller stream of the kernel in
 vectorized numpy so small-scale tests are bit-comparable (modulo
 transcendental rounding), and an analytic-moment test covers the large-N
 statistics.
@@ -14,8 +14,12 @@ import numpy as np
 
 from . import require_hip_ops
 
-# --- Philox4x32-10 in numpy (matches ops/hip/common.hpp) -------------------
+# --- Philox4x32-7 in numpy (matches ops/hip/common.hpp) --------------------
+# 7 rounds = the BigCrush-passing count from the Random123 paper; must stay
+# equal to PHILOX_ROUNDS in ops/hip/common.hpp (bit-reproducibility tests
+# compare the two streams directly).
 
+_PHILOX_ROUNDS = 7
 _M0 = np.uint64(0xD2511F53)
 _M1 = np.uint64(0xCD9E8D57)
 _W0 = np.uint32(0x9E3779B9)
@@ -23,7 +27,7 @@ _W1 = np.uint32(0xBB67AE85)
 
 
 def philox4x32_np(seed: int, ctr_lo: np.ndarray, ctr_hi: np.ndarray):
-    """Vectorized Philox4x32-10. ctr_lo/ctr_hi: uint64 arrays."""
+    """Vectorized Philox4x32-7. ctr_lo/ctr_hi: uint64 arrays."""
     u32 = np.uint32
     u64 = np.uint64
     k0 = u32(seed & 0xFFFFFFFF)
@@ -36,7 +40,7 @@ def philox4x32_np(seed: int, ctr_lo: np.ndarray, ctr_hi: np.ndarray):
     c3 = (ctr_hi >> u64(32)).astype(u32)
     k0 = np.full_like(c0, k0)
     k1 = np.full_like(c0, k1)
-    for _ in range(10):
+    for _ in range(_PHILOX_ROUNDS):
         p0 = _M0 * c0.astype(u64)
         p1 = _M1 * c2.astype(u64)
         hi0 = (p0 >> u64(32)).astype(u32)
@@ -74,6 +78,28 @@ def philox_normal4_np(seed: int, ctr_lo: np.ndarray, ctr_hi: np.ndarray):
     return np.stack([z0, z1, z2, z3], axis=-1).astype(np.float32)
 
 
+# --- shared GBM term builder (CPU reference AND GPU wrapper) ---------------
+
+_LOG2E = np.log2(np.e)     # f64; kernels work in base-2 log space
+
+
+def _gbm_terms(chol, mu, sigma, dt):
+    """(cvol_k_major, drift, both f32 and pre-scaled by log2(e)).
+
+    The kernels accumulate logS in BASE-2 so the per-asset exponential is
+    a single native v_exp_f32 (exp2) — the log2(e) factor is folded into
+    drift and cvol here, identically for the CPU reference and the GPU
+    wrapper so the two streams stay comparable."""
+    f32 = np.float32
+    chol = np.asarray(chol)
+    mu = np.asarray(mu)
+    sigma = np.asarray(sigma)
+    cvol = sigma[:, None] * chol * np.sqrt(dt) * _LOG2E          # (a, k)
+    cvol_k_major = np.ascontiguousarray(cvol.T, dtype=f32)       # (k, a)
+    drift = ((mu - 0.5 * sigma**2) * dt * _LOG2E).astype(f32)
+    return cvol_k_major, drift
+
+
 # --- CPU reference path generator -----------------------------------------
 
 def mc_paths_cpu(
@@ -92,9 +118,7 @@ def mc_paths_cpu(
     order. Returns (final_value, max_dd) each (n_paths,) f32."""
     A = chol.shape[0]
     f32 = np.float32
-    cvol = (sigma[:, None] * chol * np.sqrt(dt)).astype(f32)   # (a, k)
-    cvol_k_major = np.ascontiguousarray(cvol.T)                # (k, a)
-    drift = ((mu - 0.5 * sigma**2) * dt).astype(f32)
+    cvol_k_major, drift = _gbm_terms(chol, mu, sigma, dt)      # base-2 log
     wS0 = (weights * s0).astype(f32)
 
     logS = np.zeros((n_paths, A), f32)
@@ -114,7 +138,7 @@ def mc_paths_cpu(
                 k = k4 * 4 + dz
                 logS += np.outer(z4[:, dz], cvol_k_major[k])
         logS += drift
-        V = (wS0 * np.exp(logS)).sum(axis=1).astype(f32)
+        V = (wS0 * np.exp2(logS)).sum(axis=1).astype(f32)
         vmax = np.maximum(vmax, V)
         mdd = np.maximum(mdd, (vmax - V) / vmax)
     return V, mdd
@@ -141,9 +165,7 @@ def mc_paths_gpu(
         use_mfma = False
     pad = ((-n_paths) % 256) if use_mfma else 0   # MFMA tile = 256 paths
     f32 = np.float32
-    cvol = (np.asarray(sigma)[:, None] * np.asarray(chol) * np.sqrt(dt))
-    cvol_k_major = np.ascontiguousarray(cvol.T, dtype=f32)
-    drift = ((np.asarray(mu) - 0.5 * np.asarray(sigma) ** 2) * dt).astype(f32)
+    cvol_k_major, drift = _gbm_terms(chol, mu, sigma, dt)      # base-2 log
     wS0 = (np.asarray(weights) * s0).astype(f32)
     v0 = float(wS0.sum())
 
