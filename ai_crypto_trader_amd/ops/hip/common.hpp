@@ -64,12 +64,15 @@ DEV_INLINE float u32_to_unit(uint32_t v) {
 //    the uniform u2 in (0,1] is EXACTLY the hardware domain: no 2*pi
 //    multiply, no range reduction (__sincosf emits a 1/2pi mul + guards).
 //  - -2*ln(u) = -2ln2 * log2(u): one fused constant on the raw v_log_f32.
+//  - raw v_sqrt_f32 (1 ulp): libm sqrtf emits a ~10-instr scale/Newton
+//    adjust sequence per call for correct rounding we don't need here.
 // The numpy twin (ops/montecarlo.py) keeps sin(2*pi*u)/cos/log — identical
 // math, transcendental rounding differs by ulps (tests use tolerances).
 DEV_INLINE float2 box_muller(uint32_t a, uint32_t b) {
     float u1 = u32_to_unit(a);
     float u2 = u32_to_unit(b);
-    float r = sqrtf(-1.3862943611f * __builtin_amdgcn_logf(u1));
+    float r = __builtin_amdgcn_sqrtf(
+        -1.3862943611f * __builtin_amdgcn_logf(u1));
     float s = __builtin_amdgcn_sinf(u2);
     float c = __builtin_amdgcn_cosf(u2);
     return make_float2(r * c, r * s);
