@@ -116,7 +116,7 @@ __global__ void __launch_bounds__(256) attn_fwd_kernel(
 #pragma unroll
         for (int off = 1; off < 16; off <<= 1)
             sum += __shfl_xor(sum, off, 64);
-        const float inv = 1.0f / sum;
+        const float inv = __builtin_amdgcn_rcpf(sum);   // 1-ulp; P is bf16 anyway
 #pragma unroll
         for (int ct = 0; ct < 4; ++ct) {
             const int row = 16 * w + fq * 4 + rr;

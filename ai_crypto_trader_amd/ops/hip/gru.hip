@@ -23,9 +23,12 @@ typedef float f32x4 __attribute__((ext_vector_type(4)));
 
 namespace {
 
-DEV_INLINE float gsigmoid(float x) { return 1.0f / (1.0f + __expf(-x)); }
+// 1-ulp v_rcp instead of IEEE div (see lstm.hip rationale)
+DEV_INLINE float gsigmoid(float x) {
+    return __builtin_amdgcn_rcpf(1.0f + __expf(-x));
+}
 DEV_INLINE float gtanh(float x) {
-    return 2.0f / (1.0f + __expf(-2.0f * x)) - 1.0f;
+    return 2.0f * __builtin_amdgcn_rcpf(1.0f + __expf(-2.0f * x)) - 1.0f;
 }
 
 template <int H>

@@ -40,10 +40,15 @@ typedef float f32x4 __attribute__((ext_vector_type(4)));
 
 namespace {
 
-DEV_INLINE float sigmoidf_(float x) { return 1.0f / (1.0f + __expf(-x)); }
+// 1-ulp v_rcp instead of IEEE div: the gate math is tolerance-tested
+// against PyTorch fp32 (not bitwise), and libm-grade division costs
+// ~10 instrs per gate element per timestep.
+DEV_INLINE float sigmoidf_(float x) {
+    return __builtin_amdgcn_rcpf(1.0f + __expf(-x));
+}
 DEV_INLINE float tanhf_(float x) {
-    // tanh(x) = 2*sigmoid(2x) - 1
-    return 2.0f / (1.0f + __expf(-2.0f * x)) - 1.0f;
+    // tanh(x) = 2*sigmoid(2x) - 1, rcp as above
+    return 2.0f * __builtin_amdgcn_rcpf(1.0f + __expf(-2.0f * x)) - 1.0f;
 }
 
 template <int H>
