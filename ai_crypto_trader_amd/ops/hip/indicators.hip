@@ -77,15 +77,18 @@ __global__ void __launch_bounds__(IND_BLOCK) indicators_kernel(
 
         const float gain = fmaxf(change, 0.0f);
         const float lossv = fmaxf(-change, 0.0f);
-        avg_gain += (gain - avg_gain) / 14.0f;
-        avg_loss += (lossv - avg_loss) / 14.0f;
-        const float rsi =
-            100.0f - 100.0f / (1.0f + avg_gain / fmaxf(avg_loss, 1e-9f));
+        // reciprocal-multiply + v_rcp instead of IEEE div throughout this
+        // kernel: the CPU reference divides, the GPU-vs-CPU test runs at
+        // rtol 1e-3 (chunk-warmup reconvergence noise dominates 1-ulp rcp)
+        avg_gain += (gain - avg_gain) * (1.0f / 14.0f);
+        avg_loss += (lossv - avg_loss) * (1.0f / 14.0f);
+        const float rsi = 100.0f * avg_gain *
+            __builtin_amdgcn_rcpf(avg_gain + fmaxf(avg_loss, 1e-9f));
 
         const float tr = fmaxf(high - low,
                                fmaxf(fabsf(high - prev_close),
                                      fabsf(low - prev_close)));
-        atr += (tr - atr) / 14.0f;
+        atr += (tr - atr) * (1.0f / 14.0f);
         prev_close = close;
 
         // Bollinger ring (window 20, f64 rolling sum/sumsq)
@@ -97,11 +100,14 @@ __global__ void __launch_bounds__(IND_BLOCK) indicators_kernel(
             bb_sum2 += c64 * c64 - old * old;
             r_close[ri] = close;
         }
-        const double bcnt = (double)min(steps + 1, RING_W);
-        const double mean64 = bb_sum / bcnt;
-        const double var64 = fmax(bb_sum2 / bcnt - mean64 * mean64, 0.0);
+        // steps is the (wave-uniform) loop counter: the f64 divide only
+        // executes during the first RING_W-1 iterations of each chunk
+        double ibcnt = 1.0 / (double)RING_W;
+        if (steps + 1 < RING_W) ibcnt = 1.0 / (steps + 1.0);
+        const double mean64 = bb_sum * ibcnt;
+        const double var64 = fmax(bb_sum2 * ibcnt - mean64 * mean64, 0.0);
         const float mean = (float)mean64;
-        const float sd = sqrtf((float)var64);
+        const float sd = __builtin_amdgcn_sqrtf((float)var64);
 
         // Stoch/Williams rings (window 14): scan for min/max
         {
@@ -115,9 +121,9 @@ __global__ void __launch_bounds__(IND_BLOCK) indicators_kernel(
             hmax = fmaxf(hmax, r_high[i]);
             lmin = fminf(lmin, r_low[i]);
         }
-        const float rng = fmaxf(hmax - lmin, 1e-9f);
-        const float stoch_k = (close - lmin) / rng * 100.0f;
-        const float williams = -100.0f * (hmax - close) / rng;
+        const float irng = __builtin_amdgcn_rcpf(fmaxf(hmax - lmin, 1e-9f));
+        const float stoch_k = (close - lmin) * irng * 100.0f;
+        const float williams = -100.0f * (hmax - close) * irng;
 
         // VWAP ring (window 20, typical price)
         {
@@ -129,7 +135,7 @@ __global__ void __launch_bounds__(IND_BLOCK) indicators_kernel(
             r_pv[ri] = pv;
             r_vol[ri] = vol;
         }
-        const float vwap = pv_sum / fmaxf(vol_sum, 1e-9f);
+        const float vwap = pv_sum * __builtin_amdgcn_rcpf(fmaxf(vol_sum, 1e-9f));
 
         if (t >= t_out0) {
             float* o = out + ((long)sym * T + t) * IND_NIND;
