@@ -150,14 +150,65 @@ class AppConfig(BaseModel):
                     path = cand
                     break
         if path is None:
-            return cls()
-        with open(path) as f:
-            return cls.model_validate(json.load(f))
+            cfg = cls()
+        else:
+            with open(path) as f:
+                cfg = cls.model_validate(json.load(f))
+        return apply_env_overrides(cfg)
 
     def save(self, path: str | Path):
         Path(path).parent.mkdir(parents=True, exist_ok=True)
         with open(path, "w") as f:
             f.write(self.model_dump_json(indent=2))
+
+
+# Env-var flag layer (reference parity: per-service env flags, e.g.
+# EVOLUTION_METHOD / GA_POPULATION_SIZE strategy_evolution_service.py:75-79,
+# selection weights strategy_selection_service.py:71-78, plus .env secrets).
+# Two forms:
+#   ACT_<section>__<field>=value  — generic override for any config field
+#   legacy aliases below          — the reference's documented flag names
+_ENV_ALIASES = {
+    "EVOLUTION_METHOD": ("evolution", "method"),
+    "GA_POPULATION_SIZE": ("evolution", "population_size"),
+    "GA_GENERATIONS": ("evolution", "generations"),
+    "MIN_CONFIDENCE": ("trading", "min_confidence"),
+    "TRADING_SYMBOLS": ("trading", "symbols"),
+    "MC_NUM_SIMULATIONS": ("monte_carlo", "num_simulations"),
+    "REGIME_METHOD": ("regime", "method"),
+    "SEED": (None, "seed"),
+}
+
+
+def _coerce(cur, raw: str):
+    if isinstance(cur, bool):
+        return raw.lower() in ("1", "true", "yes", "on")
+    if isinstance(cur, int):
+        return int(raw)
+    if isinstance(cur, float):
+        return float(raw)
+    if isinstance(cur, list):
+        return [s.strip() for s in raw.split(",") if s.strip()]
+    return raw
+
+
+def apply_env_overrides(cfg: "AppConfig") -> "AppConfig":
+    import os
+
+    def set_field(section, field, raw):
+        target = cfg if section is None else getattr(cfg, section, None)
+        if target is None or not hasattr(target, field):
+            return
+        setattr(target, field, _coerce(getattr(target, field), raw))
+
+    for name, (section, field) in _ENV_ALIASES.items():
+        if name in os.environ:
+            set_field(section, field, os.environ[name])
+    for name, raw in os.environ.items():
+        if name.startswith("ACT_") and "__" in name:
+            section, _, field = name[4:].lower().partition("__")
+            set_field(section, field, raw)
+    return cfg
 
 
 _global: AppConfig | None = None

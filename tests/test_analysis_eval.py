@@ -188,3 +188,20 @@ def test_checkpoint_roundtrip(tmp_path):
         assert (await bus2.get_json("holdings"))["total_value"] == 5.0
 
     asyncio.run(bus_roundtrip())
+
+
+def test_env_overrides(monkeypatch, tmp_path):
+    """Env-var flag layer (reference: EVOLUTION_METHOD / GA_POPULATION_SIZE
+    env flags + generic ACT_section__field form)."""
+    from ai_crypto_trader_amd.config import AppConfig
+
+    monkeypatch.setenv("EVOLUTION_METHOD", "ga")
+    monkeypatch.setenv("GA_POPULATION_SIZE", "64")
+    monkeypatch.setenv("TRADING_SYMBOLS", "BTCUSDC, ETHUSDC")
+    monkeypatch.setenv("ACT_risk__max_portfolio_var_pct", "0.123")
+    monkeypatch.setenv("ACT_nosuch__field", "ignored")   # unknown: no crash
+    cfg = AppConfig.load()
+    assert cfg.evolution.method == "ga"
+    assert cfg.evolution.population_size == 64
+    assert cfg.trading.symbols == ["BTCUSDC", "ETHUSDC"]
+    assert cfg.risk.max_portfolio_var_pct == 0.123
