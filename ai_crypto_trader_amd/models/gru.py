@@ -49,17 +49,18 @@ class _FusedGRUSeq(torch.autograd.Function):
             h_out.data_ptr(), w_bf.contiguous().data_ptr(),
             dgates_x.data_ptr(), B, T, H, stream,
         )
-        # h-side dgates: the n-column is da_n * r (gru.hip header)
-        dg_h = dgates_x.float()
-        r_gate = gates[..., :H].float()
-        dg_h[..., 2 * H:] = dg_h[..., 2 * H:] * r_gate
+        # h-side dgates: the n-column is da_n * r (gru.hip header).
+        # All bf16: the weight-grad GEMM accumulates f32 inside hipBLASLt
+        # (autocast semantics), no f32 materialization of dgates
+        dg_h = dgates_x.clone()
+        dg_h[..., 2 * H:] = dg_h[..., 2 * H:] * gates[..., :H]
         h_prev = torch.cat(
             [torch.zeros((1, B, H), dtype=h_out.dtype, device=dev),
              h_out[:-1]], dim=0,
         )
-        dw_hh = h_prev.reshape(T * B, H).float().t() @ \
-            dg_h.reshape(T * B, 3 * H)
-        db_hh = dg_h.reshape(T * B, 3 * H).sum(dim=0)
+        dg = dg_h.reshape(T * B, 3 * H)
+        dw_hh = (h_prev.reshape(T * B, H).t() @ dg).float()
+        db_hh = dg.sum(dim=0, dtype=torch.float32)
         return dgates_x, dw_hh, db_hh
 
 
@@ -86,9 +87,9 @@ class FusedGRULayer(nn.Module):
             pad = (-B) % 64
             if pad:
                 x = torch.cat([x, x.new_zeros(T, pad, x.shape[2])], dim=1)
-            xproj = (x.to(self.w_ih.dtype) @ self.w_ih + self.b_ih)
-            h = _FusedGRUSeq.apply(
-                xproj.to(torch.bfloat16), self.w_hh, self.b_hh)
+            bf = torch.bfloat16    # f32 master weights, bf16 GEMMs
+            xproj = (x.to(bf) @ self.w_ih.to(bf) + self.b_ih.to(bf))
+            h = _FusedGRUSeq.apply(xproj, self.w_hh, self.b_hh)
             return h[:, :B] if pad else h
         return self._forward_reference(x)
 

@@ -65,14 +65,16 @@ class _FusedLSTMSeq(torch.autograd.Function):
             dh_up.data_ptr(), gates.data_ptr(), c_sav.data_ptr(),
             w_bf.contiguous().data_ptr(), dgates.data_ptr(), B, T, H, stream,
         )
-        # dW_hh = sum_t h_{t-1}^T dgates_t  — one hipBLASLt GEMM
+        # dW_hh = sum_t h_{t-1}^T dgates_t — one hipBLASLt bf16 GEMM
+        # (f32 accumulation inside, bf16 grad out = torch.autocast
+        # semantics; the f32 materialization of dgates was 1 GB/step)
         h_prev = torch.cat(
             [torch.zeros((1, B, H), dtype=h_out.dtype, device=dev),
              h_out[:-1]], dim=0,
         )
-        dg_f = dgates.reshape(T * B, 4 * H).float()
-        dw_hh = h_prev.reshape(T * B, H).float().t() @ dg_f
-        db_hh = dg_f.sum(dim=0)
+        dg = dgates.reshape(T * B, 4 * H)
+        dw_hh = (h_prev.reshape(T * B, H).t() @ dg).float()
+        db_hh = dg.sum(dim=0, dtype=torch.float32)
         return dgates, dw_hh, db_hh
 
 
@@ -120,14 +122,14 @@ class FusedLSTMLayer(nn.Module):
             pad = (-B) % 64              # kernel batch tile is 64 rows
             if pad:
                 x = torch.cat([x, x.new_zeros(T, pad, x.shape[2])], dim=1)
-            xproj = (x.to(self.w_ih.dtype) @ self.w_ih + self.b_ih)
+            # bf16 input projection with f32 master weights (autocast
+            # style): halves xproj traffic and makes dW_ih a bf16 GEMM
+            bf = torch.bfloat16
+            xproj = (x.to(bf) @ self.w_ih.to(bf) + self.b_ih.to(bf))
             if torch.is_grad_enabled():
-                h = _FusedLSTMSeq.apply(
-                    xproj.to(torch.bfloat16), self.w_hh, self.b_hh
-                )
+                h = _FusedLSTMSeq.apply(xproj, self.w_hh, self.b_hh)
             else:   # serving path: no backward saves
-                h = lstm_seq_infer(
-                    xproj.to(torch.bfloat16), self.w_hh, self.b_hh)
+                h = lstm_seq_infer(xproj, self.w_hh, self.b_hh)
             return h[:, :B] if pad else h
         return self._forward_reference(x)
 
