@@ -402,3 +402,21 @@ def test_strategy_evolution_service_cpu():
     svc.device = "cuda:0"
     assert svc.select_method("ranging", 0.2) == "ppo"
     assert svc.select_method("volatile", 0.9) == "ga"
+
+
+@pytest.mark.parametrize("cmd", [
+    ["run_trader.py", "--minutes", "0.25", "--candles", "3000",
+     "--symbols", "BTCUSDC,ETHUSDC", "--status-interval", "2"],
+    ["auto_trader.py", "--minutes", "0.15", "--candles", "3000"],
+    ["run_ai_model_services.py", "--model-registry", "--explainability",
+     "--seconds", "5"],
+])
+def test_entrypoint_smoke(cmd):
+    """The reference's three long-running entrypoints start, run a short
+    window over the synthetic feed, and exit 0 (SURVEY §2.1 parity)."""
+    import subprocess
+    import sys
+
+    r = subprocess.run([sys.executable] + cmd, capture_output=True,
+                       text=True, timeout=300)
+    assert r.returncode == 0, (r.stdout[-1500:], r.stderr[-1500:])
