@@ -19,7 +19,7 @@ from typing import Callable
 
 import numpy as np
 
-from ..ops.indicators import IND_NAMES, indicators_cpu
+from ..ops.indicators import IND_NAMES, indicators_fast
 from .engine import ANNUAL_CANDLES
 
 DecisionFn = Callable[[dict], str]     # context -> "BUY" | "SELL" | "HOLD"
@@ -65,7 +65,7 @@ class StrategyTester:
         cfg = self.cfg
         candles = np.asarray(candles, np.float32)
         T = len(candles)
-        ind = indicators_cpu(candles[None])[0]          # (T, 13)
+        ind = indicators_fast(candles[None])[0]         # (T, 13)
 
         cash, units = 1.0, 0.0
         entry_cost = 0.0

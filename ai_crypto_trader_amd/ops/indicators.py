@@ -122,6 +122,80 @@ def indicators_cpu(candles: np.ndarray) -> np.ndarray:
     return out
 
 
+def indicators_fast(candles: np.ndarray) -> np.ndarray:
+    """Vectorized twin of indicators_cpu for the serving/feature path.
+
+    Same formulas, evaluated with scipy lfilter (EMA-family recurrences),
+    f64 cumsum differences (Bollinger/VWAP rolling sums) and sliding-window
+    max/min (stoch/williams) instead of the per-candle python loop — ~1000x
+    faster for single-symbol requests. Differs from indicators_cpu only by
+    f32-vs-f64 rounding of the recurrences (tested at rtol 1e-4);
+    indicators_cpu remains the bit-semantics oracle for the HIP kernel."""
+    from scipy.signal import lfilter
+
+    f32 = np.float32
+    candles = np.asarray(candles, f32)
+    nsym, T, _ = candles.shape
+    W, S = 20, 14
+    c = candles[:, :, 0].astype(np.float64)
+    h = candles[:, :, 1].astype(np.float64)
+    lo = candles[:, :, 2].astype(np.float64)
+    v = candles[:, :, 3].astype(np.float64)
+
+    def ema(x, alpha, y0=None):
+        # y[t] = y[t-1] + alpha (x[t] - y[t-1]); y0 defaults to alpha*x[0]
+        zi = np.zeros((nsym, 1)) if y0 is None else ((1 - alpha) * y0)[:, None]
+        y, _ = lfilter([alpha], [1, -(1 - alpha)], x, axis=1, zi=zi)
+        return y
+
+    ema12 = ema(c, 2.0 / 13.0, y0=c[:, 0])
+    ema26 = ema(c, 2.0 / 27.0, y0=c[:, 0])
+    macd = ema12 - ema26
+    sig = ema(macd, 2.0 / 10.0)
+    change = np.diff(c, axis=1, prepend=c[:, :1])
+    gain = np.maximum(change, 0.0)
+    loss = np.maximum(-change, 0.0)
+    avg_gain = ema(gain, 1.0 / 14.0)
+    avg_loss = ema(loss, 1.0 / 14.0)
+    rsi = 100.0 - 100.0 / (1.0 + avg_gain / np.maximum(avg_loss, 1e-9))
+    pc = np.concatenate([c[:, :1], c[:, :-1]], axis=1)
+    tr = np.maximum(h - lo, np.maximum(np.abs(h - pc), np.abs(lo - pc)))
+    atr = ema(tr, 1.0 / 14.0)
+
+    def roll_sum(x, win):
+        cs = np.cumsum(x, axis=1)
+        out = cs.copy()
+        out[:, win:] = cs[:, win:] - cs[:, :-win]
+        return out
+
+    cnt = np.minimum(np.arange(T) + 1, W).astype(np.float64)
+    mean = roll_sum(c, W) / cnt
+    var = np.maximum(roll_sum(c * c, W) / cnt - mean * mean, 0.0)
+    sd = np.sqrt(var)
+
+    def roll_max(x, win, fill):
+        pad = np.full((nsym, win - 1), fill)
+        xw = np.lib.stride_tricks.sliding_window_view(
+            np.concatenate([pad, x], axis=1), win, axis=1)
+        return xw.max(axis=-1) if fill < 0 else xw.min(axis=-1)
+
+    hmax = roll_max(h, S, -np.inf)
+    lmin = roll_max(lo, S, np.inf)
+    rng = np.maximum(hmax - lmin, 1e-9)
+    stoch = (c - lmin) / rng * 100.0
+    williams = -100.0 * (hmax - c) / rng
+
+    pv = (h + lo + c) * (1.0 / 3.0) * v
+    vwap = roll_sum(pv, W) / np.maximum(roll_sum(v, W), 1e-9)
+
+    out = np.stack([
+        ema12, ema26, macd, sig, macd - sig, rsi,
+        mean, mean + 2.0 * sd, mean - 2.0 * sd,
+        atr, stoch, williams, vwap,
+    ], axis=-1)
+    return out.astype(f32)
+
+
 def indicators_gpu(candles) -> "torch.Tensor":
     """(nsym, T, 4) f32 cuda -> (nsym, T, NIND) f32 cuda."""
     import torch

@@ -27,9 +27,9 @@ def build_features(candles: np.ndarray) -> np.ndarray:
     """(T, 4) [close, high, low, vol] -> (T, 9) feature matrix: the
     reference's 9-feature set (close/high/low/volume + rsi + macd + bb
     position + returns; neural_network_service.py:530-586)."""
-    from ..ops.indicators import indicators_cpu
+    from ..ops.indicators import indicators_fast
 
-    ind = indicators_cpu(candles[None])[0]     # (T, 13)
+    ind = indicators_fast(candles[None])[0]    # (T, 13), vectorized path
     close = candles[:, 0]
     ret1 = np.zeros_like(close)
     ret1[1:] = close[1:] / close[:-1] - 1.0
@@ -175,6 +175,11 @@ class NeuralNetworkService(Service):
         if model is None or scaler is None:
             return None
         cfg = self.config.neural_network
+        # recurrent indicators (EMA/RSI/ATR) reconverge within 512 candles
+        # (same argument as indicators.hip's chunk warmup: the forgetting
+        # factor underflows f32), so serving only ever needs the tail —
+        # not the whole history per request
+        candles = candles[-(cfg.seq_len + 512):]
         feats = build_features(candles)[-cfg.seq_len:]
         if len(feats) < cfg.seq_len:
             return None

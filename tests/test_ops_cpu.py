@@ -170,3 +170,19 @@ def test_ga_improves_on_simple_objective():
     bestN = fit(pop).max()
     assert bestN > best0
     assert bestN > -0.05
+
+
+def test_indicators_fast_matches_golden():
+    """Vectorized serving-path indicators == the per-candle golden loop
+    (same formulas; only f32-vs-f64 recurrence rounding differs)."""
+    from ai_crypto_trader_amd.data.synthetic import (
+        candles_chl_v, generate_ohlcv,
+    )
+    from ai_crypto_trader_amd.ops.indicators import (
+        indicators_cpu, indicators_fast,
+    )
+
+    c = candles_chl_v(generate_ohlcv(2500, 3, seed=11, sigma=2.0))
+    ref = indicators_cpu(c)
+    fast = indicators_fast(c)
+    np.testing.assert_allclose(fast, ref, rtol=1e-4, atol=1e-3)

@@ -159,6 +159,19 @@ def bench_env(reps):
             "config": {"n_envs": 4096}}
 
 
+def bench_attn(reps):
+    from ai_crypto_trader_amd.models.attention import attn_fwd_hip
+
+    BH, S, D = 16384, 60, 16        # transformer predictor shape (4096x4)
+    q = torch.randn(BH, S, D, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn_like(q)
+    v = torch.randn_like(q)
+    dt = timed(lambda: attn_fwd_hip(q, k, v, D ** -0.5), reps)
+    return {"kernel": "attn_fwd", "ms": dt * 1e3,
+            "seqs_per_sec": BH / dt,
+            "config": {"BH": BH, "S": S, "D": D}}
+
+
 def bench_gae(reps):
     from ai_crypto_trader_amd.models.rl import gae_gpu
 
@@ -174,7 +187,7 @@ def bench_gae(reps):
 BENCHES = {
     "backtest": bench_backtest, "mc": bench_mc, "cov": bench_cov,
     "indicators": bench_indicators, "lstm": bench_lstm, "env": bench_env,
-    "gae": bench_gae,
+    "gae": bench_gae, "attn": bench_attn,
 }
 
 
