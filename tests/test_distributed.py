@@ -143,3 +143,50 @@ def test_mc_sharded_matches_single():
         assert abs(s["std"] - ref["std"]) < 1e-5
         assert abs(s["var_95"] - ref["var_95"]) < 2e-3   # histogram binning
     assert results[0] == results[1]
+
+
+def _ppo_worker(rank, world, port, out_q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from ai_crypto_trader_amd.models.rl import N_OBS, PPOAgent
+
+        torch.manual_seed(100 + rank)        # DIFFERENT data per rank
+        agent = PPOAgent("cpu", seed=0)      # same init on both ranks
+        T, E = 8, 16
+        obs_b = torch.randn(T, E, N_OBS)     # rank-local synthetic batch
+        act_b = torch.randint(0, 3, (T, E))
+        logp_b = -torch.rand(T, E)
+        rew_b = torch.randn(T, E)
+        done_b = torch.zeros(T, E)
+        val_b = torch.randn(T + 1, E)
+        agent.update(obs_b, act_b, logp_b, rew_b, done_b, val_b)
+        params = torch.cat(
+            [p.detach().reshape(-1) for p in agent.net.parameters()])
+        out_q.put((rank, params.numpy().copy()))
+    finally:
+        dist.destroy_process_group()
+
+
+def test_ppo_dp_grad_allreduce_world2():
+    """PPO DP: with identical init but DIFFERENT per-rank rollouts, the
+    gradient all-reduce keeps parameters bitwise-identical across ranks
+    after the update (the BASELINE #4 data-parallel contract)."""
+    world = 2
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [
+        ctx.Process(target=_ppo_worker, args=(r, world, 29881, q))
+        for r in range(world)
+    ]
+    for p in procs:
+        p.start()
+    out = {}
+    for _ in range(world):
+        rank, params = q.get()
+        out[rank] = params
+    for p in procs:
+        p.join(60)
+        assert p.exitcode == 0
+    np.testing.assert_array_equal(out[0], out[1])
