@@ -147,15 +147,17 @@ extern "C" void launch_cov(const float* X, float* cov, int T, int N,
     float* Sx = nullptr;
     if (hipMallocAsync((void**)&Sx, N * sizeof(float), stream) != hipSuccess)
         throw std::runtime_error("cov: workspace alloc failed");
-    hipMemsetAsync(cov, 0, (size_t)N * N * sizeof(float), stream);
-    hipMemsetAsync(Sx, 0, N * sizeof(float), stream);
+    if (hipMemsetAsync(cov, 0, (size_t)N * N * sizeof(float), stream)
+            != hipSuccess ||
+        hipMemsetAsync(Sx, 0, N * sizeof(float), stream) != hipSuccess)
+        throw std::runtime_error("cov: workspace memset failed");
     int ntiles = (T + COV_TILE_T - 1) / COV_TILE_T;
     int grid = ntiles < 512 ? ntiles : 512;
     hipLaunchKernelGGL(cov_accum_kernel, dim3(grid), dim3(256), 0, stream, X,
                        cov, Sx, T, N);
     hipLaunchKernelGGL(cov_finalize_kernel, dim3((N * N + 255) / 256),
                        dim3(256), 0, stream, cov, Sx, T, N);
-    hipFreeAsync(Sx, stream);
+    (void)hipFreeAsync(Sx, stream);
 }
 
 extern "C" void launch_mfma_gemm_test(const void* A, const void* B, float* C,
