@@ -156,3 +156,36 @@ def test_ppo_evolution_optimizer_gpu():
     lo = np.array([b[0] for b in PARAM_BOUNDS], np.float32)
     hi = np.array([b[1] for b in PARAM_BOUNDS], np.float32)
     assert np.all(params >= lo - 1e-6) and np.all(params <= hi + 1e-6)
+
+
+def test_regime_detectors_on_gpu():
+    """KMeans/GMM/HMM torch detectors run on cuda through the service's
+    detect() path and return sane labels."""
+    from ai_crypto_trader_amd.bus.message_bus import InProcessBus
+    from ai_crypto_trader_amd.config import AppConfig
+    from ai_crypto_trader_amd.services.market_regime import (
+        MarketRegimeService,
+    )
+
+    closes = np.cumprod(
+        1 + 0.003 * np.random.default_rng(2).standard_normal(900)) * 50
+    for method in ("kmeans", "gmm", "hmm"):
+        cfg = AppConfig()
+        cfg.regime.method = method
+        svc = MarketRegimeService(InProcessBus(), cfg, device="cuda:0")
+        regime, conf = svc.detect(closes)
+        assert regime in ("bull", "bear", "ranging", "volatile"), method
+        assert 0.0 <= conf <= 1.0
+
+
+def test_pattern_model_trains_on_gpu():
+    from ai_crypto_trader_amd.models.patterns import (
+        PatternRecognitionModel, generate_pattern,
+    )
+
+    m = PatternRecognitionModel("cuda:0", seed=0)
+    acc = m.train(epochs=6, n_per_class=48, seed=1)
+    assert acc > 0.8, acc
+    rng = np.random.default_rng(7)
+    out = m.detect(generate_pattern("double_top", rng))
+    assert out["confidence"] > 0.2 and out["pattern"] != "none"
