@@ -140,6 +140,68 @@ class InProcessBus:
         self._subs.clear()
 
 
+class RedisSubscription:
+    """Subscription adapter over redis pubsub with the Subscription
+    interface (get / get_batch / aiter / close). A lazy reader task pumps
+    the wire into a local queue so get_batch() can drain bursts exactly
+    like the in-process bus."""
+
+    def __init__(self, r, channels: tuple[str, ...]):
+        self.r = r
+        self.channels = channels
+        self.queue: asyncio.Queue = asyncio.Queue(maxsize=10_000)
+        self._pubsub = None
+        self._task: asyncio.Task | None = None
+
+    async def _ensure(self):
+        if self._task is None:
+            self._pubsub = self.r.pubsub()
+            pats = [c for c in self.channels if any(x in c for x in "*?[")]
+            plain = [c for c in self.channels if c not in pats]
+            if plain:
+                await self._pubsub.subscribe(*plain)
+            if pats:
+                await self._pubsub.psubscribe(*pats)
+            self._task = asyncio.create_task(self._pump())
+
+    async def _pump(self):
+        async for msg in self._pubsub.listen():
+            if msg["type"] not in ("message", "pmessage"):
+                continue
+            data = msg["data"]
+            try:
+                data = json.loads(data)
+            except (json.JSONDecodeError, TypeError):
+                pass
+            try:
+                self.queue.put_nowait((msg["channel"], data))
+            except asyncio.QueueFull:
+                pass                      # slow consumer: drop
+
+    async def __aiter__(self) -> AsyncIterator[tuple[str, Any]]:
+        await self._ensure()
+        while True:
+            yield await self.queue.get()
+
+    async def get(self, timeout: float | None = None):
+        await self._ensure()
+        if timeout is None:
+            return await self.queue.get()
+        return await asyncio.wait_for(self.queue.get(), timeout)
+
+    async def get_batch(self, max_items: int = 512) -> list:
+        await self._ensure()
+        out = [await self.queue.get()]
+        while len(out) < max_items and not self.queue.empty():
+            out.append(self.queue.get_nowait())
+        return out
+
+    def close(self):
+        if self._task is not None:
+            self._task.cancel()
+            self._task = None
+
+
 class RedisBus:
     """Thin async-redis adapter with the same interface (requires a redis
     server; wire-compatible with the reference's channel schema)."""
@@ -155,9 +217,10 @@ class RedisBus:
         return await self.r.publish(channel, payload)
 
     def subscribe(self, *channels):
-        raise NotImplementedError(
-            "use RedisBus.psubscribe_iter in a task; see services/base.py"
-        )
+        """Same interface as InProcessBus.subscribe: returns an object
+        with get/get_batch/close backed by a redis pubsub reader task
+        (glob patterns map to PSUBSCRIBE)."""
+        return RedisSubscription(self.r, channels)
 
     async def set(self, key, value, ex=None):
         if isinstance(value, (dict, list)):

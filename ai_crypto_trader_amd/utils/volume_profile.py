@@ -1,8 +1,8 @@
 """Volume-profile analytics (reference parity:
 services/utils/volume_profile_analyzer.py — price-binned volume histogram,
 POC + 70% value area, POC/VA signal rules, buy/sell delta, anomaly
-z-scores). numpy implementation; the histogram path moves to the GPU
-histogram kernel when candle counts warrant (ops/hip TODO)."""
+z-scores). numpy implementation plus the GPU histogram-kernel path
+(vp_hist_gpu -> ops/hip/histogram.hip) for bulk histories."""
 
 from __future__ import annotations
 

@@ -304,3 +304,53 @@ def test_regime_models_recover_structure():
         regime, conf = svc.detect(closes)
         assert regime in ("bull", "bear", "ranging", "volatile")
         assert 0.0 <= conf <= 1.0
+
+
+def test_redis_subscription_adapter():
+    """RedisSubscription pumps a (fake) redis pubsub into the same
+    get/get_batch interface the services consume."""
+    import json as _json
+
+    from ai_crypto_trader_amd.bus.message_bus import RedisSubscription
+
+    class FakePubSub:
+        def __init__(self):
+            self.q = asyncio.Queue()
+            self.subscribed = []
+            self.psubscribed = []
+
+        async def subscribe(self, *ch):
+            self.subscribed += list(ch)
+
+        async def psubscribe(self, *ch):
+            self.psubscribed += list(ch)
+
+        async def listen(self):
+            while True:
+                yield await self.q.get()
+
+    class FakeRedis:
+        def __init__(self):
+            self.ps = FakePubSub()
+
+        def pubsub(self):
+            return self.ps
+
+    async def go():
+        r = FakeRedis()
+        sub = RedisSubscription(r, ("market_updates", "nn_*"))
+        # feed the wire: one json message, one pattern hit, one noise frame
+        await r.ps.q.put({"type": "subscribe", "channel": "x", "data": 1})
+        await r.ps.q.put({"type": "message", "channel": "market_updates",
+                          "data": _json.dumps({"symbol": "BTCUSDC"})})
+        await r.ps.q.put({"type": "pmessage", "channel": "nn_events",
+                          "data": "plain"})
+        ch, msg = await sub.get(timeout=2)
+        assert ch == "market_updates" and msg["symbol"] == "BTCUSDC"
+        batch = await sub.get_batch()
+        assert batch == [("nn_events", "plain")]
+        assert r.ps.subscribed == ["market_updates"]
+        assert r.ps.psubscribed == ["nn_*"]
+        sub.close()
+
+    asyncio.run(go())
