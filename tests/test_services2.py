@@ -420,3 +420,34 @@ def test_entrypoint_smoke(cmd):
     r = subprocess.run([sys.executable] + cmd, capture_output=True,
                        text=True, timeout=300)
     assert r.returncode == 0, (r.stdout[-1500:], r.stderr[-1500:])
+
+
+def test_service_runner_cli():
+    """Per-service runner (one-container-per-service topology): every
+    registered service constructs; one runs end-to-end via the CLI."""
+    import subprocess
+    import sys
+
+    from ai_crypto_trader_amd.services.runner import SERVICES
+
+    assert len(SERVICES) == 21
+    r = subprocess.run(
+        [sys.executable, "-m", "ai_crypto_trader_amd.services.runner",
+         "--service", "ai_analyzer", "--minutes", "0.02"],
+        capture_output=True, text=True, timeout=120)
+    assert r.returncode == 0, r.stderr[-800:]
+
+
+def test_service_runner_builds_all():
+    import argparse
+
+    from ai_crypto_trader_amd.bus.message_bus import InProcessBus
+    from ai_crypto_trader_amd.config import AppConfig
+    from ai_crypto_trader_amd.services.runner import SERVICES, build_service
+
+    args = argparse.Namespace(device="cpu", candles=600, speed=0.0, seed=0)
+    cfg = AppConfig()
+    cfg.trading.symbols = ["BTCUSDC", "ETHUSDC"]
+    for name in SERVICES:
+        svc = build_service(name, InProcessBus(), cfg, args)
+        assert svc is not None, name
