@@ -200,6 +200,31 @@ class NeuralNetworkService(Service):
             confidence=conf,
         ).to_dict()
 
+    FEATURE_NAMES = ["close", "high", "low", "volume", "rsi",
+                     "macd_hist", "bb_position", "return_1", "atr"]
+
+    def feature_importance(self, sym: str, candles: np.ndarray) -> dict | None:
+        """Gradient x input saliency per input feature — the offline
+        counterpart of the reference's SHAP DeepExplainer importances
+        (neural_network_service.py:957-1003): |d prediction / d feature *
+        feature| averaged over the window and batch, normalized to sum 1."""
+        model = self.models.get(sym)
+        scaler = self.scalers.get(sym)
+        if model is None or scaler is None:
+            return None
+        cfg = self.config.neural_network
+        feats = build_features(candles)
+        X, _ = make_windows(scaler.transform(feats).astype(np.float32),
+                            np.zeros(len(feats), np.float32), cfg.seq_len)
+        if X is None:
+            return None
+        x = torch.from_numpy(X[-64:]).to(self.device).requires_grad_(True)
+        model(x).sum().backward()
+        sal = (x.grad * x).abs().mean(dim=(0, 1))          # (F,)
+        sal = (sal / sal.sum().clamp_min(1e-12)).detach().cpu().numpy()
+        return {name: float(v)
+                for name, v in zip(self.FEATURE_NAMES, sal)}
+
     async def _train_loop(self):
         while self.running:
             for sym, h in list(self.candles.items()):
