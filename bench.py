@@ -138,7 +138,7 @@ def main():
             "metric": "backtested_1m_candles_per_sec_ga_pop_eval",
             "value": value,
             "unit": "candles/s",
-            "n_gpus": world if on_gpu else 0,
+            "n_gpus": world,
             "steps": args.steps,
             "warmup": args.warmup,
             "ms_per_step": ms_per_step,
