@@ -137,6 +137,25 @@ def bench_lstm(reps):
             "config": {"T": T, "B": B, "H": H}}
 
 
+def bench_gru(reps):
+    from ai_crypto_trader_amd.models.gru import FusedGRULayer
+
+    T, B, H = 60, 16384, 64
+    layer = FusedGRULayer(9, H).cuda()
+    x = torch.randn(T, B, 9, device="cuda")
+    dt_f = timed(lambda: layer(x), reps)
+
+    def fwdbwd():
+        layer.zero_grad()
+        layer(x).float().pow(2).mean().backward()
+
+    dt_b = timed(fwdbwd, reps)
+    return {"kernel": "gru_seq", "fwd_ms": dt_f * 1e3,
+            "fwdbwd_ms": dt_b * 1e3,
+            "cells_per_sec_fwd": T * B / dt_f,
+            "config": {"T": T, "B": B, "H": H}}
+
+
 def bench_env(reps):
     from ai_crypto_trader_amd.data.synthetic import (
         candles_chl_v, generate_ohlcv,
@@ -186,7 +205,7 @@ def bench_gae(reps):
 
 BENCHES = {
     "backtest": bench_backtest, "mc": bench_mc, "cov": bench_cov,
-    "indicators": bench_indicators, "lstm": bench_lstm, "env": bench_env,
+    "indicators": bench_indicators, "lstm": bench_lstm, "gru": bench_gru, "env": bench_env,
     "gae": bench_gae, "attn": bench_attn,
 }
 
