@@ -15,7 +15,7 @@ from pathlib import Path
 import numpy as np
 import pandas as pd
 
-from ..data.synthetic import candles_chl_v, generate_ohlcv
+from ..data.synthetic import generate_ohlcv
 
 MARKET_COLS = ["timestamp", "open", "high", "low", "close", "volume"]
 

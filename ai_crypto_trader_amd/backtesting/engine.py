@@ -19,10 +19,10 @@ from pathlib import Path
 import numpy as np
 
 from ..backtesting.engine_cpu import (
-    METRIC_NAMES, NMETRIC, run_backtest_cpu,
+    METRIC_NAMES, run_backtest_cpu,
 )
 from ..backtesting.strategy import (
-    DEFAULT_PARAMS, NPARAM, clip_params, dict_to_params, params_to_dict,
+    clip_params, dict_to_params, params_to_dict,
 )
 from ..ops import gpu_available
 from .data_manager import HistoricalDataManager

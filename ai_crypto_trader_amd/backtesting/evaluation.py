@@ -17,7 +17,7 @@ import numpy as np
 from ..backtesting.engine_cpu import run_backtest_cpu
 from ..backtesting.strategy import clip_params, dict_to_params
 from ..data.synthetic import candles_chl_v, generate_ohlcv
-from .engine import STRATEGY_PRESETS, metrics_to_stats
+from .engine import metrics_to_stats
 
 ANNUAL = 525_600.0
 

@@ -21,8 +21,8 @@ import torch
 
 from ..ops.ga import ga_evolve_cpu
 from ..parallel import dist as pdist
-from .engine_cpu import NMETRIC, run_backtest_cpu
-from .strategy import NPARAM, PARAM_BOUNDS, random_population
+from .engine_cpu import run_backtest_cpu
+from .strategy import PARAM_BOUNDS, random_population
 
 
 class GAEngine:

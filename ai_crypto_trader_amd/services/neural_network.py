@@ -10,7 +10,6 @@ keys. Feature scaling is min-max over the training window (reference
 
 from __future__ import annotations
 
-import time
 from pathlib import Path
 
 import numpy as np

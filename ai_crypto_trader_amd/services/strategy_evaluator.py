@@ -21,8 +21,7 @@ from pathlib import Path
 import numpy as np
 
 from ..backtesting.evaluation import StrategyEvaluationSystem
-from ..backtesting.strategy import params_to_dict
-from ..bus.schema import Channels, Keys
+from ..bus.schema import Keys
 from ..config import get_config
 
 

@@ -20,7 +20,6 @@ import argparse
 import asyncio
 import time
 
-import numpy as np
 
 from ai_crypto_trader_amd.bus.message_bus import InProcessBus
 from ai_crypto_trader_amd.bus.schema import Keys

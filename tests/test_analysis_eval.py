@@ -1,7 +1,6 @@
 """Analysis toolkit, evaluation system, security/monitoring/checkpoint."""
 
 import asyncio
-import logging
 
 import numpy as np
 import pytest

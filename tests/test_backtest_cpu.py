@@ -1,13 +1,12 @@
 """CPU backtest engine behavior tests (the golden reference itself)."""
 
 import numpy as np
-import pytest
 
 from ai_crypto_trader_amd.backtesting.engine_cpu import (
-    METRIC_NAMES, NMETRIC, run_backtest_cpu,
+    NMETRIC, run_backtest_cpu,
 )
 from ai_crypto_trader_amd.backtesting.strategy import (
-    DEFAULT_PARAMS, NPARAM, WARMUP, clip_params, dict_to_params,
+    WARMUP, clip_params, dict_to_params,
     random_population,
 )
 from ai_crypto_trader_amd.data.synthetic import candles_chl_v, generate_ohlcv

@@ -11,7 +11,7 @@ from ai_crypto_trader_amd.backtesting.data_manager import (
     HistoricalDataManager, SocialDataProvider,
 )
 from ai_crypto_trader_amd.backtesting.engine import (
-    STRATEGY_PRESETS, BacktestEngine, metrics_to_stats,
+    BacktestEngine,
 )
 from ai_crypto_trader_amd.backtesting.result_analyzer import ResultAnalyzer
 

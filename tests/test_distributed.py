@@ -9,7 +9,6 @@ CPU container; on the GPU box the same logic runs over RCCL.
 import os
 
 import numpy as np
-import pytest
 import torch
 import torch.distributed as dist
 import torch.multiprocessing as mp

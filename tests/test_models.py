@@ -7,7 +7,7 @@ import torch
 from ai_crypto_trader_amd.models.gru import FusedGRULayer
 from ai_crypto_trader_amd.models.hpo import Pruned, RandomSearchStudy
 from ai_crypto_trader_amd.models.zoo import (
-    MODEL_TYPES, MultitaskPredictor, ProbabilisticPredictor, create_model,
+    MODEL_TYPES, ProbabilisticPredictor, create_model,
 )
 
 

@@ -1,7 +1,6 @@
 """CPU reference implementations: Monte-Carlo, covariance, indicators, GA."""
 
 import numpy as np
-import pytest
 
 from ai_crypto_trader_amd.data.synthetic import candles_chl_v, generate_ohlcv
 from ai_crypto_trader_amd.ops.covar import (

@@ -8,7 +8,7 @@ import pytest
 
 from ai_crypto_trader_amd.bus.message_bus import InProcessBus
 from ai_crypto_trader_amd.bus.schema import (
-    Channels, Keys, MarketUpdate, SocialUpdate, TradingSignal,
+    Keys, MarketUpdate, SocialUpdate, TradingSignal,
 )
 from ai_crypto_trader_amd.config import AppConfig
 from ai_crypto_trader_amd.data.feed import SyntheticFeed
