@@ -159,8 +159,10 @@ class BacktestEngine:
         eng = GAEngine(candles, pop_per_rank=pop_size, device=self.device,
                        seed=seed)
         t0 = time.perf_counter()
+        history = []
         for _ in range(generations):
             eng.step()
+            history.append(float(eng.last_fitness_global.max()))
         eng.eval_fitness()
         elapsed = time.perf_counter() - t0
         fit, best = eng.best()
@@ -169,7 +171,7 @@ class BacktestEngine:
                                   n_candles=n_candles)
         stats["optimize"] = {
             "pop_size": pop_size, "generations": generations,
-            "best_fitness": fit,
+            "best_fitness": fit, "fitness_history": history,
             "candle_evals_per_sec":
                 eng.candle_evals_per_step * generations / elapsed,
         }
