@@ -552,3 +552,47 @@ def test_vp_hist_gpu(dev):
         dn = vol[1:][close[1:] < close[:-1]].sum()
         np.testing.assert_allclose(updown[s].cpu().numpy(), [up, dn],
                                    rtol=1e-4)
+
+
+def test_ga_engine_memory_stable(dev):
+    """50 GA generations allocate no net GPU memory after warmup — the
+    driver's scaling runs and the evolution service loop depend on a flat
+    allocator profile."""
+    from ai_crypto_trader_amd.backtesting.ga_engine import GAEngine
+    from ai_crypto_trader_amd.data.synthetic import (
+        candles_chl_v, generate_ohlcv,
+    )
+
+    candles = candles_chl_v(generate_ohlcv(100_000, 4, seed=1))
+    eng = GAEngine(candles, pop_per_rank=256, device=dev, seed=2,
+                   segments=4)
+    for _ in range(5):
+        eng.step()
+    torch.cuda.synchronize()
+    base = torch.cuda.memory_allocated()
+    for _ in range(45):
+        eng.step()
+    torch.cuda.synchronize()
+    assert torch.cuda.memory_allocated() <= base + (1 << 20)
+
+
+def test_ppo_graphed_memory_stable(dev):
+    from ai_crypto_trader_amd.data.synthetic import (
+        candles_chl_v, generate_ohlcv,
+    )
+    from ai_crypto_trader_amd.models.rl import PPOAgent, TradingVecEnv
+
+    market = torch.from_numpy(
+        candles_chl_v(generate_ohlcv(100_000, 4, seed=2))).to(dev)
+    env = TradingVecEnv(market, n_envs=128, ep_len=512, seed=0)
+    env.reset()
+    agent = PPOAgent(dev, seed=0, use_graph=True)
+    for _ in range(3):
+        agent.train_step(env, horizon=32)
+    torch.cuda.synchronize()
+    base = torch.cuda.memory_allocated()
+    for _ in range(10):
+        agent.train_step(env, horizon=32)
+    torch.cuda.synchronize()
+    assert agent.use_graph                      # graphs actually engaged
+    assert torch.cuda.memory_allocated() <= base + (1 << 20)
