@@ -118,4 +118,25 @@ def calculate_indicator_combinations(u: dict) -> dict:
         "bullish": p5 < -0.5 and rsi > rsi5,
     }
 
+    # volatility_trend_score (:222): BB-band extremity blended with trend
+    # strength — high when a strong trend rides the band edge
+    tstr01 = min(abs(tstr) / 100.0, 1.0)
+    out["volatility_trend_score"] = round(
+        0.7 * abs(bb - 0.5) * 2.0 + 0.3 * tstr01, 4)
+
+    # volume_price_confirmation (:401): does volume confirm the 1m move
+    vol_now = u.get("volume", vol)
+    vratio = vol_now / vol if vol > 0 else 1.0
+    if abs(p1) < 0.1:
+        conf, strength = "neutral", 0.0
+    elif vratio > 1.2:
+        conf = "strong_bullish" if p1 > 0 else "strong_bearish"
+        strength = min(1.0, vratio - 1.0)
+    else:
+        conf = "weak_bullish" if p1 > 0 else "weak_bearish"
+        strength = max(0.0, min(0.5, (vratio - 0.8) / 0.4))
+    out["volume_price_confirmation"] = {
+        "confirmation": conf, "strength": round(strength, 4),
+    }
+
     return out

@@ -115,8 +115,11 @@ def test_indicator_combinations_shape():
     assert c["oscillator_consensus"]["signal"] == "oversold"
     for key in ("triple_ma", "double_rsi", "market_regime",
                 "reversal_probability", "breakout_confirmation",
-                "divergence"):
+                "divergence", "volatility_trend_score",
+                "volume_price_confirmation"):
         assert key in c
+    assert len(c) == 15          # the reference's full composite set
+    assert 0.0 <= c["volatility_trend_score"] <= 1.0
 
 
 def test_volume_profile():
