@@ -21,7 +21,7 @@ PATTERNS = [
     "ascending_triangle", "descending_triangle", "symmetric_triangle",
     "rectangle", "cup_and_handle",
     "rising_wedge", "falling_wedge",
-    "flag_bullish", "flag_bearish",
+    "flag_bullish", "flag_bearish", "pennant",
     "none",
 ]
 BULLISH = {"inverse_head_and_shoulders", "double_bottom",
@@ -73,6 +73,10 @@ def generate_pattern(name: str, rng: np.random.Generator,
         y = np.where(t < 0.5, t * 1.6, 0.8 - (t - 0.5) * 0.2)
     elif name == "flag_bearish":
         y = np.where(t < 0.5, 1.0 - t * 1.6, 0.2 + (t - 0.5) * 0.2)
+    elif name == "pennant":
+        # sharp pole then a small converging (symmetric) consolidation
+        y = np.where(t < 0.4, t * 2.0,
+                     0.8 + 0.25 * (1 - t) * np.sin((t - 0.4) * 30))
     else:  # none: random walk
         y = np.cumsum(rng.standard_normal(n)) * 0.05
     y = y + _noise(rng, n, 0.04)
