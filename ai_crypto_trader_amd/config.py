@@ -126,6 +126,39 @@ class BenchConfig(BaseModel):
     mc_assets: int = 64
 
 
+class ArbitrageConfig(BaseModel):
+    min_profit_pct: float = 0.05              # reference config.json arbitrage_detection
+    max_path_length: int = 3
+    interval_s: float = 30.0
+
+
+class OrderBookConfig(BaseModel):
+    interval_s: float = 2.0                   # reference: 60 s against live Binance
+    depth_levels: int = 20
+    impact_trade_sizes: list[float] = [10_000, 50_000, 100_000, 500_000,
+                                       1_000_000]
+
+
+class NewsConfig(BaseModel):
+    interval_s: float = 2.0                   # reference: 300 s against live feeds
+    max_items_per_source: int = 20
+
+
+class PatternConfig(BaseModel):
+    min_confidence: float = 0.5               # pattern_detection_threshold
+    interval_s: float = 5.0
+
+
+class VolumeProfileConfig(BaseModel):
+    n_bins: int = 24
+    value_area_pct: float = 0.70
+
+
+class FeatureImportanceConfig(BaseModel):
+    n_permutations: int = 30                  # reference config.json:315
+    interval_s: float = 30.0
+
+
 class AppConfig(BaseModel):
     trading: TradingConfig = Field(default_factory=TradingConfig)
     risk: RiskConfig = Field(default_factory=RiskConfig)
@@ -138,6 +171,14 @@ class AppConfig(BaseModel):
     social: SocialConfig = Field(default_factory=SocialConfig)
     bus: BusConfig = Field(default_factory=BusConfig)
     bench: BenchConfig = Field(default_factory=BenchConfig)
+    arbitrage: ArbitrageConfig = Field(default_factory=ArbitrageConfig)
+    order_book: OrderBookConfig = Field(default_factory=OrderBookConfig)
+    news: NewsConfig = Field(default_factory=NewsConfig)
+    patterns: PatternConfig = Field(default_factory=PatternConfig)
+    volume_profile: VolumeProfileConfig = Field(
+        default_factory=VolumeProfileConfig)
+    feature_importance: FeatureImportanceConfig = Field(
+        default_factory=FeatureImportanceConfig)
     data_dir: str = "backtesting_data"
     log_dir: str = "logs"
     seed: int = 0

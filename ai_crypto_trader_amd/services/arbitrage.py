@@ -61,11 +61,13 @@ class ArbitrageDetectionService(Service):
     name = "arbitrage_detection"
 
     def __init__(self, bus, exchange, config=None,
-                 min_profit_pct: float = 0.05, base_currency: str = "USDC",
+                 min_profit_pct: float | None = None,
+                 base_currency: str = "USDC",
                  pairs: list[tuple[str, str]] | None = None):
         super().__init__(bus, config)
         self.exchange = exchange
-        self.min_profit_pct = min_profit_pct
+        self.min_profit_pct = (min_profit_pct if min_profit_pct is not None
+                               else self.config.arbitrage.min_profit_pct)
         self.base = base_currency
         self.pairs = pairs or []
         self.opportunities: list[dict] = []

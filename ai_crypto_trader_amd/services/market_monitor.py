@@ -24,7 +24,9 @@ class MarketMonitorService(Service):
     def __init__(self, bus, feed, config=None):
         super().__init__(bus, config)
         self.feed = feed
-        self.vp = VolumeProfileAnalyzer()
+        vp_cfg = self.config.volume_profile
+        self.vp = VolumeProfileAnalyzer(
+            n_bins=vp_cfg.n_bins, value_area_pct=vp_cfg.value_area_pct)
         self.hist: dict[str, np.ndarray] = {}
         self.hist_len: dict[str, int] = {}
         self.last_vp: dict[str, dict] = {}

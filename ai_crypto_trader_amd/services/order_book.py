@@ -21,6 +21,9 @@ IMPACT_SIZES = [10_000, 50_000, 100_000, 500_000, 1_000_000]   # (:127)
 
 
 class OrderBookAnalyzer:
+    def __init__(self, impact_sizes=None):
+        self.impact_sizes = list(impact_sizes or IMPACT_SIZES)
+
     def analyze(self, book: dict) -> dict:
         bids = np.asarray(book.get("bids", []), dtype=np.float64)
         asks = np.asarray(book.get("asks", []), dtype=np.float64)
@@ -52,7 +55,7 @@ class OrderBookAnalyzer:
     def price_impact(self, bids, asks, mid) -> dict:
         """Walk the book for simulated sizes (:127-244)."""
         out = {}
-        for size in IMPACT_SIZES:
+        for size in self.impact_sizes:
             out[str(size)] = {
                 "buy_impact_bps": self._walk(asks, size, mid, +1),
                 "sell_impact_bps": self._walk(bids, size, mid, -1),
@@ -129,11 +132,15 @@ class OrderBookAnalyzer:
 class OrderBookAnalysisService(Service):
     name = "order_book_analysis"
 
-    def __init__(self, bus, exchange, config=None, interval_s: float = 2.0):
+    def __init__(self, bus, exchange, config=None,
+                 interval_s: float | None = None):
         super().__init__(bus, config)
         self.exchange = exchange
-        self.analyzer = OrderBookAnalyzer()
-        self.interval_s = interval_s
+        ob = self.config.order_book
+        self.analyzer = OrderBookAnalyzer(
+            impact_sizes=ob.impact_trade_sizes)
+        self.interval_s = (interval_s if interval_s is not None
+                           else ob.interval_s)
         self.analyzed = 0
 
     async def run(self):

@@ -16,10 +16,11 @@ class PatternRecognitionService(Service):
     name = "pattern_recognition"
 
     def __init__(self, bus, config=None, device="cpu",
-                 min_confidence: float = 0.5):
+                 min_confidence: float | None = None):
         super().__init__(bus, config)
         self.model = PatternRecognitionModel(device, seed=self.config.seed)
-        self.min_confidence = min_confidence
+        self.min_confidence = (min_confidence if min_confidence is not None
+                               else self.config.patterns.min_confidence)
         self.closes: dict[str, list[float]] = {}
         self.detections = 0
 
