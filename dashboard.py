@@ -14,6 +14,7 @@ standalone against a Redis bus.
 from __future__ import annotations
 
 import asyncio
+import json
 from collections import deque
 
 from ai_crypto_trader_amd.bus.message_bus import InProcessBus
@@ -133,6 +134,37 @@ def build_app(bus, store: DataStore):
         return {
             "updates": store.recent(Channels.STRATEGY_EVOLUTION_UPDATES),
             "params": await bus.get_json(Keys.STRATEGY_PARAMS),
+        }
+
+    @app.get("/api/social")
+    async def social():
+        # social panel (reference dashboard.py:759)
+        metrics = await bus.hgetall(Keys.SOCIAL_METRICS)
+        return {
+            "metrics": {k: json.loads(v) for k, v in metrics.items()},
+            "updates": store.recent(Channels.SOCIAL_UPDATES, 20),
+            "risk_adjustments":
+                await bus.get_json(Keys.SOCIAL_RISK_REPORT),
+        }
+
+    @app.get("/api/correlation")
+    async def correlation():
+        # correlation heatmap data (reference dashboard.py:1712)
+        risk_d = await bus.get_json(Keys.PORTFOLIO_RISK) or {}
+        return {
+            "correlation_matrix": risk_d.get("correlation_matrix"),
+            "symbols": risk_d.get("symbols"),
+            "avg_correlation": risk_d.get("avg_correlation"),
+        }
+
+    @app.get("/api/models")
+    async def models():
+        # AI model performance/comparison panels (reference :1180-1479)
+        return {
+            "registry": await bus.get_json(Keys.MODEL_REGISTRY),
+            "events": store.recent(Channels.MODEL_REGISTRY_EVENTS, 20),
+            "performance":
+                store.recent(Channels.MODEL_PERFORMANCE_UPDATES, 20),
         }
 
     @app.get("/", response_class=HTMLResponse)

@@ -451,3 +451,44 @@ def test_service_runner_builds_all():
     for name in SERVICES:
         svc = build_service(name, InProcessBus(), cfg, args)
         assert svc is not None, name
+
+
+def test_dashboard_api():
+    """Dashboard JSON API: every panel endpoint responds against a seeded
+    bus (reference Dash callbacks -> data endpoints)."""
+    import asyncio as _a
+
+    import httpx
+
+    from ai_crypto_trader_amd.bus.schema import Keys
+    from dashboard import DataStore, build_app
+
+    async def go():
+        bus = InProcessBus()
+        store = DataStore(bus)
+        app = build_app(bus, store)
+        await store.start()
+        await bus.set(Keys.HOLDINGS, {"total_value": 1234.5})
+        await bus.set(Keys.PORTFOLIO_RISK, {
+            "portfolio_var_pct": 2.0, "avg_correlation": 0.4,
+            "correlation_matrix": [[1.0, 0.4], [0.4, 1.0]],
+            "symbols": ["BTCUSDC", "ETHUSDC"]})
+        await bus.hset(Keys.SOCIAL_METRICS, "BTCUSDC",
+                       {"sentiment": 0.7})
+        transport = httpx.ASGITransport(app=app)
+        async with httpx.AsyncClient(transport=transport,
+                                     base_url="http://d") as c:
+            for path in ("/api/portfolio", "/api/risk", "/api/signals",
+                         "/api/trades", "/api/regime", "/api/monte_carlo",
+                         "/api/predictions", "/api/patterns",
+                         "/api/explanations", "/api/evolution",
+                         "/api/social", "/api/correlation", "/api/models",
+                         "/"):
+                r = await c.get(path)
+                assert r.status_code == 200, path
+            r = await c.get("/api/correlation")
+            assert r.json()["avg_correlation"] == 0.4
+            r = await c.get("/api/social")
+            assert r.json()["metrics"]["BTCUSDC"]["sentiment"] == 0.7
+
+    asyncio.run(go())
