@@ -127,6 +127,7 @@ class PortfolioRiskService(Service):
             "positive_definite": bool(pd_ok),
             "avg_correlation": avg_corr,
             "high_correlation_symbols": sorted(set(high_corr)),
+            "correlation_matrix": np.round(corr, 4).tolist(),
             "per_asset": per_asset,
         }
         self.last_risk = risk
