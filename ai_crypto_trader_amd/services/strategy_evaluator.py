@@ -158,6 +158,59 @@ class AIStrategyEvaluator:
         return jp
 
 
+PARAM_ALIASES = {
+    # common spellings in generated strategy code -> canonical param name
+    "rsi_period": "rsi_period", "rsiperiod": "rsi_period",
+    "rsi_oversold": "rsi_oversold", "oversold": "rsi_oversold",
+    "rsi_overbought": "rsi_overbought", "overbought": "rsi_overbought",
+    "ema_fast": "ema_fast", "fast_ema": "ema_fast",
+    "ema_slow": "ema_slow", "slow_ema": "ema_slow",
+    "macd_signal": "macd_signal", "signal_period": "macd_signal",
+    "bb_window": "bb_window", "bb_period": "bb_window",
+    "bollinger_period": "bb_window",
+    "bb_k": "bb_k", "bb_std": "bb_k", "bollinger_std": "bb_k",
+    "entry_votes": "entry_votes", "exit_votes": "exit_votes",
+    "position_size": "position_size_pct",
+    "position_size_pct": "position_size_pct",
+    "stop_loss": "stop_loss_pct", "stop_loss_pct": "stop_loss_pct",
+    "stoploss": "stop_loss_pct",
+    "take_profit": "take_profit_pct", "take_profit_pct": "take_profit_pct",
+    "takeprofit": "take_profit_pct",
+    "trailing_stop": "trailing_stop_pct",
+    "trailing_stop_pct": "trailing_stop_pct",
+    "trailing_activation": "trailing_act_pct",
+    "trailing_act_pct": "trailing_act_pct",
+    "stoch_oversold": "stoch_os", "stoch_os": "stoch_os",
+    "stoch_overbought": "stoch_ob", "stoch_ob": "stoch_ob",
+}
+
+
+def params_from_code(code: str) -> dict:
+    """Extract strategy parameters from strategy CODE text (reference
+    strategy_evolution_service.py:1512-1569: regex pulls params out of the
+    GPT-generated Cloudflare-worker JS). Accepts python/JS-ish assignments
+    and dict/object literals ('stop_loss_pct = 0.03', 'stopLoss: 0.03',
+    '"take_profit": 0.05'); returns canonical param names -> floats, which
+    clip_params/params_to_vec turn into a native kernel strategy."""
+    import re
+
+    found: dict = {}
+    pat = re.compile(
+        r'["\']?([A-Za-z_][A-Za-z0-9_]*)["\']?\s*[:=]\s*'
+        r'(-?\d+(?:\.\d+)?(?:e-?\d+)?)')
+    for name, raw in pat.findall(code):
+        # camelCase -> snake_case, then alias lookup
+        snake = re.sub(r'(?<=[a-z0-9])([A-Z])', r'_\1', name).lower()
+        canon = PARAM_ALIASES.get(snake)
+        if canon is not None:
+            v = float(raw)
+            # percent-style values ("stop_loss = 3" meaning 3%)
+            if canon.endswith("_pct") and v >= 1.0:
+                v /= 100.0
+            found[canon] = v
+    return found
+
+
 class MarketRegimeDataCollector:
     """Assembles regime-training datasets from bus history
     (market_regime_data_collector.py:44-395: price/signal/outcome history

@@ -204,3 +204,39 @@ def test_env_overrides(monkeypatch, tmp_path):
     assert cfg.evolution.population_size == 64
     assert cfg.trading.symbols == ["BTCUSDC", "ETHUSDC"]
     assert cfg.risk.max_portfolio_var_pct == 0.123
+
+
+def test_params_from_code():
+    """Code-to-params bridge (reference regex extraction from generated
+    strategy code): JS and python spellings, percent normalization, and
+    the extracted dict evaluates on the native engine."""
+    from ai_crypto_trader_amd.backtesting.strategy import (
+        DEFAULT_PARAMS, clip_params, params_to_dict,
+    )
+    from ai_crypto_trader_amd.services.strategy_evaluator import (
+        params_from_code,
+    )
+
+    code = """
+    // LLM-generated strategy
+    const stopLoss = 0.03;
+    const takeProfit = 6;       // percent
+    let rsiPeriod = 21;
+    "bollinger_period": 24,
+    oversold: 22
+    entry_votes = 3
+    """
+    got = params_from_code(code)
+    assert got["stop_loss_pct"] == 0.03
+    assert got["take_profit_pct"] == 0.06
+    assert got["rsi_period"] == 21.0
+    assert got["bb_window"] == 24.0
+    assert got["rsi_oversold"] == 22.0
+    assert got["entry_votes"] == 3.0
+    # merge onto defaults -> valid native strategy vector
+    base = params_to_dict(DEFAULT_PARAMS)
+    base.update(got)
+    import numpy as np
+    from ai_crypto_trader_amd.backtesting.strategy import dict_to_params
+    vec = clip_params(dict_to_params(base)[None])[0]
+    assert np.isfinite(vec).all()
