@@ -147,6 +147,47 @@ class SocialMetricsAnalyzer:
         score = float(forest.score_samples(X[-1:])[0])
         return {"anomaly": bool(labels[-1] == -1), "score": score}
 
+    def direction_accuracy(self, symbol: str, horizon: int = 3) -> dict:
+        """Sentiment direction-accuracy / information-coefficient
+        evaluation (reference :457-634): how often sign(sentiment - 0.5)
+        predicted the sign of the forward `horizon`-step return, plus the
+        Pearson IC between sentiment and forward return."""
+        s = np.asarray([x[1] for x in self.history.get(symbol, [])])
+        p = np.asarray(self.price_history.get(symbol, []))
+        n = min(len(s), len(p))
+        if n < horizon + 16:
+            return {"accuracy": 0.5, "ic": 0.0, "n": 0}
+        s = s[-n:]
+        p = p[-n:]
+        fwd = np.log(np.maximum(p[horizon:], 1e-12)) - \
+            np.log(np.maximum(p[:-horizon], 1e-12))
+        sig = s[:-horizon] - 0.5
+        mask = (np.abs(sig) > 1e-6) & (np.abs(fwd) > 1e-12)
+        if mask.sum() < 8:
+            return {"accuracy": 0.5, "ic": 0.0, "n": int(mask.sum())}
+        acc = float((np.sign(sig[mask]) == np.sign(fwd[mask])).mean())
+        a = sig[mask] - sig[mask].mean()
+        b = fwd[mask] - fwd[mask].mean()
+        den = a.std() * b.std()
+        ic = float((a * b).mean() / den) if den > 0 else 0.0
+        return {"accuracy": acc, "ic": ic, "n": int(mask.sum())}
+
+    def update_source_weights(self, per_source_accuracy: dict) -> dict:
+        """Adaptive source-weight updates (reference :635-750): shift
+        weight toward sources whose sentiment has been predictive,
+        blended 80/20 with the current weights and floored so no source
+        dies permanently."""
+        names = list(self.source_weights)
+        acc = np.asarray([per_source_accuracy.get(k, 0.5) for k in names])
+        raw = np.exp((acc - 0.5) * 4.0)
+        new = raw / raw.sum()
+        cur = np.asarray([self.source_weights[k] for k in names])
+        w = 0.8 * cur + 0.2 * new
+        w = np.maximum(w, 0.05)
+        w = w / w.sum()
+        self.source_weights = {k: float(v) for k, v in zip(names, w)}
+        return self.source_weights
+
     def enhanced_sentiment(self, symbol: str) -> dict:
         """Main API (:751): decayed sentiment corrected by lead/lag."""
         base = self.decayed_sentiment(symbol)

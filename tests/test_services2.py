@@ -492,3 +492,34 @@ def test_dashboard_api():
             assert r.json()["metrics"]["BTCUSDC"]["sentiment"] == 0.7
 
     asyncio.run(go())
+
+
+def test_social_accuracy_and_adaptive_weights():
+    """Direction-accuracy/IC self-assessment + adaptive source weights
+    (reference social_metrics_analyzer.py:457-750)."""
+    rng = np.random.default_rng(4)
+    an = SocialMetricsAnalyzer()
+    # sentiment that genuinely leads price by construction
+    price = 100.0
+    for t in range(400):
+        s = float(np.clip(0.5 + 0.3 * np.sin(t / 9.0)
+                          + 0.05 * rng.standard_normal(), 0, 1))
+        an.record("BTCUSDC", s, price, ts=float(t * 60))
+        price *= float(np.exp(0.004 * (s - 0.5)
+                              + 0.0005 * rng.standard_normal()))
+    d = an.direction_accuracy("BTCUSDC")
+    assert d["n"] > 100
+    assert d["accuracy"] > 0.6 and d["ic"] > 0.2
+
+    # adaptive weights shift toward the accurate source, never below floor
+    w0 = dict(an.source_weights)
+    w1 = an.update_source_weights(
+        {"twitter": 0.8, "reddit": 0.45, "news": 0.5})
+    assert w1["twitter"] > w0["twitter"]
+    assert abs(sum(w1.values()) - 1.0) < 1e-9
+    for _ in range(50):
+        w1 = an.update_source_weights(
+            {"twitter": 0.9, "reddit": 0.1, "news": 0.5})
+    # floored pre-normalization, so the post-normalized weight sits just
+    # under the floor — but never collapses to zero
+    assert w1["reddit"] >= 0.04
