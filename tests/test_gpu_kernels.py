@@ -596,3 +596,28 @@ def test_ppo_graphed_memory_stable(dev):
     torch.cuda.synchronize()
     assert agent.use_graph                      # graphs actually engaged
     assert torch.cuda.memory_allocated() <= base + (1 << 20)
+
+
+@pytest.mark.parametrize("seed", [101, 202, 303, 404, 505])
+def test_backtest_bitwise_parity_fuzz(dev, seed):
+    """Multi-seed stress of the bitwise CPU/GPU decision parity: fresh
+    random market + population per seed, trade/win counts must agree
+    EXACTLY on every (strategy, symbol)."""
+    from ai_crypto_trader_amd.backtesting.engine_cpu import run_backtest_cpu
+    from ai_crypto_trader_amd.backtesting.strategy import random_population
+    from ai_crypto_trader_amd.data.synthetic import (
+        candles_chl_v, generate_ohlcv,
+    )
+    from ai_crypto_trader_amd.ops.backtest import run_backtest_gpu
+
+    market = candles_chl_v(
+        generate_ohlcv(3000, 3, seed=seed, sigma=1.0 + (seed % 3)))
+    pop = random_population(96, seed=seed + 7)
+    ref = run_backtest_cpu(market, pop)
+    got = run_backtest_gpu(
+        torch.from_numpy(market).to(dev), torch.from_numpy(pop).to(dev))
+    torch.cuda.synchronize()
+    got = got.cpu().numpy()
+    np.testing.assert_array_equal(got[..., 1], ref[..., 1])
+    np.testing.assert_array_equal(got[..., 2], ref[..., 2])
+    np.testing.assert_allclose(got[..., 0], ref[..., 0], rtol=2e-5)
