@@ -265,7 +265,7 @@ class DQNAgent:
         if self.updates % self.target_sync == 0:
             self.target.load_state_dict(self.q.state_dict())
         self.eps = max(self.eps_min, self.eps * self.eps_decay)
-        return float(loss)
+        return float(loss.detach())
 
 
 class ActorCritic(nn.Module):

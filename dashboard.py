@@ -62,14 +62,18 @@ class DataStore:
 
 
 def build_app(bus, store: DataStore):
+    from contextlib import asynccontextmanager
+
     from fastapi import FastAPI
     from fastapi.responses import HTMLResponse
 
-    app = FastAPI(title="ai-crypto-trader-amd dashboard")
-
-    @app.on_event("startup")
-    async def _startup():
+    @asynccontextmanager
+    async def lifespan(_app):
         await store.start()
+        yield
+
+    app = FastAPI(title="ai-crypto-trader-amd dashboard",
+                  lifespan=lifespan)
 
     @app.get("/api/portfolio")
     async def portfolio():

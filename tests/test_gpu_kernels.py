@@ -279,7 +279,7 @@ def test_lstm_predictor_trains(dev):
         loss = ((pred - y) ** 2).mean()
         loss.backward()
         opt.step()
-        losses.append(float(loss))
+        losses.append(float(loss.detach()))
     assert losses[-1] < losses[0] * 0.7, losses[::10]
 
 
