@@ -458,9 +458,9 @@ class PPOAgent:
                 loss.backward()
                 self._allreduce_grads()
                 self.opt.step()
-                stats["pi_loss"] += float(pi_loss)
-                stats["v_loss"] += float(v_loss)
-                stats["entropy"] += float(ent)
+                stats["pi_loss"] += float(pi_loss.detach())
+                stats["v_loss"] += float(v_loss.detach())
+                stats["entropy"] += float(ent.detach())
         k = self.epochs * self.minibatches
         return {k2: v2 / k for k2, v2 in stats.items()}
 
