@@ -165,6 +165,7 @@ class DCAStrategy(Service):
         self.purchases: list[dict] = []
         self.invested = 0.0
         self.units = 0.0
+        self.banked = 0.0              # rebalance proceeds
         self.recent: list[float] = []
 
     def order_size(self, price: float, sentiment: float = 0.5,
@@ -215,6 +216,30 @@ class DCAStrategy(Service):
         })
         return rec
 
+    def maybe_rebalance(self, price: float,
+                        max_alloc_pct: float = 0.6,
+                        period_mult: int = 30) -> dict | None:
+        """Periodic ("monthly") rebalancing (reference :864-1022): when
+        the position has run far past the target allocation of total
+        DCA capital, trim back to target and bank the proceeds. Runs every
+        `period_mult` DCA periods."""
+        if self.counter == 0 or self.counter % (self.period * period_mult):
+            return None
+        value = self.units * price
+        capital = value + self.banked
+        if capital <= 0 or value / capital <= max_alloc_pct:
+            return None
+        target_value = capital * max_alloc_pct
+        sell_units = (value - target_value) / price
+        self.units -= sell_units
+        proceeds = sell_units * price
+        self.banked += proceeds
+        rec = {"symbol": self.symbol, "action": "rebalance_sell",
+               "qty": sell_units, "price": price, "usd": proceeds,
+               "at": time.time()}
+        self.purchases.append(rec)
+        return rec
+
     async def run(self):
         sub = self.bus.subscribe(Channels.MARKET_UPDATES)
 
@@ -224,6 +249,10 @@ class DCAStrategy(Service):
             price = m["current_price"]
             self.recent.append(price)
             del self.recent[:-512]
+            reb = self.maybe_rebalance(price)
+            if reb:
+                await self.bus.publish(
+                    Channels.DCA_PURCHASE_NOTIFICATIONS, reb)
             sent_raw = await self.bus.hget(Keys.SOCIAL_METRICS, self.symbol)
             sentiment = 0.5
             if sent_raw:

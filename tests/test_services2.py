@@ -523,3 +523,24 @@ def test_social_accuracy_and_adaptive_weights():
     # floored pre-normalization, so the post-normalized weight sits just
     # under the floor — but never collapses to zero
     assert w1["reddit"] >= 0.04
+
+
+def test_dca_rebalance():
+    """Periodic rebalancing trims an over-grown position back to the
+    target allocation (reference dca_strategy.py:864-1022)."""
+    from ai_crypto_trader_amd.config import AppConfig as _AC
+    from ai_crypto_trader_amd.services.grid_dca import DCAStrategy
+
+    svc = DCAStrategy(InProcessBus(), FakeExchange(), "BTCUSDC", _AC(),
+                      candles_per_period=10)
+    svc.units = 10.0
+    svc.invested = 100.0          # value will dwarf invested
+    svc.counter = 10 * 30         # exactly a rebalance tick
+    rec = svc.maybe_rebalance(price=100.0)
+    assert rec is not None and rec["action"] == "rebalance_sell"
+    value = svc.units * 100.0
+    capital = value + svc.banked
+    assert abs(value / capital - 0.6) < 1e-6
+    # off-tick: no rebalance
+    svc.counter += 1
+    assert svc.maybe_rebalance(price=100.0) is None
