@@ -187,14 +187,33 @@ class DCAStrategy(Service):
         hi = max(self.recent[-120:])
         return price < hi * (1 - self.config.dca.dip_threshold_pct)
 
+    def _effective_period(self, regime: str) -> int:
+        """regime_based schedule (:347-451): accumulate twice as often in
+        bear markets, half as often in bull runs."""
+        if self.config.dca.schedule != "regime_based":
+            return self.period
+        mult = {"bear": 0.5, "volatile": 0.75, "bull": 2.0}.get(regime, 1.0)
+        return max(int(self.period * mult), 1)
+
     async def maybe_buy(self, price: float, sentiment: float,
                         regime: str):
         self.counter += 1
-        scheduled = self.counter % self.period == 0
+        scheduled = self.counter % self._effective_period(regime) == 0
         dip = self.is_dip(price)
         if not (scheduled or dip):
             return None
-        usd = self.order_size(price, sentiment, regime)
+        if scheduled and self.config.dca.schedule == "value_averaging":
+            # value averaging (:347-451): buy the gap to a linearly
+            # growing target value instead of a fixed amount
+            periods_done = self.counter // self.period
+            target = self.config.dca.base_order_usd * periods_done
+            gap = target - self.units * price
+            usd = float(np.clip(gap, 0.0,
+                                self.config.dca.base_order_usd * 4))
+            if usd < 1e-6 and not dip:
+                return None
+        else:
+            usd = self.order_size(price, sentiment, regime)
         if dip:
             usd *= self.config.dca.dip_multiplier
         qty = usd / price

@@ -544,3 +544,44 @@ def test_dca_rebalance():
     # off-tick: no rebalance
     svc.counter += 1
     assert svc.maybe_rebalance(price=100.0) is None
+
+
+def test_dca_schedules():
+    """All three DCA schedules behave distinctly (reference :347-451):
+    regime_based buys more often in bear markets; value_averaging buys
+    the gap to a growing target value."""
+    from ai_crypto_trader_amd.config import AppConfig as _AC
+    from ai_crypto_trader_amd.services.grid_dca import DCAStrategy
+
+    async def run_sched(schedule, regime, price_fn):
+        cfg = _AC()
+        cfg.dca.schedule = schedule
+        svc = DCAStrategy(InProcessBus(), FakeExchange(), "BTCUSDC", cfg,
+                          candles_per_period=10)
+        for t in range(100):
+            await svc.maybe_buy(price_fn(t), 0.5, regime)
+        return svc
+
+    async def go():
+        fixed = await run_sched("fixed", "bear", lambda t: 100.0)
+        bear = await run_sched("regime_based", "bear", lambda t: 100.0)
+        bull = await run_sched("regime_based", "bull", lambda t: 100.0)
+        assert len(bear.purchases) > len(fixed.purchases) \
+            > len(bull.purchases)
+
+        va = await run_sched("value_averaging", "ranging", lambda t: 100.0)
+        # flat price: each buy tops value up to base*periods -> invested
+        # grows ~linearly and matches target at each tick
+        assert len(va.purchases) >= 9
+        assert abs(va.units * 100.0
+                   - va.config.dca.base_order_usd * 10) < 1.0
+
+        # rising market: value averaging buys LESS than fixed (the market
+        # does the growing)
+        va_up = await run_sched("value_averaging", "ranging",
+                                lambda t: 100.0 * (1 + 0.01 * t))
+        fixed_up = await run_sched("fixed", "ranging",
+                                   lambda t: 100.0 * (1 + 0.01 * t))
+        assert va_up.invested < fixed_up.invested
+
+    asyncio.run(go())
