@@ -66,6 +66,29 @@ class TechnicalAnalyzer:
     def atr(self) -> float:
         return float(self._col("atr14")[-1])
 
+    def ichimoku(self) -> dict:
+        """Ichimoku cloud (reference binance_ml_strategy.py:40-182 via the
+        `ta` lib): tenkan(9) / kijun(26) midpoints, senkou A/B spans,
+        chikou = close 26 back, and the position vs the cloud."""
+        h, lo = self.candles[:, 1], self.candles[:, 2]
+
+        def midpoint(n):
+            return (h[-n:].max() + lo[-n:].min()) / 2.0
+
+        tenkan = float(midpoint(min(9, len(h))))
+        kijun = float(midpoint(min(26, len(h))))
+        senkou_a = (tenkan + kijun) / 2.0
+        senkou_b = float(midpoint(min(52, len(h))))
+        close = float(self.candles[-1, 0])
+        chikou = float(self.candles[-min(26, len(h)), 0])
+        top, bot = max(senkou_a, senkou_b), min(senkou_a, senkou_b)
+        position = ("above_cloud" if close > top else
+                    "below_cloud" if close < bot else "in_cloud")
+        return {"tenkan": tenkan, "kijun": kijun, "senkou_a": senkou_a,
+                "senkou_b": senkou_b, "chikou": chikou,
+                "position": position,
+                "bullish": close > top and tenkan > kijun}
+
     def vwap(self) -> float:
         return float(self._col("vwap20")[-1])
 

@@ -39,6 +39,10 @@ def test_technical_analyzer(candles):
     assert ta.volatility() > 0
     sr = ta.support_resistance()
     assert sr["support"] < sr["resistance"]
+    ich = ta.ichimoku()
+    assert ich["senkou_a"] == pytest.approx(
+        (ich["tenkan"] + ich["kijun"]) / 2)
+    assert ich["position"] in ("above_cloud", "below_cloud", "in_cloud")
 
 
 def test_position_sizer():
