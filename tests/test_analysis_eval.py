@@ -244,3 +244,33 @@ def test_params_from_code():
     from ai_crypto_trader_amd.backtesting.strategy import dict_to_params
     vec = clip_params(dict_to_params(base)[None])[0]
     assert np.isfinite(vec).all()
+
+
+def test_ga_checkpoint_resume_deterministic(tmp_path):
+    """Checkpoint/resume reproduces the uninterrupted GA trajectory
+    exactly (Philox-seeded evolution is a pure function of (pop, gen))."""
+    from ai_crypto_trader_amd.backtesting.ga_engine import GAEngine
+    from ai_crypto_trader_amd.data.synthetic import (
+        candles_chl_v, generate_ohlcv,
+    )
+    from ai_crypto_trader_amd.utils.checkpoint import CheckpointManager
+
+    candles = candles_chl_v(generate_ohlcv(1200, 2, seed=6))
+
+    ref = GAEngine(candles, pop_per_rank=16, device="cpu", seed=9)
+    for _ in range(6):
+        ref.step()
+
+    a = GAEngine(candles, pop_per_rank=16, device="cpu", seed=9)
+    for _ in range(3):
+        a.step()
+    cm = CheckpointManager(str(tmp_path / "ck"))
+    cm.save_ga(a, tag="resume_test")
+
+    b = GAEngine(candles, pop_per_rank=16, device="cpu", seed=9)
+    assert cm.load_ga(b, tag="resume_test")
+    assert b.gen == 3
+    for _ in range(3):
+        b.step()
+
+    np.testing.assert_array_equal(b.pop_t.numpy(), ref.pop_t.numpy())
