@@ -9,8 +9,9 @@ evolves the parameter set. Optimizers:
   rl      DQN/PPO parameter-nudging (reference :696-975): trains the RL
           agent on recent market snapshots and maps its action preference
           to parameter nudges
-  gpt     optional LLM optimizer seam (reference :364-511) — requires an
-          OpenAI-compatible endpoint; disabled offline
+  gpt     LLM optimizer seam (reference :364-511): an injected llm_fn
+          proposes a param dict or strategy CODE (params_from_code
+          extracts); GA fallback when no llm_fn is configured
   hybrid  regime/volatility-based method selection (:1151-1184)
 Hot-swap via `strategy_params` key + `strategy_update` channel (:349-362);
 per-regime parameter adjustment tables (:145-174); model-version registry
@@ -46,10 +47,12 @@ REGIME_ADJUSTMENTS = {
 class StrategyEvolutionService(Service):
     name = "strategy_evolution"
 
-    def __init__(self, bus, config=None, candles=None, device="cpu"):
+    def __init__(self, bus, config=None, candles=None, device="cpu",
+                 llm_fn=None):
         super().__init__(bus, config)
         self.candles = candles        # (nsym, T, 4) evaluation market
         self.device = device
+        self.llm_fn = llm_fn          # optional GPT-optimizer seam
         self.current_params = DEFAULT_PARAMS.copy()
         self.current_strategy_id = "default"
         self.engine = None
@@ -171,6 +174,25 @@ class StrategyEvolutionService(Service):
             "ppo_entropy": stats.get("entropy", 0.0),
         }
 
+    def optimize_with_llm(self):
+        """GPT optimizer seam (reference :364-511): an injected `llm_fn`
+        receives the current params + measured performance and returns a
+        param dict (or strategy CODE — params_from_code extracts them).
+        Offline default: no llm_fn -> GA fallback."""
+        if self.llm_fn is None:
+            return self.optimize_with_ga()
+        perf = self.evaluate_params(self.current_params)
+        out = self.llm_fn(params_to_dict(self.current_params), perf)
+        if isinstance(out, str):                 # strategy code text
+            from .strategy_evaluator import params_from_code
+            out = params_from_code(out)
+        from ..backtesting.strategy import dict_to_params
+        base = params_to_dict(self.current_params)
+        base.update(out or {})
+        vec = clip_params(dict_to_params(base)[None])[0]
+        return vec, {"llm": True, **{k: v for k, v in perf.items()
+                                     if k in ("sharpe", "win_rate")}}
+
     def adjust_for_regime(self, params: np.ndarray, regime: str):
         """per-regime multiplier tables (reference :302, :145-174)."""
         adj = REGIME_ADJUSTMENTS.get(regime, {})
@@ -211,6 +233,8 @@ class StrategyEvolutionService(Service):
             params, perf = self.optimize_with_rl()
         elif method == "ppo":
             params, perf = self.optimize_with_ppo()
+        elif method == "gpt":
+            params, perf = self.optimize_with_llm()
         else:
             params, perf = self.optimize_with_ga()
         params = self.adjust_for_regime(params, regime)

@@ -585,3 +585,35 @@ def test_dca_schedules():
         assert va_up.invested < fixed_up.invested
 
     asyncio.run(go())
+
+
+def test_llm_evolution_seam():
+    """method='gpt': an injected llm_fn proposing params (as a dict OR as
+    strategy code text) drives the evolution (reference :364-511)."""
+    from ai_crypto_trader_amd.data.synthetic import (
+        candles_chl_v, generate_ohlcv,
+    )
+    from ai_crypto_trader_amd.services.strategy_evolution import (
+        StrategyEvolutionService,
+    )
+
+    candles = candles_chl_v(generate_ohlcv(1200, 2, seed=2))
+    cfg = AppConfig()
+    cfg.evolution.method = "gpt"
+
+    def llm_fn(params, perf):
+        assert "sharpe" in perf
+        return "const stopLoss = 0.07;\nrsi_period = 9\n"
+
+    async def go():
+        bus = InProcessBus()
+        svc = StrategyEvolutionService(bus, cfg, candles=candles,
+                                       device="cpu", llm_fn=llm_fn)
+        out = await svc.evolve_once()
+        assert out["method"] == "gpt" and out["llm"]
+        from ai_crypto_trader_amd.backtesting.strategy import params_to_dict
+        d = params_to_dict(svc.current_params)
+        assert abs(d["stop_loss_pct"] - 0.07) < 1e-6
+        assert d["rsi_period"] == 9.0
+
+    asyncio.run(go())
