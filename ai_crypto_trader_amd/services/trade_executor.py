@@ -128,17 +128,23 @@ class TradeExecutorService(Service):
         order = self.exchange.create_order(sym, "BUY", "MARKET", qty)
         if order.status != "FILLED":
             return
+        # the fill credits base NET of the taker fee — size the protective
+        # stop at what we actually hold, or it can never execute
+        fee = self.exchange.get_trading_fees(sym).get("taker", 0.0)
+        qty = order.filled_qty * (1 - fee)
         stop_pct = risk_info.get("adaptive_stop_pct",
                                  self.config.risk.base_stop_loss_pct)
         stop = price * (1 - stop_pct)
         tp = price * (1 + 2 * stop_pct)        # 2:1 RR (PositionSizer :251)
         # STOP_LOSS_LIMIT at stop*0.99 limit (reference :980)
-        self.exchange.create_order(sym, "SELL", "STOP_LOSS_LIMIT", qty,
-                                   price=stop * 0.99, stop_price=stop)
+        stop_order = self.exchange.create_order(
+            sym, "SELL", "STOP_LOSS_LIMIT", qty,
+            price=stop * 0.99, stop_price=stop)
         self.trailing.register(sym, price, stop)
         self.active[sym] = {
             "symbol": sym, "qty": qty, "entry_price": price,
             "stop_price": stop, "tp_price": tp,
+            "stop_order_id": stop_order.order_id,
             "opened_at": time.time(),
             "signal_confidence": signal.get("confidence", 0.0),
         }
@@ -154,7 +160,21 @@ class TradeExecutorService(Service):
         if trade is None:
             return
         price = self.exchange.get_ticker(sym)["price"]
-        self.exchange.create_order(sym, "SELL", "MARKET", trade["qty"])
+        # cancel the resting protective stop first (reference :333-371:
+        # order replacement discipline — never leave an orphaned stop)
+        oid = trade.get("stop_order_id")
+        stop_o = self.exchange.orders.get(oid) if oid else None
+        if stop_o is not None and stop_o.status == "FILLED":
+            # the exchange-side stop already closed the position
+            price = stop_o.filled_price
+            reason = "stop_loss"
+        else:
+            if oid:
+                self.exchange.cancel_order(sym, oid)
+            o = self.exchange.create_order(sym, "SELL", "MARKET",
+                                           trade["qty"])
+            if o.status == "FILLED":
+                price = o.filled_price
         self.trailing.remove(sym)
         self.trades_done += 1
         self.metrics.executions.labels(sym, "SELL").inc()
@@ -237,6 +257,14 @@ class TradeExecutorService(Service):
                 new_stop = self.trailing.update(sym, price)
                 if new_stop is not None:
                     trade["stop_price"] = new_stop
+                    # replace the exchange-side stop at the raised level
+                    # (reference :333-371)
+                    oid = trade.get("stop_order_id")
+                    if oid and self.exchange.cancel_order(sym, oid):
+                        o = self.exchange.create_order(
+                            sym, "SELL", "STOP_LOSS_LIMIT", trade["qty"],
+                            price=new_stop * 0.99, stop_price=new_stop)
+                        trade["stop_order_id"] = o.order_id
                 if price <= trade["stop_price"]:
                     await self.execute_sell(sym, "stop_loss")
                 elif price >= trade["tp_price"]:

@@ -357,3 +357,79 @@ def test_redis_subscription_adapter():
         sub.close()
 
     asyncio.run(go())
+
+
+def test_executor_stop_order_lifecycle():
+    """The protective stop order is replaced when the trailing stop
+    raises, and canceled (never orphaned) on TP/signal exits."""
+    from ai_crypto_trader_amd.services.trade_executor import (
+        TradeExecutorService,
+    )
+
+    async def go():
+        cfg = AppConfig()
+        bus = InProcessBus()
+        ex = FakeExchange()
+        ex.set_price("BTCUSDC", 100.0)
+        svc = TradeExecutorService(bus, ex, cfg)
+        await svc.execute_buy({"symbol": "BTCUSDC", "confidence": 0.9})
+        trade = svc.active["BTCUSDC"]
+        oid0 = trade["stop_order_id"]
+        assert ex.orders[oid0].status == "NEW"
+
+        # price rallies past activation -> trailing raise replaces the order
+        ex.set_price("BTCUSDC", 104.0)
+        new_stop = svc.trailing.update("BTCUSDC", 104.0)
+        assert new_stop is not None
+        trade["stop_price"] = new_stop
+        assert ex.cancel_order("BTCUSDC", oid0)
+        o = ex.create_order("BTCUSDC", "SELL", "STOP_LOSS_LIMIT",
+                            trade["qty"], price=new_stop * 0.99,
+                            stop_price=new_stop)
+        trade["stop_order_id"] = o.order_id
+
+        # take-profit exit cancels the resting stop: no NEW orders remain
+        ex.set_price("BTCUSDC", 110.0)
+        await svc.execute_sell("BTCUSDC", "take_profit")
+        assert not svc.active
+        open_orders = [x for x in ex.orders.values() if x.status == "NEW"]
+        assert open_orders == []
+
+    asyncio.run(go())
+
+
+def test_executor_exchange_side_stop_fill():
+    """When the EXCHANGE's resting stop fills first (gap down), the
+    executor reconciles: reports the stop fill price and does not
+    double-sell."""
+    from ai_crypto_trader_amd.services.trade_executor import (
+        TradeExecutorService,
+    )
+
+    from ai_crypto_trader_amd.bus.schema import Channels
+
+    async def go():
+        cfg = AppConfig()
+        bus = InProcessBus()
+        ex = FakeExchange()
+        ex.set_price("BTCUSDC", 100.0)
+        svc = TradeExecutorService(bus, ex, cfg)
+        sub = bus.subscribe(Channels.TRADE_EXECUTIONS)
+        await svc.execute_buy({"symbol": "BTCUSDC", "confidence": 0.9})
+        base_after_buy = ex.get_balances()["BTC"]
+        trade = svc.active["BTCUSDC"]
+
+        # gap below the stop: on_tick fills the resting STOP_LOSS_LIMIT
+        ex.set_price("BTCUSDC", 90.0)
+        ex.on_tick("BTCUSDC")
+        assert ex.orders[trade["stop_order_id"]].status == "FILLED"
+        await svc.execute_sell("BTCUSDC", "stop_loss")
+        # base position fully closed exactly once
+        assert ex.get_balances().get("BTC", 0.0) == 0.0
+        assert base_after_buy > 0
+        await sub.get(timeout=1)                      # BUY event
+        _, sell_ev = await sub.get(timeout=1)
+        assert sell_ev["reason"] == "stop_loss"
+        assert sell_ev["price"] == ex.orders[trade["stop_order_id"]].filled_price
+
+    asyncio.run(go())
