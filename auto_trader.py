@@ -190,7 +190,7 @@ def main():
     ap.add_argument("--candles", type=int, default=20_000)
     ap.add_argument("--seed", type=int, default=0)
     args = ap.parse_args()
-    cfg = AppConfig()
+    cfg = AppConfig.load()
     cfg.seed = args.seed
     syms = cfg.trading.symbols
     data = candles_chl_v(generate_ohlcv(args.candles, len(syms),

@@ -133,7 +133,7 @@ async def print_status(bus: InProcessBus, services: list, interval: float):
 async def main():
     args = parse_args()
     symbols = args.symbols.split(",")
-    cfg = AppConfig()
+    cfg = AppConfig.load()
     cfg.trading.symbols = symbols
     cfg.trading.ai_analysis_interval = 0.5     # replay-speed analysis
     cfg.monte_carlo.interval_s = 20.0

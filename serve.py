@@ -37,7 +37,7 @@ def build_server(model_dir: str, device: str):
     from ai_crypto_trader_amd.utils.metrics import GpuTimer, get_metrics
 
     app = FastAPI(title="ai-crypto-trader-amd serving")
-    cfg = AppConfig()
+    cfg = AppConfig.load()
     nn = NeuralNetworkService(InProcessBus(), cfg, device=device)
     n_loaded = nn.load(model_dir)
     analyst = LocalAnalyst()
