@@ -156,13 +156,8 @@ class BacktestEngine:
 
         df = self.dm.load_market_data(symbol, n_candles=n_candles)
         candles = self.dm.to_chlv(df.iloc[:n_candles])
-        # single-symbol optimize launches only pop_size lanes — segment the
-        # history so the GPU sees enough waves (same CV trick as bench.py)
-        T = candles.shape[1]
-        seg = next((s for s in (64, 32, 16, 8, 4, 2)
-                    if T % s == 0 and T // s >= 4096), 1)
         eng = GAEngine(candles, pop_per_rank=pop_size, device=self.device,
-                       seed=seed, segments=seg if self.device != "cpu" else 1)
+                       seed=seed, segments="auto")
         t0 = time.perf_counter()
         history = []
         for _ in range(generations):

@@ -40,7 +40,7 @@ class GAEngine:
         cx_rate: float = 0.5,
         mut_rate: float = 0.15,
         mut_scale: float = 0.1,
-        segments: int = 1,
+        segments: int | str = 1,
     ):
         """segments > 1 splits each symbol's history into `segments`
         independent backtest segments (a pure reshape of the candle
@@ -61,6 +61,17 @@ class GAEngine:
         self.mut_scale = mut_scale
         self.gen = 0
         self.nsym, self.T, _ = candles.shape
+        if segments == "auto":
+            # largest divisor of T (<= 64) keeping segments >= 4096
+            # candles: measured optimum on MI355X (profiles/
+            # backtest_seg64_pmc.json); 1 on CPU (the numpy reference
+            # gains nothing from extra lanes)
+            if str(device) != "cpu":
+                segments = next(
+                    (s for s in (64, 32, 16, 8, 4, 2)
+                     if self.T % s == 0 and self.T // s >= 4096), 1)
+            else:
+                segments = 1
         self.segments = max(int(segments), 1)
         if self.segments > 1:
             assert self.T % self.segments == 0, \

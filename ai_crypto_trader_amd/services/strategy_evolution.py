@@ -92,7 +92,7 @@ class StrategyEvolutionService(Service):
                 device=self.device, seed=self.config.seed + 7,
                 elite_k=e.elite_k, tournament=e.tournament,
                 cx_rate=e.cx_rate, mut_rate=e.mut_rate,
-                mut_scale=e.mut_scale,
+                mut_scale=e.mut_scale, segments="auto",
             )
             # seed the population with the current params (elitism anchor)
             import torch
