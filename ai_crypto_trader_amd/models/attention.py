@@ -1,11 +1,11 @@
 """Fused attention op + transformer encoder layer on the gfx950 kernel
 (ops/hip/attention.hip).
 
-fused_attention(): multi-head attention forward on the HIP kernel for
-S <= 64 (the predictors' shape). Training gradients come from a
-differentiable torch recompute (the kernel forward defines the value;
-backward rebuilds the same math with torch ops — bf16-level agreement is
-covered by tests), so the fused path serves both inference and training.
+fused_attention(): multi-head attention on the HIP kernels for S <= 64
+(the predictors' shape). The forward saves the softmax matrix P when
+grads are enabled; the fused backward kernel (attn_bwd) computes
+dV = P^T dO, the softmax-jacobian correction, and dQ/dK on the same MFMA
+fragments — no torch recompute on the training path.
 """
 
 from __future__ import annotations
@@ -19,8 +19,10 @@ from ..ops import require_hip_ops
 
 
 def attn_fwd_hip(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
-                 scale: float | None = None) -> torch.Tensor:
-    """q/k/v: (BH, S, D) -> (BH, S, D). bf16 in/out, S <= 64."""
+                 scale: float | None = None, save_p: bool = False):
+    """q/k/v: (BH, S, D) -> (BH, S, D). bf16 in/out, S <= 64.
+    save_p=True additionally returns the (BH, 64, 64) softmax matrix for
+    the fused backward."""
     ops = require_hip_ops()
     bh, s, d = q.shape
     if scale is None:
@@ -29,32 +31,52 @@ def attn_fwd_hip(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     k = k.to(torch.bfloat16).contiguous()
     v = v.to(torch.bfloat16).contiguous()
     o = torch.empty_like(q)
+    p = torch.empty((bh, 64, 64), dtype=torch.bfloat16, device=q.device) \
+        if save_p else None
     stream = torch.cuda.current_stream(q.device).cuda_stream
     ops.attn_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
-                 bh, s, d, float(scale), stream)
-    return o
+                 p.data_ptr() if save_p else 0, bh, s, d, float(scale),
+                 stream)
+    return (o, p) if save_p else o
+
+
+def attn_bwd_hip(q, k, v, p, grad_o, scale):
+    """Fused backward: returns (dQ, dK, dV) bf16."""
+    ops = require_hip_ops()
+    bh, s, d = q.shape
+    go = grad_o.to(torch.bfloat16).contiguous()
+    dq = torch.empty_like(q)
+    dk = torch.empty_like(k)
+    dv = torch.empty_like(v)
+    stream = torch.cuda.current_stream(q.device).cuda_stream
+    ops.attn_bwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), p.data_ptr(),
+                 go.data_ptr(), dq.data_ptr(), dk.data_ptr(),
+                 dv.data_ptr(), bh, s, d, float(scale), stream)
+    return dq, dk, dv
 
 
 class _FusedAttention(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, scale):
-        ctx.save_for_backward(q, k, v)
+        need_grad = any(t.requires_grad for t in (q, k, v))
+        qb = q.detach().to(torch.bfloat16).contiguous()
+        kb = k.detach().to(torch.bfloat16).contiguous()
+        vb = v.detach().to(torch.bfloat16).contiguous()
+        if need_grad:
+            o, p = attn_fwd_hip(qb, kb, vb, scale, save_p=True)
+            ctx.save_for_backward(qb, kb, vb, p)
+        else:
+            o = attn_fwd_hip(qb, kb, vb, scale)
         ctx.scale = scale
-        return attn_fwd_hip(q, k, v, scale)
+        ctx.dtypes = (q.dtype, k.dtype, v.dtype)
+        return o
 
     @staticmethod
     def backward(ctx, grad_o):
-        q, k, v = ctx.saved_tensors
-        with torch.enable_grad():
-            q2 = q.detach().float().requires_grad_(True)
-            k2 = k.detach().float().requires_grad_(True)
-            v2 = v.detach().float().requires_grad_(True)
-            s = torch.softmax(
-                (q2 @ k2.transpose(1, 2)) * ctx.scale, dim=-1)
-            o = s @ v2
-            gq, gk, gv = torch.autograd.grad(o, (q2, k2, v2),
-                                             grad_o.float())
-        return gq.to(q.dtype), gk.to(k.dtype), gv.to(v.dtype), None
+        q, k, v, p = ctx.saved_tensors
+        dq, dk, dv = attn_bwd_hip(q, k, v, p, grad_o, ctx.scale)
+        qt, kt, vt = ctx.dtypes
+        return dq.to(qt), dk.to(kt), dv.to(vt), None
 
 
 def fused_attention(q, k, v, scale=None):

@@ -46,6 +46,10 @@ extern "C" void launch_gru_seq_bwd(const float*, const void*, const float*,
                                    const void*, const void*, void*, int, int,
                                    int, hipStream_t);
 extern "C" void launch_attn_fwd(const void*, const void*, const void*,
+                                void*, void*, long, int, int, float,
+                                hipStream_t);
+extern "C" void launch_attn_bwd(const void*, const void*, const void*,
+                                const void*, const void*, void*, void*,
                                 void*, long, int, int, float, hipStream_t);
 extern "C" void launch_lagged_corr(const float*, const float*, float*, int,
                                    int, hipStream_t);
@@ -267,14 +271,32 @@ PYBIND11_MODULE(_hip_ops, m) {
 
     m.def("attn_fwd",
           [](uintptr_t Q, uintptr_t K, uintptr_t V, uintptr_t O,
-             long bh_count, int s_len, int d_head, float scale,
-             uintptr_t stream) {
+             uintptr_t P_out, long bh_count, int s_len, int d_head,
+             float scale, uintptr_t stream) {
               launch_attn_fwd(reinterpret_cast<const void*>(Q),
                               reinterpret_cast<const void*>(K),
                               reinterpret_cast<const void*>(V),
-                              reinterpret_cast<void*>(O), bh_count, s_len,
-                              d_head, scale, as_stream(stream));
+                              reinterpret_cast<void*>(O),
+                              reinterpret_cast<void*>(P_out), bh_count,
+                              s_len, d_head, scale, as_stream(stream));
               check(hipGetLastError(), "attn_fwd launch");
+          });
+
+    m.def("attn_bwd",
+          [](uintptr_t Q, uintptr_t K, uintptr_t V, uintptr_t P,
+             uintptr_t dO, uintptr_t dQ, uintptr_t dK, uintptr_t dV,
+             long bh_count, int s_len, int d_head, float scale,
+             uintptr_t stream) {
+              launch_attn_bwd(reinterpret_cast<const void*>(Q),
+                              reinterpret_cast<const void*>(K),
+                              reinterpret_cast<const void*>(V),
+                              reinterpret_cast<const void*>(P),
+                              reinterpret_cast<const void*>(dO),
+                              reinterpret_cast<void*>(dQ),
+                              reinterpret_cast<void*>(dK),
+                              reinterpret_cast<void*>(dV), bh_count,
+                              s_len, d_head, scale, as_stream(stream));
+              check(hipGetLastError(), "attn_bwd launch");
           });
 
     m.def("lagged_corr",
