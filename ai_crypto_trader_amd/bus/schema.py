@@ -147,10 +147,11 @@ class MarketUpdate:
 
 @dataclass
 class SocialMetricsBlock:
+    # field names verbatim per reference README.md:377-401
     social_volume: float = 0.0
-    engagement: float = 0.0
-    contributors: float = 0.0
-    sentiment: float = 0.5
+    social_engagement: float = 0.0
+    social_contributors: float = 0.0
+    social_sentiment: float = 0.5
     twitter_volume: float = 0.0
     reddit_volume: float = 0.0
     news_volume: float = 0.0

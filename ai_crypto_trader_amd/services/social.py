@@ -49,9 +49,9 @@ class SyntheticSocialSource:
         base = abs(self._rng.standard_normal())
         return SocialMetricsBlock(
             social_volume=1000 * (1 + base),
-            engagement=10_000 * (1 + base),
-            contributors=100 * (1 + base),
-            sentiment=float(sent),
+            social_engagement=10_000 * (1 + base),
+            social_contributors=100 * (1 + base),
+            social_sentiment=float(sent),
             twitter_volume=600 * (1 + base),
             reddit_volume=300 * (1 + base),
             news_volume=100 * (1 + base),
@@ -237,15 +237,15 @@ class SocialMonitorService(Service):
             for sym in list(self.tracked):
                 mb = self.source.metrics(sym, t)
                 upd = SocialUpdate(symbol=sym, metrics=mb,
-                                   weighted_sentiment=mb.sentiment)
+                                   weighted_sentiment=mb.social_sentiment)
                 await self.bus.publish(Channels.SOCIAL_UPDATES,
                                        upd.to_dict())
                 await self.bus.hset(Keys.SOCIAL_METRICS, sym, {
-                    "sentiment": mb.sentiment,
+                    "sentiment": mb.social_sentiment,
                     "social_volume": mb.social_volume,
                 })
                 self.published += 1
-                self.metrics.social_sentiment.labels(sym).set(mb.sentiment)
+                self.metrics.social_sentiment.labels(sym).set(mb.social_sentiment)
             t += 1
             await self.sleep(
                 min(self.config.social.update_interval_s, 1.0))

@@ -57,7 +57,7 @@ def test_schema_payload_shapes():
         assert k in d
     s = SocialUpdate(symbol="BTCUSDC").to_dict()
     assert "data" in s and "metrics" in s["data"]
-    assert "sentiment" in s["data"]["metrics"]
+    assert "social_sentiment" in s["data"]["metrics"]
     t = TradingSignal(symbol="X", decision="BUY", confidence=0.9).to_dict()
     for k in ("decision", "confidence", "reasoning", "risk_level",
               "explanation", "factor_weights", "model_version"):
