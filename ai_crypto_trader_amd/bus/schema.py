@@ -112,6 +112,7 @@ class Keys:
     DCA_PERFORMANCE = "dca_performance"
     DCA_PURCHASE_LIST = "dca_purchase_list"
     ARBITRAGE_OPPORTUNITIES = "arbitrage_opportunities"
+    SELECTED_STRATEGY = "selected_strategy"
 
 
 @dataclass
@@ -190,6 +191,8 @@ class TradingSignal:
     reasoning: str = ""
     risk_level: str = "medium"
     key_indicators: list = field(default_factory=list)
+    social_impact: str = ""                      # README.md:525
+    selected_strategy: dict = field(default_factory=dict)   # :529-543
     explanation: dict = field(default_factory=dict)
     factor_weights: dict = field(default_factory=dict)
     model_version: str = "rule-1.0"
@@ -270,10 +273,17 @@ class NNPrediction:
     predicted_change_pct: float
     confidence: float
     model_type: str = "lstm"
+    status: str = "success"                       # README.md:499
+    prediction_time: str = field(default_factory=now_iso)
+    reference_time: str = field(default_factory=now_iso)
+    training_metrics: dict = field(default_factory=dict)      # :501-505
+    features_used: list = field(default_factory=list)         # :506-512
     timestamp: str = field(default_factory=now_iso)
 
     def to_dict(self) -> dict:
-        return asdict(self)
+        d = asdict(self)
+        d["change_pct"] = self.predicted_change_pct   # reference name
+        return d
 
 
 @dataclass

@@ -148,6 +148,12 @@ class AnalyzerService(Service):
             news_d = json.loads(news) if news else None
             analysis = self.analyst.analyze(
                 market, self.social_cache.get(sym), news_d, nn)
+            soc = analysis["explanation"].get("social_sentiment", 0.0)
+            social_impact = ("positive sentiment" if soc > 0.1 else
+                             "negative sentiment" if soc < -0.1 else
+                             "neutral sentiment")
+            selected = await self.bus.get_json(
+                Keys.SELECTED_STRATEGY) or {}
             sig = TradingSignal(
                 symbol=sym,
                 decision=analysis["decision"],
@@ -155,6 +161,8 @@ class AnalyzerService(Service):
                 reasoning=analysis["reasoning"],
                 risk_level=analysis["risk_level"],
                 key_indicators=analysis["key_indicators"],
+                social_impact=social_impact,
+                selected_strategy=selected,
                 explanation=analysis["explanation"],
                 factor_weights=analysis["factor_weights"],
                 model_version=analysis["model_version"],

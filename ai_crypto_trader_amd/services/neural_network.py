@@ -197,6 +197,8 @@ class NeuralNetworkService(Service):
             current_price=price,
             predicted_change_pct=change_pct,
             confidence=conf,
+            training_metrics={"val_loss": round(vl, 6)},
+            features_used=list(self.FEATURE_NAMES),
         ).to_dict()
 
     FEATURE_NAMES = ["close", "high", "low", "volume", "rsi",

@@ -156,6 +156,18 @@ class StrategySelectionService(Service):
                 sent = sum(vals) / len(vals)
             fi = await self.bus.get_json(Keys.FEATURE_IMPORTANCE)
             best, sc, scores = self.select_optimal(regime, sent, vol, fi)
+            # the selected_strategy block the reference embeds in signals
+            # (README.md:529-543)
+            await self.bus.set(Keys.SELECTED_STRATEGY, {
+                "name": best,
+                "market_regime": regime,
+                "performance_score": round(sc, 4),
+                "risk_profile": self.risk_profiles.get(
+                    best, "moderate") if hasattr(self, "risk_profiles")
+                    else "moderate",
+                "selection_factors": {k: round(v, 4)
+                                      for k, v in scores.items()},
+            })
             if self.should_switch(best, scores):
                 old = self.current
                 self.current = best
