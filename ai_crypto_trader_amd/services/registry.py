@@ -199,13 +199,26 @@ class FeatureImportanceAnalyzer(Service):
             for c, fs in self.CATEGORIES.items()
         }
         top = sorted(imp, key=lambda f: -imp[f]["permutation"])
+        cat_sorted = sorted(cats, key=lambda c: -cats[c])
+        # shape per reference README.md:420-468 (+ the richer per-feature
+        # importances block the integrator consumes)
         self.report = {
-            "at": time.time(), "n_samples": len(self.samples),
+            "at": time.time(),
+            "analysis_type": "feature_importance",
+            "model_type": "RandomForest",
+            "feature_count": len(self.FEATURES),
+            "n_samples": len(self.samples),
             "importances": imp, "categories": cats,
             "top_features": top[:5],
-            "recommendations": [
-                f"weight '{top[0]}' highest", f"consider dropping '{top[-1]}'",
-            ],
+            "top_features_permutation": {
+                ff: round(imp[ff]["permutation"], 5) for ff in top[:10]},
+            "top_categories": {c: round(cats[c], 5) for c in cat_sorted},
+            "recommendations": {
+                "features_to_prioritize": top[:5],
+                "features_to_reconsider": top[-3:],
+                "categories_to_prioritize": cat_sorted[:2],
+                "categories_to_reconsider": cat_sorted[-1:],
+            },
         }
         return self.report
 
