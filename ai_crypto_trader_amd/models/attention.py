@@ -80,7 +80,7 @@ class _FusedAttention(torch.autograd.Function):
 
 
 def fused_attention(q, k, v, scale=None):
-    """(B, H, S, D) or (BH, S, D); HIP forward + recompute backward."""
+    """(B, H, S, D) or (BH, S, D); fused HIP forward AND backward."""
     shape4 = q.dim() == 4
     if shape4:
         B, H, S, D = q.shape
