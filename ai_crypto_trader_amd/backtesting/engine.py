@@ -29,6 +29,13 @@ from .data_manager import HistoricalDataManager
 
 ANNUAL_CANDLES = 525_600.0      # 1m bars
 
+# bars per year by interval (interval-aware Sharpe annualization)
+INTERVAL_BARS_PER_YEAR = {
+    "1m": 525_600.0, "3m": 175_200.0, "5m": 105_120.0,
+    "15m": 35_040.0, "30m": 17_520.0, "1h": 8_760.0,
+    "4h": 2_190.0, "1d": 365.0,
+}
+
 
 # Named strategy presets (the reference ships dca/grid/threshold bots as
 # separate services; in backtests they are parameter presets of the same
@@ -58,12 +65,23 @@ STRATEGY_PRESETS = {
 }
 
 
-def metrics_to_stats(m: np.ndarray, T: int) -> dict:
+def metrics_to_stats(m: np.ndarray, T: int, interval: str = "1m") -> dict:
     """One lane's metric vector -> result dict
-    (strategy_tester.py:403-430 / strategy_evaluation.py:32-228)."""
+    (strategy_tester.py:403-430 / strategy_evaluation.py:32-228).
+    Sharpe is recomputed HOST-SIDE from the kernel's raw return moments
+    (sum_ret/sum_ret2) with the interval's bars-per-year, so non-1m
+    backtests annualize correctly (the in-kernel sharpe/fitness are
+    1m-annualized — the GA fitness convention)."""
     d = dict(zip(METRIC_NAMES, (float(x) for x in m)))
     gp, gl = d["gross_profit"], d["gross_loss"]
-    years = T / ANNUAL_CANDLES
+    bars_py = INTERVAL_BARS_PER_YEAR.get(interval, ANNUAL_CANDLES)
+    years = T / bars_py
+    n = max(T, 1)
+    mean_r = d["sum_ret"] / n
+    var_r = max(d["sum_ret2"] / n - mean_r * mean_r, 0.0)
+    sharpe = (mean_r / max(np.sqrt(var_r), 1e-9) * np.sqrt(bars_py)
+              if d["n_trades"] > 0 else 0.0)
+    d["sharpe"] = float(sharpe)
     return {
         "final_equity": d["final_equity"],
         "total_return_pct": (d["final_equity"] - 1.0) * 100.0,
@@ -106,7 +124,7 @@ class BacktestEngine:
         elapsed = time.perf_counter() - t0
         metrics = out[0] if record_equity else out
         T = candles.shape[1]
-        stats = metrics_to_stats(metrics[0, 0], T)
+        stats = metrics_to_stats(metrics[0, 0], T, interval)
         stats.update({
             "symbol": symbol, "strategy": strategy, "interval": interval,
             "n_candles": T, "engine": self.device,

@@ -139,3 +139,27 @@ def test_custom_strategy_tester():
     tester.backtest_strategy(candles, probe)
     for k in ("rsi14", "macd", "bb_up", "close", "in_position"):
         assert k in seen
+
+
+def test_interval_aware_sharpe(tmp_path):
+    """Sharpe annualizes by the interval's bars-per-year (recomputed
+    host-side from the kernel's raw return moments): the same per-bar
+    return stream annualizes sqrt(525600/8760) ~ 7.75x higher at 1m than
+    at 1h."""
+    import numpy as np
+
+    from ai_crypto_trader_amd.backtesting.engine import metrics_to_stats
+    from ai_crypto_trader_amd.backtesting.engine_cpu import METRIC_NAMES
+
+    m = np.zeros(len(METRIC_NAMES), np.float32)
+    d = dict(zip(METRIC_NAMES, range(len(METRIC_NAMES))))
+    m[d["final_equity"]] = 1.1
+    m[d["n_trades"]] = 5
+    m[d["wins"]] = 3
+    m[d["sum_ret"]] = 0.10          # T=1000 bars of identical moments
+    m[d["sum_ret2"]] = 0.001
+    s1m = metrics_to_stats(m, 1000, "1m")["sharpe"]
+    s1h = metrics_to_stats(m, 1000, "1h")["sharpe"]
+    assert s1m == pytest.approx(s1h * np.sqrt(525_600 / 8_760), rel=1e-6)
+    # default matches the kernel's 1m convention closely
+    assert s1m > 0
