@@ -21,10 +21,10 @@ extern "C" void launch_ga_evolve(const float*, const float*, const int*,
                                  hipStream_t);
 extern "C" void launch_mc_paths(const float*, const float*, const float*,
                                 const float*, float*, float*, int, int, long,
-                                float, uint64_t, long, hipStream_t);
+                                float, uint64_t, long, int, hipStream_t);
 extern "C" void launch_mc_paths_mfma(const float*, const float*,
                                      const float*, float*, float*, int, int,
-                                     long, float, uint64_t, long,
+                                     long, float, uint64_t, long, int,
                                      hipStream_t);
 extern "C" void launch_cov(const float*, float*, int, int, hipStream_t);
 extern "C" void launch_indicators(const float*, float*, int, int, int,
@@ -109,7 +109,8 @@ PYBIND11_MODULE(_hip_ops, m) {
           [](uintptr_t chol, uintptr_t drift, uintptr_t vol_sqrt_dt,
              uintptr_t weights, uintptr_t final_value, uintptr_t max_dd,
              int n_assets, int n_steps, long n_paths, float s0,
-             uint64_t seed, long path_base, uintptr_t stream) {
+             uint64_t seed, long path_base, int antithetic,
+             uintptr_t stream) {
               launch_mc_paths(reinterpret_cast<const float*>(chol),
                               reinterpret_cast<const float*>(drift),
                               reinterpret_cast<const float*>(vol_sqrt_dt),
@@ -117,7 +118,7 @@ PYBIND11_MODULE(_hip_ops, m) {
                               reinterpret_cast<float*>(final_value),
                               reinterpret_cast<float*>(max_dd), n_assets,
                               n_steps, n_paths, s0, seed, path_base,
-                              as_stream(stream));
+                              antithetic, as_stream(stream));
               check(hipGetLastError(), "mc_paths launch");
           });
 
@@ -125,14 +126,14 @@ PYBIND11_MODULE(_hip_ops, m) {
           [](uintptr_t chol, uintptr_t drift, uintptr_t vol_sqrt_dt,
              uintptr_t final_value, uintptr_t max_dd, int n_assets,
              int n_steps, long n_paths, float s0, uint64_t seed,
-             long path_base, uintptr_t stream) {
+             long path_base, int antithetic, uintptr_t stream) {
               launch_mc_paths_mfma(reinterpret_cast<const float*>(chol),
                                    reinterpret_cast<const float*>(drift),
                                    reinterpret_cast<const float*>(vol_sqrt_dt),
                                    reinterpret_cast<float*>(final_value),
                                    reinterpret_cast<float*>(max_dd),
                                    n_assets, n_steps, n_paths, s0, seed,
-                                   path_base, as_stream(stream));
+                                   path_base, antithetic, as_stream(stream));
               check(hipGetLastError(), "mc_paths_mfma launch");
           });
 

@@ -38,7 +38,8 @@ __global__ void __launch_bounds__(256) mc_paths_kernel(
     const float* __restrict__ wS0,      // (A,)   weight_a * S0_a
     float* __restrict__ final_value,    // (n_paths,)
     float* __restrict__ max_dd,         // (n_paths,)
-    int n_steps, long n_paths, float v0, uint64_t seed, long path_base)
+    int n_steps, long n_paths, float v0, uint64_t seed, long path_base,
+    int antithetic)
 {
     __shared__ float lds_cvol[A * A];
     __shared__ float lds_drift[A];
@@ -59,11 +60,20 @@ __global__ void __launch_bounds__(256) mc_paths_kernel(
         for (int a = 0; a < A; ++a) logS[a] = 0.0f;
         float vmax = v0, mdd = 0.0f, V = v0;
 
+        // antithetic pairing: the top half of the path range mirrors the
+        // bottom half's normals with flipped sign (classic MC variance
+        // reduction; counter-based RNG makes the pairing stateless)
+        const long half = n_paths / 2;
+        const bool mirror = antithetic && path >= half;
+        const long zsrc = (mirror ? path - half : path) + path_base;
+        const float zsign = mirror ? -1.0f : 1.0f;
         for (int step = 0; step < n_steps; ++step) {
             for (int k4 = 0; k4 < A / 4; ++k4) {
                 float4 z4 = philox_normal4(
-                    seed, (uint64_t)(path + path_base),
+                    seed, (uint64_t)zsrc,
                     ((uint64_t)step << 32) | (uint64_t)k4);
+                z4.x *= zsign; z4.y *= zsign;
+                z4.z *= zsign; z4.w *= zsign;
 #pragma unroll
                 for (int dz = 0; dz < 4; ++dz) {
                     const float z = dz == 0   ? z4.x
@@ -127,7 +137,8 @@ __global__ void __launch_bounds__(512) mc_paths_mfma_kernel(
     const float* __restrict__ wS0,      // (A,)
     float* __restrict__ final_value,    // (n_paths,)
     float* __restrict__ max_dd,         // (n_paths,)
-    int n_steps, long n_paths, float v0, uint64_t seed, long path_base)
+    int n_steps, long n_paths, float v0, uint64_t seed, long path_base,
+    int antithetic)
 {
     // 8 waves: wave w owns path-columns [32w, 32w+32) -> 8 accumulator
     // fragments per lane (32 f32) instead of 16, so 2+ waves/SIMD fit;
@@ -156,7 +167,12 @@ __global__ void __launch_bounds__(512) mc_paths_mfma_kernel(
     if (path0 >= n_paths) return;
     const int zrow = tid & (MCM_PB - 1);      // this thread's Z^T row
     const int zhalf = tid >> 8;               // which 32-normal half
-    const long zpath = path0 + zrow + path_base;
+    // antithetic pairing across the global path range (see VALU kernel)
+    const long gpath = path0 + zrow;
+    const long half = n_paths / 2;
+    const bool mirror = antithetic && gpath >= half;
+    const long zpath = (mirror ? gpath - half : gpath) + path_base;
+    const float zsign = mirror ? -1.0f : 1.0f;
 
     mc_f32x4 logS[4][2];                      // [m_tile][col_tile]
 #pragma unroll
@@ -189,8 +205,9 @@ __global__ void __launch_bounds__(512) mc_paths_mfma_kernel(
                 ((uint64_t)step << 32) | (uint64_t)k4);
             // one 8-byte store (offset = zrow*144 + k4*8 bytes, 8-aligned)
             *reinterpret_cast<mc_bf16x4*>(lds_zt + zrow * MCM_KP + k4 * 4) =
-                mc_bf16x4{(mc_bf16)z4.x, (mc_bf16)z4.y, (mc_bf16)z4.z,
-                          (mc_bf16)z4.w};
+                mc_bf16x4{(mc_bf16)(z4.x * zsign), (mc_bf16)(z4.y * zsign),
+                          (mc_bf16)(z4.z * zsign),
+                          (mc_bf16)(z4.w * zsign)};
         }
         __syncthreads();
 
@@ -255,7 +272,7 @@ extern "C" void launch_mc_paths_mfma(const float* cvol, const float* drift,
                                      float* max_dd, int n_assets,
                                      int n_steps, long n_paths, float v0,
                                      uint64_t seed, long path_base,
-                                     hipStream_t stream) {
+                                     int antithetic, hipStream_t stream) {
     if (n_assets != MCM_A)
         throw std::runtime_error("mc_paths_mfma: n_assets must be 64");
     if (n_paths % MCM_PB != 0)
@@ -264,7 +281,8 @@ extern "C" void launch_mc_paths_mfma(const float* cvol, const float* drift,
     long blocks = n_paths / MCM_PB;
     hipLaunchKernelGGL(mc_paths_mfma_kernel, dim3((unsigned)blocks),
                        dim3(512), 0, stream, cvol, drift, wS0, final_value,
-                       max_dd, n_steps, n_paths, v0, seed, path_base);
+                       max_dd, n_steps, n_paths, v0, seed, path_base,
+                       antithetic);
 }
 
 extern "C" void launch_mc_paths(const float* cvol, const float* drift,
@@ -272,7 +290,7 @@ extern "C" void launch_mc_paths(const float* cvol, const float* drift,
                                 float* final_value, float* max_dd,
                                 int n_assets, int n_steps, long n_paths,
                                 float v0, uint64_t seed, long path_base,
-                                hipStream_t stream) {
+                                int antithetic, hipStream_t stream) {
     (void)weights_unused;
     long want = (n_paths + 255) / 256;
     int blocks = (int)(want < 8192 ? want : 8192);
@@ -281,7 +299,7 @@ extern "C" void launch_mc_paths(const float* cvol, const float* drift,
     case AA:                                                                  \
         hipLaunchKernelGGL(mc_paths_kernel<AA>, grid, block, 0, stream, cvol, \
                            drift, wS0, final_value, max_dd, n_steps, n_paths, \
-                           v0, seed, path_base);                              \
+                           v0, seed, path_base, antithetic);                  \
         break;
     switch (n_assets) {
         MC_CASE(4)
