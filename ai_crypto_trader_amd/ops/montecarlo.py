@@ -226,7 +226,7 @@ def risk_stats(final_values, v0: float, confidences=(0.95, 0.99)):
 def mc_paths_sharded(
     chol, mu, sigma, weights, *, n_paths_total: int, n_steps: int,
     dt: float, rank: int = 0, world: int = 1, seed: int = 0, s0: float = 1.0,
-    device="cuda", n_bins: int = 2048,
+    device="cuda", n_bins: int = 2048, antithetic: bool = False,
 ):
     """Monte-Carlo sharded across ranks (path batches DP
     across GPUs, stats via RCCL all-reduce).
@@ -260,7 +260,8 @@ def mc_paths_sharded(
     else:
         fv, dd = mc_paths_gpu(
             chol, mu, sigma, weights, n_steps=n_steps, n_paths=n_local,
-            dt=dt, s0=s0, seed=seed, device=device, path_base=lo)
+            dt=dt, s0=s0, seed=seed, device=device, path_base=lo,
+            antithetic=antithetic)    # pairs mirror within each shard
 
     v0 = float(np.sum(np.asarray(weights) * s0))
     # all-reduced moments
