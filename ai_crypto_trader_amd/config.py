@@ -39,6 +39,7 @@ class RiskConfig(BaseModel):
 
 
 class MonteCarloConfig(BaseModel):
+    simulation_method: str = "gbm"            # gbm | historical (:275-298)
     num_simulations: int = 10_000             # reference default 1000
     time_horizon_days: int = 30               # config.json:87-89
     n_assets: int = 16

@@ -26,6 +26,9 @@ extern "C" void launch_mc_paths_mfma(const float*, const float*,
                                      const float*, float*, float*, int, int,
                                      long, float, uint64_t, long, int,
                                      hipStream_t);
+extern "C" void launch_mc_bootstrap(const float*, const float*, float*,
+                                    float*, int, int, int, long, float,
+                                    uint64_t, long, hipStream_t);
 extern "C" void launch_cov(const float*, float*, int, int, hipStream_t);
 extern "C" void launch_indicators(const float*, float*, int, int, int,
                                   hipStream_t);
@@ -135,6 +138,20 @@ PYBIND11_MODULE(_hip_ops, m) {
                                    n_assets, n_steps, n_paths, s0, seed,
                                    path_base, antithetic, as_stream(stream));
               check(hipGetLastError(), "mc_paths_mfma launch");
+          });
+
+    m.def("mc_bootstrap",
+          [](uintptr_t logret, uintptr_t wS0, uintptr_t final_value,
+             uintptr_t max_dd, int n_assets, int t_hist, int n_steps,
+             long n_paths, float v0, uint64_t seed, long path_base,
+             uintptr_t stream) {
+              launch_mc_bootstrap(reinterpret_cast<const float*>(logret),
+                                  reinterpret_cast<const float*>(wS0),
+                                  reinterpret_cast<float*>(final_value),
+                                  reinterpret_cast<float*>(max_dd),
+                                  n_assets, t_hist, n_steps, n_paths, v0,
+                                  seed, path_base, as_stream(stream));
+              check(hipGetLastError(), "mc_bootstrap launch");
           });
 
     m.def("cov",

@@ -313,3 +313,96 @@ extern "C" void launch_mc_paths(const float* cvol, const float* drift,
     }
 #undef MC_CASE
 }
+
+// ---------------------------------------------------------------------------
+// Historical bootstrap simulation (reference monte_carlo_service.py:275-298
+// 'historical' method): paths resample observed log-return ROWS with
+// replacement instead of drawing GBM normals. One lane = one portfolio
+// path; the same sampled time index applies to every asset of a step, so
+// the empirical cross-asset correlation is preserved exactly (the
+// reference bootstraps per asset independently and then ignores
+// correlation in its portfolio sum — joint row resampling is the
+// correlation-correct version of the same estimator). The (T_hist, A)
+// log-return table is read through L2 (well under a slice of it); the
+// Philox counter is the (path, step) pair so streams are reproducible
+// and DP-shardable exactly like the GBM kernel.
+// ---------------------------------------------------------------------------
+
+namespace {
+
+template <int A>
+__global__ void __launch_bounds__(256) mc_bootstrap_kernel(
+    const float* __restrict__ logret,   // (T_hist, A) log2-returns
+    const float* __restrict__ wS0,      // (A,) weight * S0
+    float* __restrict__ final_value,    // (n_paths,)
+    float* __restrict__ max_dd,
+    int t_hist, int n_steps, long n_paths, float v0, uint64_t seed,
+    long path_base)
+{
+    __shared__ float lds_w[A];
+    for (int i = threadIdx.x; i < A; i += blockDim.x)
+        lds_w[i] = wS0[i];
+    __syncthreads();
+
+    const long stride = (long)gridDim.x * blockDim.x;
+    for (long path = (long)blockIdx.x * blockDim.x + threadIdx.x;
+         path < n_paths; path += stride) {
+        float logS[A];
+#pragma unroll
+        for (int a = 0; a < A; ++a) logS[a] = 0.0f;
+        float vmax = v0, mdd = 0.0f, V = v0;
+
+        for (int s4 = 0; s4 < (n_steps + 3) / 4; ++s4) {
+            Philox4 r = philox4x32(seed, (uint64_t)(path + path_base),
+                                   0x8000000000000000ull | (uint64_t)s4);
+            const uint32_t ix[4] = {r.x, r.y, r.z, r.w};
+#pragma unroll
+            for (int j = 0; j < 4; ++j) {
+                const int step = s4 * 4 + j;
+                if (step >= n_steps) break;
+                const int t = (int)(ix[j] % (uint32_t)t_hist);
+                const float* row = logret + (long)t * A;
+                V = 0.0f;
+#pragma unroll
+                for (int a = 0; a < A; ++a) {
+                    logS[a] += row[a];
+                    V += lds_w[a] * __builtin_amdgcn_exp2f(logS[a]);
+                }
+                vmax = fmaxf(vmax, V);
+                mdd = fmaxf(mdd,
+                            (vmax - V) * __builtin_amdgcn_rcpf(vmax));
+            }
+        }
+        final_value[path] = V;
+        max_dd[path] = mdd;
+    }
+}
+
+}  // namespace
+
+extern "C" void launch_mc_bootstrap(const float* logret, const float* wS0,
+                                    float* final_value, float* max_dd,
+                                    int n_assets, int t_hist, int n_steps,
+                                    long n_paths, float v0, uint64_t seed,
+                                    long path_base, hipStream_t stream) {
+    long want = (n_paths + 255) / 256;
+    int blocks = (int)(want < 8192 ? want : 8192);
+    dim3 grid(blocks), block(256);
+#define MCB_CASE(AA)                                                          \
+    case AA:                                                                  \
+        hipLaunchKernelGGL(mc_bootstrap_kernel<AA>, grid, block, 0, stream,   \
+                           logret, wS0, final_value, max_dd, t_hist,          \
+                           n_steps, n_paths, v0, seed, path_base);            \
+        break;
+    switch (n_assets) {
+        MCB_CASE(4)
+        MCB_CASE(8)
+        MCB_CASE(16)
+        MCB_CASE(32)
+        MCB_CASE(64)
+        default:
+            throw std::runtime_error(
+                "mc_bootstrap: n_assets must be one of 4/8/16/32/64");
+    }
+#undef MCB_CASE
+}

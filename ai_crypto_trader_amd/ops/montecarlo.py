@@ -197,6 +197,64 @@ def mc_paths_gpu(
     return fv, dd
 
 
+def mc_bootstrap_cpu(log_returns, weights, *, n_steps: int, n_paths: int,
+                     s0: float = 1.0, seed: int = 0):
+    """CPU twin of mc_bootstrap_kernel (reference 'historical' method,
+    monte_carlo_service.py:275-298): paths resample historical log-return
+    ROWS with replacement (joint resampling preserves the empirical
+    cross-asset correlation). Same Philox index stream as the kernel."""
+    f32 = np.float32
+    lr = np.asarray(log_returns, f32)           # (T_hist, A) natural logs
+    lr2 = (lr * np.float32(_LOG2E)).astype(f32)  # base-2 like the kernel
+    T_hist, A = lr2.shape
+    wS0 = (np.asarray(weights) * s0).astype(f32)
+    v0 = f32(wS0.sum())
+    logS = np.zeros((n_paths, A), f32)
+    vmax = np.full(n_paths, v0, f32)
+    mdd = np.zeros(n_paths, f32)
+    V = np.full(n_paths, v0, f32)
+    paths = np.arange(n_paths, dtype=np.uint64)
+    for s4 in range((n_steps + 3) // 4):
+        ctr_hi = np.full(n_paths,
+                         np.uint64(0x8000000000000000) | np.uint64(s4),
+                         np.uint64)
+        u = philox4x32_np(seed, paths, ctr_hi)       # 4 x (P,) uint32
+        for j in range(4):
+            step = s4 * 4 + j
+            if step >= n_steps:
+                break
+            t = (u[j] % np.uint32(T_hist)).astype(np.int64)
+            logS += lr2[t]
+            V = (wS0 * np.exp2(logS)).sum(axis=1).astype(f32)
+            vmax = np.maximum(vmax, V)
+            mdd = np.maximum(mdd, (vmax - V) / vmax)
+    return V, mdd
+
+
+def mc_bootstrap_gpu(log_returns, weights, *, n_steps: int, n_paths: int,
+                     s0: float = 1.0, seed: int = 0, device="cuda",
+                     path_base: int = 0):
+    """GPU historical-bootstrap paths; returns (final_value, max_dd)."""
+    import torch
+
+    ops = require_hip_ops()
+    f32 = np.float32
+    lr = np.asarray(log_returns, f32)
+    lr2 = np.ascontiguousarray(lr * np.float32(_LOG2E), dtype=f32)
+    T_hist, A = lr2.shape
+    wS0 = (np.asarray(weights) * s0).astype(f32)
+    v0 = float(wS0.sum())
+    t_lr = torch.from_numpy(lr2).to(device)
+    t_w = torch.from_numpy(wS0).to(device)
+    fv = torch.empty(n_paths, dtype=torch.float32, device=device)
+    dd = torch.empty(n_paths, dtype=torch.float32, device=device)
+    stream = torch.cuda.current_stream(fv.device).cuda_stream
+    ops.mc_bootstrap(t_lr.data_ptr(), t_w.data_ptr(), fv.data_ptr(),
+                     dd.data_ptr(), A, T_hist, n_steps, n_paths, v0,
+                     seed, path_base, stream)
+    return fv, dd
+
+
 def risk_stats(final_values, v0: float, confidences=(0.95, 0.99)):
     """VaR/CVaR/percentiles/prob-profit from per-path final values
     (monte_carlo_service.py:304-325 semantics). Works on numpy arrays or

@@ -73,6 +73,16 @@ class MonteCarloService(Service):
             corr = np.eye(1)
         return mu, sigma, np.linalg.cholesky(corr)
 
+    def _log_return_rows(self, syms) -> np.ndarray | None:
+        """(T-1, A) joint log-return rows for the historical bootstrap
+        (None when histories are too short)."""
+        n = min(len(self.prices[s]) for s in syms)
+        if n < 32:
+            return None
+        px = np.stack([np.asarray(self.prices[s][-n:], np.float64)
+                       for s in syms], axis=1)
+        return np.diff(np.log(px), axis=0).astype(np.float32)
+
     def simulate(self, syms: list[str], scenario: str = "base",
                  n_paths: int | None = None) -> dict:
         """One portfolio simulation under a scenario (reference
@@ -98,6 +108,45 @@ class MonteCarloService(Service):
         dt = 1.0 / 365.0
         n_paths = n_paths or self.config.monte_carlo.num_simulations
         seed = (self.runs * 977 + 13) & 0xFFFFFFFF
+
+        if self.config.monte_carlo.simulation_method == "historical":
+            lr = self._log_return_rows(syms)
+            if lr is not None:
+                # scenario multipliers rescale the empirical rows around
+                # their mean (shape-preserving; reference :275-298)
+                lm = lr.mean(axis=0, keepdims=True)
+                lr_s = lm * m_mu + (lr - lm) * m_sig
+                if pad:
+                    lr_s = np.concatenate(
+                        [lr_s, np.zeros((len(lr_s), pad), np.float32)],
+                        axis=1)
+                if gpu_available() and lr_s.shape[1] in (4, 8, 16, 32, 64):
+                    import torch
+                    from ..ops.montecarlo import mc_bootstrap_gpu
+                    fv, dd = mc_bootstrap_gpu(lr_s, w, n_steps=days,
+                                              n_paths=n_paths, seed=seed)
+                    torch.cuda.synchronize()
+                    stats = risk_stats(fv, v0=1.0)
+                    stats["max_drawdown_mean"] = float(dd.mean())
+                    stats["max_drawdown_p95"] = float(
+                        torch.quantile(dd, 0.95))
+                else:
+                    from ..ops.montecarlo import mc_bootstrap_cpu
+                    fv, dd = mc_bootstrap_cpu(
+                        lr_s, w, n_steps=days,
+                        n_paths=min(n_paths, 20_000), seed=seed)
+                    stats = risk_stats(fv, v0=1.0)
+                    stats["max_drawdown_mean"] = float(dd.mean())
+                    stats["max_drawdown_p95"] = float(
+                        np.percentile(dd, 95))
+                stats["method"] = "historical"
+                self.runs += 1
+                stats.update({
+                    "scenario": scenario, "symbols": syms,
+                    "n_paths": int(n_paths), "horizon_days": days,
+                    "timestamp": time.time(),
+                })
+                return stats
 
         if gpu_available() and len(mu) in (4, 8, 16, 32, 64):
             import torch
@@ -140,12 +189,19 @@ class MonteCarloService(Service):
         streams')."""
         import torch
 
-        from ..ops.montecarlo import mc_paths_gpu, risk_stats
+        from ..ops.montecarlo import (
+            mc_bootstrap_gpu, mc_paths_gpu, risk_stats,
+        )
 
         mu0, sigma0, chol = self._estimate_params(syms)
         w = np.full(len(syms), 1.0 / len(syms))
         days = self.config.monte_carlo.time_horizon_days
         n_paths = self.config.monte_carlo.num_simulations
+        historical = (self.config.monte_carlo.simulation_method
+                      == "historical")
+        lr = self._log_return_rows(syms) if historical else None
+        if lr is not None:
+            lr_mean = lr.mean(axis=0, keepdims=True)
         from ..utils.metrics import GpuTimer
 
         streams = {}
@@ -155,11 +211,21 @@ class MonteCarloService(Service):
             st = torch.cuda.Stream()
             streams[scen] = st
             with torch.cuda.stream(st):
-                fv, dd = mc_paths_gpu(
-                    chol, mu0 * m_mu, np.maximum(sigma0 * m_sig, 1e-4), w,
-                    n_steps=days, n_paths=n_paths, dt=1.0 / 365.0,
-                    seed=(self.runs * 977 + 13) & 0xFFFFFFFF,
-                    antithetic=True)
+                if lr is not None:
+                    # historical bootstrap (reference :275-298): scenario
+                    # multipliers rescale the empirical return rows around
+                    # their mean (shape-preserving)
+                    lr_s = lr_mean * m_mu + (lr - lr_mean) * m_sig
+                    fv, dd = mc_bootstrap_gpu(
+                        lr_s, w, n_steps=days, n_paths=n_paths,
+                        seed=(self.runs * 977 + 13) & 0xFFFFFFFF)
+                else:
+                    fv, dd = mc_paths_gpu(
+                        chol, mu0 * m_mu,
+                        np.maximum(sigma0 * m_sig, 1e-4), w,
+                        n_steps=days, n_paths=n_paths, dt=1.0 / 365.0,
+                        seed=(self.runs * 977 + 13) & 0xFFFFFFFF,
+                        antithetic=True)
                 results[scen] = (fv, dd)
             self.runs += 1
         for st in streams.values():
