@@ -113,6 +113,7 @@ class Keys:
     DCA_PURCHASE_LIST = "dca_purchase_list"
     ARBITRAGE_OPPORTUNITIES = "arbitrage_opportunities"
     SELECTED_STRATEGY = "selected_strategy"
+    MC_FAN_CHART = "monte_carlo_fan_chart"
 
 
 @dataclass

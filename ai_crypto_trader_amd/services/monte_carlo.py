@@ -242,6 +242,52 @@ class MonteCarloService(Service):
             report[scen] = stats
         return report
 
+    def fan_chart_data(self, syms: list[str], scenario: str = "base",
+                       n_paths: int = 512) -> dict | None:
+        """Per-step percentile bands of the portfolio value — the data
+        behind the reference's matplotlib fan chart (:396-490), served as
+        JSON for the dashboard. Small-n run on the numpy twin recording
+        V_t per step."""
+        if not all(len(self.prices.get(s, [])) >= 64 for s in syms):
+            return None
+        mu, sigma, chol = self._estimate_params(syms)
+        m_mu, m_sig = self.config.monte_carlo.scenarios.get(
+            scenario, (1.0, 1.0))
+        mu = mu * m_mu
+        sigma = np.maximum(sigma * m_sig, 1e-4)
+        A = len(syms)
+        pad = (-A) % 4
+        w = np.full(A, 1.0 / A)
+        if pad:
+            mu = np.concatenate([mu, np.zeros(pad)])
+            sigma = np.concatenate([sigma, np.full(pad, 1e-4)])
+            w = np.concatenate([w, np.zeros(pad)])
+            c2 = np.eye(A + pad)
+            c2[:A, :A] = chol
+            chol = c2
+        days = self.config.monte_carlo.time_horizon_days
+        from ..ops.montecarlo import _gbm_terms, philox_normal4_np
+        cvol_k, drift = _gbm_terms(chol, mu, sigma, 1.0 / 365.0)
+        Af = len(mu)
+        logS = np.zeros((n_paths, Af), np.float32)
+        paths = np.arange(n_paths, dtype=np.uint64)
+        bands = {p: [1.0] for p in (5, 25, 50, 75, 95)}
+        for step in range(days):
+            for k4 in range(Af // 4):
+                ctr = np.full(n_paths,
+                              (np.uint64(step) << np.uint64(32))
+                              | np.uint64(k4), np.uint64)
+                z4 = philox_normal4_np(11, paths, ctr)
+                for dz in range(4):
+                    logS += np.outer(z4[:, dz], cvol_k[k4 * 4 + dz])
+            logS += drift
+            V = (w * np.exp2(logS)).sum(axis=1)
+            for p in bands:
+                bands[p].append(float(np.percentile(V, p)))
+        return {"scenario": scenario, "symbols": syms,
+                "horizon_days": days, "n_paths": n_paths,
+                "percentiles": {str(p): v for p, v in bands.items()}}
+
     async def run_portfolio_mc(self) -> dict | None:
         syms = sorted(s for s, h in self.prices.items() if len(h) >= 64)
         if not syms:
@@ -258,6 +304,12 @@ class MonteCarloService(Service):
             for scen in self.config.monte_carlo.scenarios:
                 report[scen] = self.simulate(syms, scen)
         await self.bus.set(Keys.MONTE_CARLO_RESULTS, report)
+        try:
+            fan = self.fan_chart_data(syms)
+            if fan:
+                await self.bus.set(Keys.MC_FAN_CHART, fan)
+        except Exception as e:
+            self.log.debug("fan chart skipped: %r", e)
         await self.bus.set(Keys.MONTE_CARLO_LATEST_REPORT, {
             "generated_at": time.time(),
             "scenarios": list(report),

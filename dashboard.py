@@ -112,7 +112,10 @@ def build_app(bus, store: DataStore):
 
     @app.get("/api/monte_carlo")
     async def monte_carlo():
-        return await bus.get_json(Keys.MONTE_CARLO_RESULTS)
+        return {
+            "results": await bus.get_json(Keys.MONTE_CARLO_RESULTS),
+            "fan_chart": await bus.get_json(Keys.MC_FAN_CHART),
+        }
 
     @app.get("/api/predictions")
     async def predictions():

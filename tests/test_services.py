@@ -462,3 +462,27 @@ def test_mc_historical_method_cpu():
         assert bear["mean"] < out["mean"]
 
     asyncio.run(go())
+
+
+def test_mc_fan_chart_data():
+    """Fan-chart percentile bands (reference matplotlib fan :396-490,
+    served as data): bands widen over the horizon and stay ordered."""
+    import numpy as np
+
+    from ai_crypto_trader_amd.services.monte_carlo import MonteCarloService
+
+    svc = MonteCarloService(InProcessBus(), AppConfig())
+    rng = np.random.default_rng(2)
+    for s in ("AUSDC", "BUSDC", "CUSDC", "DUSDC"):
+        steps = 0.0003 + 0.002 * rng.standard_normal(256)
+        svc.prices[s] = list(np.exp(np.cumsum(steps)))
+    fan = svc.fan_chart_data(["AUSDC", "BUSDC", "CUSDC", "DUSDC"],
+                             n_paths=256)
+    assert fan is not None
+    p = fan["percentiles"]
+    days = fan["horizon_days"]
+    assert all(len(v) == days + 1 for v in p.values())
+    for t in range(days + 1):
+        assert p["5"][t] <= p["25"][t] <= p["50"][t] \
+            <= p["75"][t] <= p["95"][t]
+    assert (p["95"][-1] - p["5"][-1]) > (p["95"][1] - p["5"][1])
