@@ -349,9 +349,23 @@ def mc_paths_sharded(
             q * n, device=fv.device)))
         return float(edges[min(idx + 1, n_bins)])
 
+    def cvar(q):
+        """E[v0 - V | V <= quantile(q)] from the all-reduced histogram
+        (bin midpoints weighted by counts below the q-quantile)."""
+        idx = int(torch.searchsorted(cum, torch.tensor(
+            q * n, device=fv.device)))
+        idx = max(idx, 1)
+        mids = (edges[:-1] + edges[1:]) / 2
+        w_tail = hist[:idx]
+        m = float(w_tail.sum())
+        if m <= 0:
+            return v0 - quantile(q)
+        return v0 - float((mids[:idx] * w_tail).sum()) / m
+
     stats = {
         "mean": mean, "std": var ** 0.5,
         "var_95": v0 - quantile(0.05), "var_99": v0 - quantile(0.01),
+        "cvar_95": cvar(0.05), "cvar_99": cvar(0.01),
         "p5": quantile(0.05), "p50": quantile(0.5), "p95": quantile(0.95),
         "max_drawdown_mean": float(sums[2]) / n,
         "n_paths": int(n), "world": world,

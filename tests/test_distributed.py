@@ -141,6 +141,8 @@ def test_mc_sharded_matches_single():
         assert abs(s["mean"] - ref["mean"]) < 1e-5
         assert abs(s["std"] - ref["std"]) < 1e-5
         assert abs(s["var_95"] - ref["var_95"]) < 2e-3   # histogram binning
+        assert abs(s["cvar_95"] - ref["cvar_95"]) < 2e-3
+        assert s["cvar_95"] >= s["var_95"] - 2e-3        # tail mean >= VaR
     assert results[0] == results[1]
 
 
