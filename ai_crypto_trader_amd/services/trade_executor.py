@@ -16,6 +16,20 @@ from ..utils.exchange import ExchangeInterface
 from .base import Service
 
 
+def round_to_filters(qty: float, price: float, filters: dict):
+    """Tick/step rounding (reference trade_executor_service.py:789-797):
+    qty floored to step_size, price floored to tick_size; returns
+    (qty, price, ok) with ok=False below min notional."""
+    step = filters.get("step_size", 0.0) or 0.0
+    tick = filters.get("tick_size", 0.0) or 0.0
+    if step > 0:
+        qty = int(qty / step) * step
+    if tick > 0:
+        price = int(price / tick) * tick
+    ok = qty * price >= filters.get("min_notional", 0.0)
+    return qty, price, ok
+
+
 class TrailingStopManager:
     """4 trailing strategies (trade_executor_service.py:168-247):
     percent_based / atr_based / volatility_based / fixed_amount, with an
@@ -125,6 +139,11 @@ class TradeExecutorService(Service):
         if cost < 1e-6:
             return
         qty = cost / price
+        filters = self.exchange.get_symbol_filters(sym) \
+            if hasattr(self.exchange, "get_symbol_filters") else {}
+        qty, _, ok = round_to_filters(qty, price, filters)
+        if not ok or qty <= 0:
+            return
         order = self.exchange.create_order(sym, "BUY", "MARKET", qty)
         if order.status != "FILLED":
             return

@@ -152,6 +152,11 @@ class FakeExchange(ExchangeInterface):
     def get_trading_fees(self, symbol) -> dict:
         return {"maker": self.fee, "taker": self.fee}
 
+    def get_symbol_filters(self, symbol) -> dict:
+        """Exchange trading filters (reference tick/step rounding
+        :789-797): price tick size, quantity step size, min notional."""
+        return {"tick_size": 1e-6, "step_size": 1e-6, "min_notional": 1e-6}
+
     def portfolio_value(self) -> float:
         v = self.balances.get(self.quote, 0.0)
         for asset, qty in self.balances.items():

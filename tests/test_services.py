@@ -486,3 +486,19 @@ def test_mc_fan_chart_data():
         assert p["5"][t] <= p["25"][t] <= p["50"][t] \
             <= p["75"][t] <= p["95"][t]
     assert (p["95"][-1] - p["5"][-1]) > (p["95"][1] - p["5"][1])
+
+
+def test_order_filter_rounding():
+    from ai_crypto_trader_amd.services.trade_executor import (
+        round_to_filters,
+    )
+
+    q, p, ok = round_to_filters(0.123456789, 101.2345678,
+                                {"tick_size": 0.01, "step_size": 0.001,
+                                 "min_notional": 10.0})
+    assert q == pytest.approx(0.123)
+    assert p == pytest.approx(101.23)
+    assert ok
+    _, _, ok2 = round_to_filters(0.01, 100.0, {"step_size": 0.001,
+                                               "min_notional": 10.0})
+    assert not ok2            # 1 USD < min notional
