@@ -52,6 +52,11 @@ class ExchangeInterface(ABC):
     @abstractmethod
     def get_trading_fees(self, symbol: str) -> dict: ...
 
+    def get_symbol_filters(self, symbol: str) -> dict:
+        """Tick/step/min-notional filters; adapters override with the
+        venue's real values (reference exchange filters :789-797)."""
+        return {"tick_size": 0.0, "step_size": 0.0, "min_notional": 0.0}
+
 
 class FakeExchange(ExchangeInterface):
     """Deterministic in-memory exchange driven by set_price() ticks.
