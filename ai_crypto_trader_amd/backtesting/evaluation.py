@@ -37,6 +37,8 @@ def calculate_metrics(equity_curve: np.ndarray,
         "max_drawdown_pct": float(dd.max() * 100),
         "volatility_ann": float(rets.std() * np.sqrt(ANNUAL)),
         "n_candles": len(eq),
+        "daily": period_returns(eq, 1440),
+        "monthly": period_returns(eq, 43_200),
     }
     if trades:
         pnls = [t["pnl"] for t in trades]
@@ -51,6 +53,28 @@ def calculate_metrics(equity_curve: np.ndarray,
             "avg_loss": float(np.mean(losses)) if losses else 0.0,
         })
     return out
+
+
+def period_returns(equity_curve: np.ndarray,
+                   candles_per_period: int = 1440) -> dict:
+    """Per-period (daily at 1440 1m-candles, monthly at 43200) return
+    buckets (strategy_evaluation.py:180-188): the return of each complete
+    period plus best/worst/positive-share summaries."""
+    eq = np.asarray(equity_curve, np.float64)
+    k = max(int(candles_per_period), 1)
+    n = (len(eq) - 1) // k
+    if n < 1:
+        return {"returns_pct": [], "best_pct": 0.0, "worst_pct": 0.0,
+                "positive_share": 0.0, "n_periods": 0}
+    marks = eq[:n * k + 1:k]
+    rets = (marks[1:] / marks[:-1] - 1.0) * 100.0
+    return {
+        "returns_pct": [round(float(r), 6) for r in rets],
+        "best_pct": float(rets.max()),
+        "worst_pct": float(rets.min()),
+        "positive_share": float((rets > 0).mean()),
+        "n_periods": int(n),
+    }
 
 
 def calculate_advanced_metrics(equity_curve: np.ndarray,

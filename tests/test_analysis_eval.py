@@ -274,3 +274,22 @@ def test_ga_checkpoint_resume_deterministic(tmp_path):
         b.step()
 
     np.testing.assert_array_equal(b.pop_t.numpy(), ref.pop_t.numpy())
+
+
+def test_period_return_buckets():
+    """Daily/monthly return buckets (reference :180-188)."""
+    from ai_crypto_trader_amd.backtesting.evaluation import (
+        calculate_metrics, period_returns,
+    )
+
+    # 3 exact days of 0.1%/day compounding at 1-minute resolution
+    daily_r = 0.001
+    per_min = (1 + daily_r) ** (1 / 1440)
+    eq = per_min ** np.arange(3 * 1440 + 1)
+    d = period_returns(eq, 1440)
+    assert d["n_periods"] == 3
+    np.testing.assert_allclose(d["returns_pct"], [0.1] * 3, rtol=1e-5)
+    assert d["positive_share"] == 1.0
+    m = calculate_metrics(eq)
+    assert m["daily"]["n_periods"] == 3
+    assert m["monthly"]["n_periods"] == 0    # < one month of candles
