@@ -211,6 +211,34 @@ def params_from_code(code: str) -> dict:
     return found
 
 
+def validate_strategy(params: dict) -> tuple[bool, list[str]]:
+    """Strategy validator (reference STRATEGY_EVOLUTION.md 'Strategy
+    Validator Service' + ai_strategy_evaluator's min-requirements gate):
+    sanity-checks a proposed parameter set — risk management present,
+    values inside the native bounds, sizes/ratios coherent — before it
+    is allowed to trade or evolve."""
+    from ..backtesting.strategy import PARAM_BOUNDS, PARAM_NAMES
+
+    issues = []
+    bounds = {n: (float(b[0]), float(b[1]))
+              for n, b in zip(PARAM_NAMES, PARAM_BOUNDS)}
+    for name, v in params.items():
+        if name in bounds:
+            lo, hi = bounds[name]
+            if not (lo <= float(v) <= hi):
+                issues.append(f"{name}={v} outside [{lo}, {hi}]")
+    if float(params.get("stop_loss_pct", 0.0)) <= 0:
+        issues.append("no stop loss (risk management required)")
+    tp = float(params.get("take_profit_pct", 0.0))
+    sl = float(params.get("stop_loss_pct", 1.0))
+    if tp > 0 and sl > 0 and tp / sl < 0.5:
+        issues.append("take profit under half the stop "
+                      "(reward:risk < 0.5)")
+    if float(params.get("position_size_pct", 0.0)) > 1.0:
+        issues.append("position size above 100%")
+    return (not issues), issues
+
+
 class MarketRegimeDataCollector:
     """Assembles regime-training datasets from bus history
     (market_regime_data_collector.py:44-395: price/signal/outcome history

@@ -293,3 +293,20 @@ def test_period_return_buckets():
     m = calculate_metrics(eq)
     assert m["daily"]["n_periods"] == 3
     assert m["monthly"]["n_periods"] == 0    # < one month of candles
+
+
+def test_strategy_validator():
+    from ai_crypto_trader_amd.backtesting.strategy import (
+        DEFAULT_PARAMS, params_to_dict,
+    )
+    from ai_crypto_trader_amd.services.strategy_evaluator import (
+        validate_strategy,
+    )
+
+    ok, issues = validate_strategy(params_to_dict(DEFAULT_PARAMS))
+    assert ok, issues
+    bad = params_to_dict(DEFAULT_PARAMS)
+    bad["stop_loss_pct"] = 0.0
+    bad["rsi_oversold"] = 500.0
+    ok2, issues2 = validate_strategy(bad)
+    assert not ok2 and len(issues2) >= 2
