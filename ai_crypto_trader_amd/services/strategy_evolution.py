@@ -187,11 +187,16 @@ class StrategyEvolutionService(Service):
             from .strategy_evaluator import params_from_code
             out = params_from_code(out)
         from ..backtesting.strategy import dict_to_params
+        from .strategy_evaluator import validate_strategy
         base = params_to_dict(self.current_params)
         base.update(out or {})
+        ok, issues = validate_strategy(base)   # validator gate (reference
+        if not ok:                             # STRATEGY_EVOLUTION.md)
+            self.log.warning("llm proposal issues (clipped): %s", issues)
         vec = clip_params(dict_to_params(base)[None])[0]
-        return vec, {"llm": True, **{k: v for k, v in perf.items()
-                                     if k in ("sharpe", "win_rate")}}
+        return vec, {"llm": True, "validation_issues": issues,
+                     **{k: v for k, v in perf.items()
+                        if k in ("sharpe", "win_rate")}}
 
     def adjust_for_regime(self, params: np.ndarray, regime: str):
         """per-regime multiplier tables (reference :302, :145-174)."""
