@@ -24,6 +24,7 @@ class TradingConfig(BaseModel):
     fee_rate: float = 0.001
     max_positions: int = 5
     min_confidence: float = 0.7
+    min_trade_usd: float = 40.0               # trading_strategy.md minimum
 
 
 class RiskConfig(BaseModel):
@@ -36,6 +37,7 @@ class RiskConfig(BaseModel):
     adaptive_stop_vol_factor_max: float = 2.0
     base_stop_loss_pct: float = 0.02
     correlation_threshold: float = 0.7
+    max_daily_drawdown_pct: float = 0.06      # trading_strategy.md halt
 
 
 class MonteCarloConfig(BaseModel):

@@ -95,14 +95,41 @@ class TradeExecutorService(Service):
         self.active: dict[str, dict] = {}     # symbol -> trade record
         self.strategy_params: dict = {}
         self.trades_done = 0
+        self.day_start_value: float | None = None
+        self.day_start_at = 0.0
 
     def run_tasks(self):
         return [self._consume_signals(), self._monitor_trades()]
+
+    def _portfolio_value(self) -> float:
+        balances = self.exchange.get_balances()
+        total = balances.get(self.config.trading.quote_asset, 0.0)
+        for asset, qty in balances.items():
+            if asset != self.config.trading.quote_asset:
+                total += qty * self.exchange.get_ticker(
+                    asset + self.config.trading.quote_asset)["price"]
+        return total
+
+    def daily_drawdown_exceeded(self) -> bool:
+        """Max-daily-drawdown halt (trading_strategy.md: 6%): stop
+        opening positions when today's portfolio drawdown from the day
+        anchor passes the configured limit."""
+        now = time.time()
+        value = self._portfolio_value()
+        if self.day_start_value is None or now - self.day_start_at > 86_400:
+            self.day_start_value = value
+            self.day_start_at = now
+            return False
+        self.day_start_value = max(self.day_start_value, value)
+        dd = 1.0 - value / max(self.day_start_value, 1e-9)
+        return dd > self.config.risk.max_daily_drawdown_pct
 
     # --- gates (trade_executor_service.py:719-787) -----------------------
     async def check_trading_conditions(self, signal: dict) -> tuple[bool, str]:
         if len(self.active) >= self.config.trading.max_positions:
             return False, "max_positions"
+        if self.daily_drawdown_exceeded():
+            return False, "daily_drawdown_halt"
         if signal["symbol"] in self.active:
             return False, "already_in_position"
         risk = await self.bus.get_json(Keys.PORTFOLIO_RISK)
@@ -136,8 +163,8 @@ class TradeExecutorService(Service):
             adj = json.loads(sra)
             pos_pct *= adj.get("position_multiplier", 1.0)
         cost = min(cash * pos_pct, cash)
-        if cost < 1e-6:
-            return
+        if cost < self.config.trading.min_trade_usd:
+            return      # minimum trade amount (trading_strategy.md)
         qty = cost / price
         filters = self.exchange.get_symbol_filters(sym) \
             if hasattr(self.exchange, "get_symbol_filters") else {}

@@ -502,3 +502,32 @@ def test_order_filter_rounding():
     _, _, ok2 = round_to_filters(0.01, 100.0, {"step_size": 0.001,
                                                "min_notional": 10.0})
     assert not ok2            # 1 USD < min notional
+
+
+def test_daily_drawdown_halt():
+    """BUYs stop when today's portfolio drawdown exceeds the configured
+    max (trading_strategy.md: 6%)."""
+    from ai_crypto_trader_amd.services.trade_executor import (
+        TradeExecutorService,
+    )
+
+    async def go():
+        cfg = AppConfig()
+        bus = InProcessBus()
+        ex = FakeExchange()
+        ex.set_price("BTCUSDC", 100.0)
+        svc = TradeExecutorService(bus, ex, cfg)
+        ok, _ = await svc.check_trading_conditions({"symbol": "BTCUSDC"})
+        assert ok                                 # anchors the day value
+        await svc.execute_buy({"symbol": "BTCUSDC", "confidence": 0.9})
+        assert "BTCUSDC" in svc.active
+        # a 2% gap-down is absorbed by the protective stop: no halt
+        ex.set_price("BTCUSDC", 97.0)
+        ok_mid, _ = await svc.check_trading_conditions({"symbol": "XUSDC"})
+        assert ok_mid
+        # a loss the stops could NOT cap (e.g. fees/slippage cascade)
+        ex.balances["USDC"] -= 1500.0             # -15% of the day anchor
+        ok2, why = await svc.check_trading_conditions({"symbol": "ETHUSDC"})
+        assert not ok2 and why == "daily_drawdown_halt"
+
+    asyncio.run(go())
