@@ -64,6 +64,8 @@ class ServiceSupervisor(Service):
     async def run(self):
         while self.running:
             for s in self.services:
+                self.metrics.service_health.labels(s.name).set(
+                    1.0 if s.healthy else 0.0)
                 if self._needs_restart(s):
                     try:
                         await self.restart(s)

@@ -59,6 +59,10 @@ class Metrics:
         self.portfolio_value = Gauge(
             "portfolio_value_usd", "Portfolio value", registry=r)
         self.equity = Gauge("equity", "Account equity", registry=r)
+        self.service_health = Gauge(
+            "crypto_trader_service_health",
+            "1 = service healthy (reference PRODUCTION_READINESS metric)",
+            ["service_name"], registry=r)
         self.active_trades = Gauge(
             "active_trades", "Open positions", registry=r)
         self.ai_confidence = Gauge(
