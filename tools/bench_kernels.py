@@ -86,6 +86,22 @@ def bench_mc(reps):
             "config": {"A": A, "steps": steps, "paths": paths}}
 
 
+def bench_mc_bootstrap(reps):
+    from ai_crypto_trader_amd.ops.montecarlo import mc_bootstrap_gpu
+
+    rng = np.random.default_rng(3)
+    T_hist, A = 365, 64
+    lr = (rng.standard_normal((T_hist, A)) * 0.01).astype(np.float32)
+    w = np.full(A, 1.0 / A)
+    n_paths, steps = 10_000_000, 30
+    dt = timed(lambda: mc_bootstrap_gpu(lr, w, n_steps=steps,
+                                        n_paths=n_paths, seed=1), reps)
+    return {"kernel": "mc_bootstrap", "ms": dt * 1e3,
+            "paths_per_sec": n_paths / dt,
+            "config": {"paths": n_paths, "assets": A, "steps": steps,
+                       "t_hist": T_hist}}
+
+
 def bench_cov(reps):
     from ai_crypto_trader_amd.ops.covar import cov_gpu
 
@@ -204,7 +220,7 @@ def bench_gae(reps):
 
 
 BENCHES = {
-    "backtest": bench_backtest, "mc": bench_mc, "cov": bench_cov,
+    "backtest": bench_backtest, "mc": bench_mc, "mc_bootstrap": bench_mc_bootstrap, "cov": bench_cov,
     "indicators": bench_indicators, "lstm": bench_lstm, "gru": bench_gru, "env": bench_env,
     "gae": bench_gae, "attn": bench_attn,
 }
