@@ -163,3 +163,18 @@ def test_interval_aware_sharpe(tmp_path):
     assert s1m == pytest.approx(s1h * np.sqrt(525_600 / 8_760), rel=1e-6)
     # default matches the kernel's 1m convention closely
     assert s1m > 0
+
+
+def test_multi_interval_backtest(tmp_path):
+    """Backtests run at non-1m intervals end-to-end with interval-aware
+    annualization."""
+    eng = BacktestEngine(str(tmp_path / "d"), device="cpu")
+    s1h = eng.run_backtest("BTCUSDC", "momentum", interval="1h",
+                           n_candles=1500)
+    assert s1h["interval"] == "1h" and s1h["n_candles"] == 1500
+    assert np.isfinite(s1h["sharpe"])
+    s1m = eng.run_backtest("BTCUSDC", "momentum", interval="1m",
+                           n_candles=1500)
+    # different interval data => independent results, both well-formed
+    assert s1m["interval"] == "1m"
+    assert np.isfinite(s1m["sharpe"])
