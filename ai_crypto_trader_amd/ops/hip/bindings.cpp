@@ -44,7 +44,7 @@ extern "C" void launch_mfma_gemm_test_bf16(const void*, const void*, float*,
                                            int, int, int, hipStream_t);
 extern "C" void launch_gru_seq_fwd(const void*, const void*, const float*,
                                    void*, void*, float*, int, int, int,
-                                   hipStream_t);
+                                   int, hipStream_t);
 extern "C" void launch_gru_seq_bwd(const float*, const void*, const float*,
                                    const void*, const void*, void*, int, int,
                                    int, hipStream_t);
@@ -262,14 +262,14 @@ PYBIND11_MODULE(_hip_ops, m) {
     m.def("gru_seq_fwd",
           [](uintptr_t xproj, uintptr_t Wt, uintptr_t bias, uintptr_t h_out,
              uintptr_t gates_out, uintptr_t hpn_out, int B, int T, int H,
-             uintptr_t stream) {
+             int save_mode, uintptr_t stream) {
               launch_gru_seq_fwd(reinterpret_cast<const void*>(xproj),
                                  reinterpret_cast<const void*>(Wt),
                                  reinterpret_cast<const float*>(bias),
                                  reinterpret_cast<void*>(h_out),
                                  reinterpret_cast<void*>(gates_out),
                                  reinterpret_cast<float*>(hpn_out), B, T, H,
-                                 as_stream(stream));
+                                 save_mode, as_stream(stream));
               check(hipGetLastError(), "gru_seq_fwd launch");
           });
 

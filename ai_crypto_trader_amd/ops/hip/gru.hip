@@ -39,7 +39,7 @@ __global__ void __launch_bounds__((H / 16) * 64) gru_seq_fwd_kernel(
     bf16* __restrict__ h_out,          // (T, B, H)
     bf16* __restrict__ gates_out,      // (T, B, 3H)  post-act r, z, n
     float* __restrict__ hpn_out,       // (T, B, H)   hp_n (pre-r-multiply)
-    int B, int T)
+    int B, int T, int save_mode)       // 0 = inference: skip bwd saves
 {
     constexpr int NW = H / 16;
     constexpr int THREEH = 3 * H;
@@ -121,11 +121,13 @@ __global__ void __launch_bounds__((H / 16) * 64) gru_seq_fwd_kernel(
                 hprev[mt][r] = h;
                 const bf16 hb = (bf16)h;
                 lds_h[row * HP + ncol] = hb;
-                const long gb = (base_tb + row) * THREEH;
-                gates_out[gb + 0 * H + ncol] = (bf16)gr;
-                gates_out[gb + 1 * H + ncol] = (bf16)gz;
-                gates_out[gb + 2 * H + ncol] = (bf16)gn;
-                hpn_out[(base_tb + row) * H + ncol] = hpn;
+                if (save_mode) {       // training: save for backward
+                    const long gb = (base_tb + row) * THREEH;
+                    gates_out[gb + 0 * H + ncol] = (bf16)gr;
+                    gates_out[gb + 1 * H + ncol] = (bf16)gz;
+                    gates_out[gb + 2 * H + ncol] = (bf16)gn;
+                    hpn_out[(base_tb + row) * H + ncol] = hpn;
+                }
                 h_out[(base_tb + row) * H + ncol] = hb;
             }
         }
@@ -238,7 +240,8 @@ __global__ void __launch_bounds__((H / 16) * 64) gru_seq_bwd_kernel(
 extern "C" void launch_gru_seq_fwd(const void* xproj, const void* Wt,
                                    const float* bias, void* h_out,
                                    void* gates_out, float* hpn_out, int B,
-                                   int T, int H, hipStream_t stream) {
+                                   int T, int H, int save_mode,
+                                   hipStream_t stream) {
     if (B % GRU_BM != 0)
         throw std::runtime_error("gru_fwd: B must be a multiple of 64");
     dim3 grid(B / GRU_BM);
@@ -246,12 +249,12 @@ extern "C" void launch_gru_seq_fwd(const void* xproj, const void* Wt,
         hipLaunchKernelGGL(gru_seq_fwd_kernel<64>, grid, dim3(256), 0,
                            stream, (const bf16*)xproj, (const bf16*)Wt,
                            bias, (bf16*)h_out, (bf16*)gates_out, hpn_out,
-                           B, T);
+                           B, T, save_mode);
     } else if (H == 32) {
         hipLaunchKernelGGL(gru_seq_fwd_kernel<32>, grid, dim3(128), 0,
                            stream, (const bf16*)xproj, (const bf16*)Wt,
                            bias, (bf16*)h_out, (bf16*)gates_out, hpn_out,
-                           B, T);
+                           B, T, save_mode);
     } else {
         throw std::runtime_error("gru_fwd: H must be 32 or 64");
     }

@@ -685,3 +685,20 @@ def test_mc_bootstrap_matches_cpu(dev):
     # sanity: mean final value ~ exp of n_steps * mean log-return
     expect = float(np.exp(n_steps * lr.mean() * A / A))
     assert abs(float(fv.mean()) - expect) < 0.05
+
+
+def test_gru_infer_matches_training_forward(dev):
+    """The no-save GRU inference forward produces the same hidden states
+    as the training forward (mirrors the LSTM serving path)."""
+    from ai_crypto_trader_amd.models.gru import FusedGRULayer
+
+    torch.manual_seed(3)
+    layer = FusedGRULayer(9, 64).to(dev)
+    x = torch.randn(24, 128, 9, device=dev)
+    with torch.enable_grad():
+        h_train = layer(x)
+    with torch.no_grad():
+        h_infer = layer(x)
+    torch.cuda.synchronize()
+    np.testing.assert_array_equal(h_train.detach().cpu().float().numpy(),
+                                  h_infer.cpu().float().numpy())
