@@ -90,7 +90,10 @@ def metrics_to_stats(m: np.ndarray, T: int, interval: str = "1m") -> dict:
             if d["final_equity"] > 0 else -100.0,
         "n_trades": int(d["n_trades"]),
         "win_rate": d["wins"] / max(d["n_trades"], 1.0),
-        "profit_factor": gp / gl if gl > 0 else float("inf"),
+        # capped: float('inf') serializes as the non-standard 'Infinity'
+        # token that strict JSON parsers (and dashboards) reject
+        "profit_factor": min(gp / gl, 1e9) if gl > 0 else (
+            1e9 if gp > 0 else 0.0),
         "max_drawdown_pct": d["max_drawdown"] * 100.0,
         "sharpe": d["sharpe"],
         "fitness": d["fitness"],

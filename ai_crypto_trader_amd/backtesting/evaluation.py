@@ -85,16 +85,22 @@ def calculate_advanced_metrics(equity_curve: np.ndarray,
     peak = np.maximum.accumulate(eq)
     mdd = float(((peak - eq) / peak).max())
     years = len(eq) / ANNUAL
-    ann_ret = (eq[-1] / eq[0]) ** (1 / max(years, 1e-9)) - 1
+    # log-space exponentiation: the naive ratio**(1/years) overflows to
+    # inf for short curves (years ~ 1e-6); clamp the annualized log-return
+    # so calmar stays finite and JSON-serializable
+    ratio = max(eq[-1] / eq[0], 1e-12)
+    ann_ret = float(np.exp(np.clip(np.log(ratio) / max(years, 1e-9),
+                                   -50.0, 50.0))) - 1
     downside = rets[rets < 0]
-    sortino = float(rets.mean() / (downside.std() + 1e-12)
-                    * np.sqrt(ANNUAL)) if len(downside) else float("inf")
+    # finite caps instead of inf: keeps every metric JSON-serializable
+    sortino = min(float(rets.mean() / (downside.std() + 1e-12)
+                        * np.sqrt(ANNUAL)), 1e9) if len(downside) else 1e9
     out = {
-        "calmar": float(ann_ret / mdd) if mdd > 0 else float("inf"),
+        "calmar": min(float(ann_ret / mdd), 1e9) if mdd > 0 else 1e9,
         "sortino": sortino,
         "recovery_factor":
-            float((eq[-1] - eq[0]) / (mdd * eq[0])) if mdd > 0
-            else float("inf"),
+            min(float((eq[-1] - eq[0]) / (mdd * eq[0])), 1e9) if mdd > 0
+            else 1e9,
     }
     if trades:
         pnls = np.asarray([t["pnl"] for t in trades])

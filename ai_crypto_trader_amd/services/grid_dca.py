@@ -75,7 +75,7 @@ class GridTradingStrategy(Service):
         re-arm the grid (:418-509)."""
         fills = []
         for oid, lv in list(self._orders.items()):
-            o = self.exchange.orders.get(oid)
+            o = self.exchange.get_order(self.symbol, oid)
             if o is None:
                 self._orders.pop(oid, None)
                 continue

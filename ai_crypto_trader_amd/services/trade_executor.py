@@ -10,22 +10,33 @@ from __future__ import annotations
 
 import json
 import time
+from decimal import Decimal
 
 from ..bus.schema import Channels, Keys
 from ..utils.exchange import ExchangeInterface
 from .base import Service
 
 
+def _quantize_down(value: float, step: float) -> float:
+    """Floor `value` to a multiple of `step` in decimal arithmetic —
+    binary-float division (0.3/0.1 == 2.999...) would floor one step low
+    and carry residue a live venue's filters reject."""
+    v = Decimal(str(value))
+    s = Decimal(str(step))
+    return float((v // s) * s)
+
+
 def round_to_filters(qty: float, price: float, filters: dict):
     """Tick/step rounding (reference trade_executor_service.py:789-797):
-    qty floored to step_size, price floored to tick_size; returns
+    qty floored to step_size, price floored to tick_size (Decimal
+    quantization, as live-venue filter code must); returns
     (qty, price, ok) with ok=False below min notional."""
     step = filters.get("step_size", 0.0) or 0.0
     tick = filters.get("tick_size", 0.0) or 0.0
     if step > 0:
-        qty = int(qty / step) * step
+        qty = _quantize_down(qty, step)
     if tick > 0:
-        price = int(price / tick) * tick
+        price = _quantize_down(price, tick)
     ok = qty * price >= filters.get("min_notional", 0.0)
     return qty, price, ok
 
@@ -112,15 +123,16 @@ class TradeExecutorService(Service):
 
     def daily_drawdown_exceeded(self) -> bool:
         """Max-daily-drawdown halt (trading_strategy.md: 6%): stop
-        opening positions when today's portfolio drawdown from the day
-        anchor passes the configured limit."""
+        opening positions when today's portfolio drawdown from the
+        day-open anchor passes the configured limit. The anchor is the
+        value at the start of the 24h window (NOT the intraday peak —
+        day-anchor semantics per trading_strategy.md)."""
         now = time.time()
         value = self._portfolio_value()
         if self.day_start_value is None or now - self.day_start_at > 86_400:
             self.day_start_value = value
             self.day_start_at = now
             return False
-        self.day_start_value = max(self.day_start_value, value)
         dd = 1.0 - value / max(self.day_start_value, 1e-9)
         return dd > self.config.risk.max_daily_drawdown_pct
 
@@ -209,7 +221,7 @@ class TradeExecutorService(Service):
         # cancel the resting protective stop first (reference :333-371:
         # order replacement discipline — never leave an orphaned stop)
         oid = trade.get("stop_order_id")
-        stop_o = self.exchange.orders.get(oid) if oid else None
+        stop_o = self.exchange.get_order(sym, oid) if oid else None
         if stop_o is not None and stop_o.status == "FILLED":
             # the exchange-side stop already closed the position
             price = stop_o.filled_price

@@ -47,6 +47,13 @@ class ExchangeInterface(ABC):
     def cancel_order(self, symbol: str, order_id: str) -> bool: ...
 
     @abstractmethod
+    def get_order(self, symbol: str, order_id: str) -> Order | None:
+        """Order status lookup (reference BinanceExchange get_order,
+        exchange_interface.py:67-207) — callers must use this instead of
+        touching adapter internals."""
+        ...
+
+    @abstractmethod
     def get_balances(self) -> dict[str, float]: ...
 
     @abstractmethod
@@ -150,6 +157,9 @@ class FakeExchange(ExchangeInterface):
             o.status = "CANCELED"
             return True
         return False
+
+    def get_order(self, symbol, order_id) -> Order | None:
+        return self.orders.get(order_id)
 
     def get_balances(self) -> dict[str, float]:
         return dict(self.balances)
