@@ -18,7 +18,7 @@ from __future__ import annotations
 
 import numpy as np
 
-from .strategy import FEE, MAX_WIN, NPARAM, WARMUP
+from .strategy import FEE, MAX_WIN, NPARAM, RESNAP, WARMUP
 
 NMETRIC = 10
 METRIC_NAMES = [
@@ -70,6 +70,7 @@ def run_backtest_cpu(
     *,
     initial_equity: float = 1.0,
     record_equity: bool = False,
+    record_net: bool = False,
 ):
     """Run every param-set against every symbol.
 
@@ -140,6 +141,7 @@ def run_backtest_cpu(
     sum_ret2 = np.zeros(L, f32)
 
     curves = np.empty((L, T), f32) if record_equity else None
+    nets = np.empty((L, T), np.int32) if record_net else None
 
     close_all = np.ascontiguousarray(candles[:, :, 0])   # (nsym, T)
     high_all = np.ascontiguousarray(candles[:, :, 1])
@@ -175,10 +177,25 @@ def run_backtest_cpu(
         rsi_den = avg_gain + np.maximum(avg_loss, EPS)
 
         ridx = t % bb_w                                   # per-lane ring slot
-        old = ring[lanes, ridx].astype(np.float64)
-        c64 = close.astype(np.float64)
-        bb_sum += c64 - old
-        bb_sum2 += c64 * c64 - old * old
+        if t > 0 and t % RESNAP == 0:
+            # drift-free Bollinger resnap (strategy.py RESNAP): recompute
+            # the window sums directly, oldest->newest, instead of the
+            # incremental +new-old update. Keeps the BB state exactly
+            # restartable at aligned boundaries (time-parallel kernel).
+            acc = np.zeros(L, np.float64)
+            acc2 = np.zeros(L, np.float64)
+            for j in range(MAX_WIN - 1, -1, -1):
+                cj = close_all[sym_idx, t - j].astype(np.float64)
+                m = j < bb_w
+                acc = acc + np.where(m, cj, 0.0)
+                acc2 = acc2 + np.where(m, cj * cj, 0.0)
+            bb_sum = acc
+            bb_sum2 = acc2
+        else:
+            old = ring[lanes, ridx].astype(np.float64)
+            c64 = close.astype(np.float64)
+            bb_sum += c64 - old
+            bb_sum2 += c64 * c64 - old * old
         ring[lanes, ridx] = close
         inv_cnt = np.where(t + 1 < bb_w, 1.0 / (t + 1.0), inv_w)
         mean64 = bb_sum * inv_cnt
@@ -222,6 +239,8 @@ def run_backtest_cpu(
             net = buy - sell
         else:
             net = np.zeros(L, np.int32)
+        if record_net:
+            nets[:, t] = net
 
         # --- 3. position management ---------------------------------------
         pos = in_pos
@@ -275,9 +294,122 @@ def run_backtest_cpu(
     metrics = finalize_metrics(
         T, equity, n_trades, wins, gross_p, gross_l, max_dd, sum_ret, sum_ret2
     ).reshape(P, nsym, NMETRIC)
+    extras = []
     if record_equity:
-        return metrics, curves.reshape(P, nsym, T)
+        extras.append(curves.reshape(P, nsym, T))
+    if record_net:
+        extras.append(nets.reshape(P, nsym, T))
+    if extras:
+        return (metrics, *extras)
     return metrics
+
+
+def run_trades_from_flags_cpu(
+    candles: np.ndarray,       # (nsym, T, 4) f32
+    population: np.ndarray,    # (P, NPARAM) f32
+    entry_flags: np.ndarray,   # (P, nsym, T) bool  (net >= entry_votes, t>=WARMUP)
+    exit_flags: np.ndarray,    # (P, nsym, T) bool  (net <= -exit_votes)
+    *,
+    initial_equity: float = 1.0,
+) -> np.ndarray:
+    """Numpy twin of the bt_trades phase (ops/hip/backtest_tp.hip):
+    the exact position state machine + equity accounting driven by
+    precomputed vote flags. Identical op order to run_backtest_cpu
+    sections 3-4, so given flags extracted from run_backtest_cpu
+    (record_net=True) the metrics are bitwise equal — the decomposition
+    property the time-parallel GPU path rests on
+    (tests/test_backtest_cpu.py::test_flag_decomposition_bitwise)."""
+    f32 = np.float32
+    candles = np.asarray(candles, dtype=f32)
+    population = np.asarray(population, dtype=f32)
+    nsym, T, _ = candles.shape
+    P = population.shape[0]
+    L = P * nsym
+
+    par = np.repeat(population, nsym, axis=0)
+    size_pct, sl_pct, tp_pct = par[:, 12], par[:, 13], par[:, 14]
+    trail_pct, trail_act = par[:, 15], par[:, 16]
+
+    eflags = entry_flags.reshape(L, T)
+    xflags = exit_flags.reshape(L, T)
+
+    cash = np.full(L, initial_equity, f32)
+    units = np.zeros(L, f32)
+    in_pos = np.zeros(L, bool)
+    entry_cost = np.zeros(L, f32)
+    entry_price = np.zeros(L, f32)
+    stop = np.zeros(L, f32)
+    tp = np.zeros(L, f32)
+    peak = np.zeros(L, f32)
+    equity = np.full(L, initial_equity, f32)
+    max_eq = np.full(L, initial_equity, f32)
+    max_dd = np.zeros(L, f32)
+    n_trades = np.zeros(L, f32)
+    wins = np.zeros(L, f32)
+    gross_p = np.zeros(L, f32)
+    gross_l = np.zeros(L, f32)
+    sum_ret = np.zeros(L, f32)
+    sum_ret2 = np.zeros(L, f32)
+
+    close_all = np.ascontiguousarray(candles[:, :, 0])
+    high_all = np.ascontiguousarray(candles[:, :, 1])
+    low_all = np.ascontiguousarray(candles[:, :, 2])
+    sym_idx = np.tile(np.arange(nsym), P)
+
+    for t in range(T):
+        close = close_all[sym_idx, t]
+        high = high_all[sym_idx, t]
+        low = low_all[sym_idx, t]
+
+        pos = in_pos
+        peak = np.where(pos, np.maximum(peak, high), peak)
+        trail_on = pos & (trail_pct > 0) & (
+            peak >= entry_price * (f32(1.0) + trail_act)
+        )
+        stop = np.where(
+            trail_on, np.maximum(stop, peak * (f32(1.0) - trail_pct)), stop
+        )
+
+        hit_sl = pos & (low <= stop)
+        hit_tp = pos & ~hit_sl & (high >= tp)
+        hit_sig = pos & ~hit_sl & ~hit_tp & xflags[:, t]
+        exiting = hit_sl | hit_tp | hit_sig
+        exit_price = np.where(hit_sl, stop, np.where(hit_tp, tp, close))
+
+        proceeds = units * exit_price * (f32(1.0) - f32(FEE))
+        pnl = proceeds - entry_cost
+        cash = np.where(exiting, cash + proceeds, cash)
+        n_trades += exiting
+        wins += exiting & (pnl > 0)
+        gross_p += np.where(exiting, np.maximum(pnl, f32(0.0)), f32(0.0))
+        gross_l += np.where(exiting, np.maximum(-pnl, f32(0.0)), f32(0.0))
+        units = np.where(exiting, f32(0.0), units)
+        in_pos = pos & ~exiting
+
+        entering = (~pos) & eflags[:, t]
+        cost = np.minimum(size_pct * equity, cash)
+        new_units = cost * (f32(1.0) - f32(FEE)) / close
+        cash = np.where(entering, cash - cost, cash)
+        units = np.where(entering, new_units, units)
+        entry_cost = np.where(entering, cost, entry_cost)
+        entry_price = np.where(entering, close, entry_price)
+        stop = np.where(entering, close * (f32(1.0) - sl_pct), stop)
+        tp = np.where(entering, close * (f32(1.0) + tp_pct), tp)
+        peak = np.where(entering, close, peak)
+        in_pos = in_pos | entering
+
+        new_eq = cash + units * close
+        r = new_eq / equity - f32(1.0)
+        sum_ret += r
+        sum_ret2 += r * r
+        equity = new_eq
+        max_eq = np.maximum(max_eq, equity)
+        max_dd = np.maximum(max_dd, (max_eq - equity) / max_eq)
+
+    return finalize_metrics(
+        T, equity, n_trades, wins, gross_p, gross_l, max_dd, sum_ret,
+        sum_ret2
+    ).reshape(P, nsym, NMETRIC)
 
 
 def finalize_metrics(

@@ -41,6 +41,7 @@ class GAEngine:
         mut_rate: float = 0.15,
         mut_scale: float = 0.1,
         segments: int | str = 1,
+        continuous: bool = False,
     ):
         """segments > 1 splits each symbol's history into `segments`
         independent backtest segments (a pure reshape of the candle
@@ -73,6 +74,10 @@ class GAEngine:
             else:
                 segments = 1
         self.segments = max(int(segments), 1)
+        # continuous=True routes GPU fitness through the time-parallel
+        # kernel pair (ops/hip/backtest_tp.hip): unsegmented multi-year
+        # fitness at full occupancy. Requires segments == 1.
+        self.continuous = continuous and self.segments == 1
         if self.segments > 1:
             assert self.T % self.segments == 0, \
                 "T must divide evenly into segments"
@@ -103,8 +108,13 @@ class GAEngine:
         """Backtest the local shard; all-gather to global fitness."""
         shard = self.pop_t[self._my_slice()]
         if self.use_gpu:
-            from ..ops.backtest import run_backtest_gpu
-            metrics = run_backtest_gpu(self.candles_t, shard)
+            if self.continuous:
+                from ..ops.backtest import run_backtest_continuous_gpu
+                metrics = run_backtest_continuous_gpu(
+                    self.candles_t, shard)
+            else:
+                from ..ops.backtest import run_backtest_gpu
+                metrics = run_backtest_gpu(self.candles_t, shard)
             fitness_local = metrics[..., 9].mean(dim=1)
         else:
             metrics_np = run_backtest_cpu(

@@ -81,6 +81,15 @@ MAX_WIN = 32           # max Bollinger window — sized to the kernel's LDS ring
 SHARED_HALO = 64       # kernel halo: covers bb<=32 and sma50 windows
 WARMUP = 128           # candles before the first vote (covers 3x max period)
 FEE = 0.001            # taker fee per side (reference: strategy_tester.py 0.1%)
+# Every RESNAP candles (t % RESNAP == 0, t > 0) the f64 Bollinger rolling
+# sums are recomputed directly from the window (oldest->newest add order)
+# instead of incrementally. This makes the BB state *exactly restartable*
+# at any RESNAP-aligned boundary — the property the time-parallel GPU
+# backtest (ops/hip/backtest_tp.hip) relies on to split the time axis
+# across waves while staying bit-identical to the sequential engines.
+# Multiple of the kernel tile (256). Cost: ~2x32 f64 adds per 4096
+# candles — amortized noise.
+RESNAP = 4096
 
 PARAM_NAMES = [
     "rsi_period", "rsi_oversold", "rsi_overbought",

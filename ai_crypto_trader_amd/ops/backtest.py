@@ -41,6 +41,71 @@ def run_backtest_gpu(
     return metrics
 
 
+# Flag-bitmap buffers reused across GA generations (16.4 GB at the
+# flagship 1024x64x1M shape — allocated once, not per step).
+_flag_cache: dict = {}
+
+
+def pick_nshards(nsym: int, T: int, P: int, *, target_blocks: int = 2048,
+                 tail: int = 2048) -> int:
+    """Time shards for bt_flags: enough blocks to give every SIMD 4-8
+    resident waves (256 CUs x 8 blocks), bounded so each shard body is
+    >= one RESNAP period and >= 4x the warm tail."""
+    chunks = (P + 255) // 256
+    base = nsym * chunks
+    want = max(1, -(-target_blocks // base))
+    max_shards = max(1, T // max(4096, 4 * tail))
+    return min(want, max_shards, 64)
+
+
+def run_backtest_continuous_gpu(
+    candles: torch.Tensor,     # (nsym, T, 4) f32 cuda
+    population: torch.Tensor,  # (P, NPARAM) f32 cuda
+    *,
+    initial_equity: float = 1.0,
+    nshards: int = 0,          # 0 = auto
+    tail: int = 2048,
+) -> torch.Tensor:             # (P, nsym, NMETRIC) f32 cuda
+    """Continuous (unsegmented) backtest via the time-parallel kernel
+    pair (ops/hip/backtest_tp.hip): indicator/vote flags computed with
+    the time axis split across shards (warm-tail reconverged EMA/RSI +
+    RESNAP-exact Bollinger), then the exact sequential position state
+    machine over packed flag words. Same metrics contract as
+    run_backtest_gpu / engine_cpu.run_backtest_cpu."""
+    ops = require_hip_ops()
+    assert candles.is_cuda and population.is_cuda
+    assert candles.dtype == torch.float32
+    assert population.dtype == torch.float32
+    candles = candles.contiguous()
+    population = population.contiguous()
+    nsym, T, _ = candles.shape
+    P = population.shape[0]
+    if nshards <= 0:
+        nshards = pick_nshards(nsym, T, P, tail=tail)
+    nwords = (T + 63) // 64
+    key = (nsym, nwords, P, candles.device.index)
+    bufs = _flag_cache.get(key)
+    if bufs is None or bufs[0].device != candles.device:
+        eflags = torch.empty((nsym, nwords, P), dtype=torch.int64,
+                             device=candles.device)
+        xflags = torch.empty_like(eflags)
+        _flag_cache.clear()      # one shape live at a time (16 GB-class)
+        _flag_cache[key] = (eflags, xflags)
+    else:
+        eflags, xflags = bufs
+    metrics = torch.empty((P, nsym, NMETRIC), dtype=torch.float32,
+                          device=candles.device)
+    stream = torch.cuda.current_stream(candles.device).cuda_stream
+    ops.bt_flags(candles.data_ptr(), population.data_ptr(),
+                 eflags.data_ptr(), xflags.data_ptr(),
+                 nsym, T, P, nshards, tail, stream)
+    ops.bt_trades(candles.data_ptr(), population.data_ptr(),
+                  eflags.data_ptr(), xflags.data_ptr(),
+                  metrics.data_ptr(), nsym, T, P,
+                  float(initial_equity), stream)
+    return metrics
+
+
 def fitness_from_metrics(metrics: torch.Tensor) -> torch.Tensor:
     """Aggregate per-(param, symbol) fitness to per-param GA fitness:
     mean across symbols (the reference's GA evaluates one fitness per

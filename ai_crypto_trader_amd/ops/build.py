@@ -20,6 +20,7 @@ SO_PATH = OPS_DIR / "_hip_ops.so"
 
 SOURCES = [
     "backtest.hip",
+    "backtest_tp.hip",
     "ga.hip",
     "montecarlo.hip",
     "covar.hip",

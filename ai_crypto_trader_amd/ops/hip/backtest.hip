@@ -35,6 +35,7 @@
 #define BT_NPARAM 19
 #define BT_NMETRIC 10
 #define BT_WARMUP 128          // == strategy.py WARMUP
+#define BT_RESNAP 4096         // == strategy.py RESNAP (BB sum resnap)
 #define BT_FEE 0.001f
 #define BT_EPS 1e-9f
 #define BT_ANNUALIZE 724.9827573f   // float32(sqrt(525600))
@@ -92,8 +93,27 @@ struct BtState {
         sum_ret = sum_ret2 = 0.f;
     }
 
+    // Drift-free Bollinger resnap (strategy.py RESNAP): recompute the
+    // window sums directly, oldest->newest, from the close history.
+    // `hist` points at close[t] (the resnap candle itself, j = 0 term).
+    __device__ void resnap(const float* __restrict__ hist)
+    {
+#pragma clang fp contract(off)
+        double s = 0.0, s2 = 0.0;
+        const int w = q.bb_w;
+        for (int j = BT_MAXWIN - 1; j >= 0; --j) {
+            if (j < w) {
+                double c = (double)hist[-j];
+                s += c;
+                s2 += c * c;
+            }
+        }
+        bb_sum = s;
+        bb_sum2 = s2;
+    }
+
     __device__ void step(int t, float close, float high, float low,
-                         float oldc, float change, float4 sv)
+                         float oldc, float change, float4 sv, bool skip_bb)
     {
 #pragma clang fp contract(off)           // match the numpy f32 reference
         // --- 1. indicators -------------------------------------------
@@ -116,11 +136,14 @@ struct BtState {
         float rsi_den = avg_gain + fmaxf(avg_loss, BT_EPS);
 
         // Bollinger: close[t - W] comes from the shared halo tile
-        // (== the zero-initialized per-lane ring of engine_cpu.py)
-        double old = (double)oldc;
-        double c64 = (double)close;
-        bb_sum += c64 - old;
-        bb_sum2 += c64 * c64 - old * old;
+        // (== the zero-initialized per-lane ring of engine_cpu.py).
+        // skip_bb: a resnap() just replaced the sums for this candle.
+        if (!skip_bb) {
+            double old = (double)oldc;
+            double c64 = (double)close;
+            bb_sum += c64 - old;
+            bb_sum2 += c64 * c64 - old * old;
+        }
         double inv_cnt = inv_w;
         if (t < BT_MAXWIN && t + 1 < q.bb_w)   // uniformly skipped t>=32
             inv_cnt = 1.0 / (t + 1.0);
@@ -288,6 +311,14 @@ __global__ void __launch_bounds__(BT_BLOCK) backtest_kernel(
         }
         __syncthreads();
         const int tend = min(BT_TILE, T - t0);
+        // BB resnap at RESNAP-aligned tiles (engine_cpu.py lockstep):
+        // window closes for candle t0 are chist[BT_HALO - j], j < bb_w.
+        const bool resnap_tile = (t0 > 0) && ((t0 & (BT_RESNAP - 1)) == 0);
+        if (resnap_tile) {
+#pragma unroll
+            for (int i = 0; i < ILP; ++i)
+                st[i].resnap(&chist[BT_HALO]);
+        }
         // cooperative shared-series precompute: one thread per tile candle
         if (tid < tend) {
             const int t = t0 + tid;
@@ -325,7 +356,8 @@ __global__ void __launch_bounds__(BT_BLOCK) backtest_kernel(
 #pragma unroll
             for (int i = 0; i < ILP; ++i)
                 st[i].step(t, close, high, low,
-                           chist[tt + BT_HALO - st[i].q.bb_w], change, sv);
+                           chist[tt + BT_HALO - st[i].q.bb_w], change, sv,
+                           resnap_tile && tt == 0);
             prev_close = close;
         }
     }

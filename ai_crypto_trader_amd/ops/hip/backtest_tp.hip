@@ -1,0 +1,529 @@
+// Time-parallel continuous backtest — the seg=1 fast path.
+//
+// The classic kernel (backtest.hip) marches one lane per (param x symbol)
+// through all T candles. At the flagship shape (pop=1024 x 64 symbols)
+// that is 65,536 lanes = 1,024 waves = exactly 1 wave/SIMD on MI355X:
+// the per-candle dependency chain (EMA/RSI/f64-Bollinger -> votes ->
+// position machine) is fully latency-exposed and the launch runs ~2.7x
+// below the chip's issue wall (round-1 PMC analysis,
+// profiles/backtest_seg64_pmc.json).
+//
+// This file splits the problem into two kernels so the expensive part
+// fills the chip:
+//
+//   1. bt_flags_kernel — indicators + votes only, time-split into S
+//      shards per lane (S x more waves -> 4-8 waves/SIMD). Exactness:
+//        - EMA/MACD-signal/Wilder-RSI are contractive recurrences; each
+//          shard warm-starts them TAIL (default 2048) candles early,
+//          which drives the difference from the continuous trajectory
+//          below half an f32 ulp (worst contraction 1-1/64 per candle:
+//          0.984^2048 ~ 1e-14) — bitwise reconvergence in practice,
+//          asserted by tests/test_gpu_kernels GPU parity tests.
+//        - Bollinger f64 rolling sums are NOT contractive; instead both
+//          sequential engines resnap them from the raw window every
+//          RESNAP=4096 candles (strategy.py), and shard bodies start on
+//          RESNAP-aligned boundaries, so the shard's BB state is
+//          *bit-identical* to the continuous run by construction.
+//        - stoch/Williams/trend vote inputs are finite-window shared
+//          series recomputed from the tile halo — exact by construction.
+//      Output: per-lane entry/exit flag bitmaps (net >= entry_votes,
+//      net <= -exit_votes), 1 bit per candle packed into uint64 words.
+//
+//   2. bt_trades_kernel — the exact position state machine + equity
+//      accounting (identical op order to engine_cpu.py), reading flags
+//      instead of recomputing votes. Still 1 wave/SIMD, but the
+//      per-candle chain is ~5x shorter (no indicators, no f64, no
+//      sqrt), flat candles are provably no-ops (equity==cash -> r==0,
+//      engine_cpu semantics), and whole 64-candle words are skipped
+//      when no lane of the wave is in a position and no entry flag is
+//      set (__ballot).
+//
+// Flag layout: [sym][word][param] (uint64), word = t/64 — phase 1
+// writes and phase 2 reads are both coalesced across the param axis.
+//
+// Reference semantics: backtesting/strategy.py (shared CPU/GPU
+// contract); replaces the per-candle loop of the reference's
+// backtesting/strategy_tester.py:190-300 at continuous (multi-year
+// single-history) shape.
+
+#include "common.hpp"
+
+#define BT_BLOCK 256
+#define BT_TILE 256
+#define BT_MAXWIN 32
+#define BT_HALO 64
+#define BT_SPAN (BT_TILE + BT_HALO)
+#define BT_NPARAM 19
+#define BT_NMETRIC 10
+#define BT_WARMUP 128
+#define BT_RESNAP 4096
+#define BT_FEE 0.001f
+#define BT_EPS 1e-9f
+#define BT_ANNUALIZE 724.9827573f   // float32(sqrt(525600))
+
+namespace {
+
+// ---------------------------------------------------------------------
+// Phase 1: indicator/vote state (no position, no metrics)
+// ---------------------------------------------------------------------
+struct FlagState {
+    float inv_rsi_p, rsi_os, rsi_ob;
+    float a_f, a_s, a_sig;
+    int bb_w;
+    float bb_k, bb_bth, bb_sth;
+    int entry_v, exit_v;
+    float stoch_os, stoch_ob, will_os, will_ob;
+
+    float ema_f, ema_s, sig;
+    float avg_gain, avg_loss;
+    double bb_sum, bb_sum2, inv_w;
+
+    __device__ void load(const float* __restrict__ pr)
+    {
+        inv_rsi_p = 1.0f / fmaxf(floorf(pr[0]), 1.0f);
+        rsi_os = pr[1]; rsi_ob = pr[2];
+        a_f = 2.0f / (pr[3] + 1.0f);
+        a_s = 2.0f / (pr[4] + 1.0f);
+        a_sig = 2.0f / (pr[5] + 1.0f);
+        bb_w = min(max((int)pr[6], 2), BT_MAXWIN);
+        bb_k = pr[7]; bb_bth = pr[8]; bb_sth = pr[9];
+        entry_v = (int)pr[10]; exit_v = (int)pr[11];
+        stoch_os = pr[17]; stoch_ob = pr[18];
+        will_os = stoch_os - 100.0f;
+        will_ob = stoch_ob - 100.0f;
+        ema_f = ema_s = sig = 0.f;
+        avg_gain = avg_loss = 0.f;
+        bb_sum = bb_sum2 = 0.0;
+        inv_w = 1.0 / (double)bb_w;
+    }
+
+    // warm-tail march: the same f32 recurrence ops as step(), minus
+    // Bollinger (resnapped at body start) and votes
+    __device__ void warm_step(float close, float change)
+    {
+#pragma clang fp contract(off)
+        ema_f += a_f * (close - ema_f);
+        ema_s += a_s * (close - ema_s);
+        float macd = ema_f - ema_s;
+        sig += a_sig * (macd - sig);
+        float gain = fmaxf(change, 0.0f);
+        float loss = fmaxf(-change, 0.0f);
+        avg_gain += (gain - avg_gain) * inv_rsi_p;
+        avg_loss += (loss - avg_loss) * inv_rsi_p;
+    }
+
+    __device__ void resnap(const float* __restrict__ hist)
+    {
+#pragma clang fp contract(off)
+        double s = 0.0, s2 = 0.0;
+        const int w = bb_w;
+        for (int j = BT_MAXWIN - 1; j >= 0; --j) {
+            if (j < w) {
+                double c = (double)hist[-j];
+                s += c;
+                s2 += c * c;
+            }
+        }
+        bb_sum = s;
+        bb_sum2 = s2;
+    }
+
+    // (the full body step — indicators + votes + flag bits — is inlined
+    // in bt_flags_kernel so the packed-word bookkeeping stays local)
+};
+
+// ---------------------------------------------------------------------
+// Phase 2: position machine + metrics (exact engine_cpu op order)
+// ---------------------------------------------------------------------
+struct TradeState {
+    float size_pct, sl_pct, tp_pct, trail_pct, trail_act;
+    float cash, units;
+    bool in_pos;
+    float entry_cost, entry_price;
+    float stop, tp, peak;
+    float equity, max_eq, max_dd;
+    float n_trades, wins, gross_p, gross_l;
+    float sum_ret, sum_ret2;
+
+    __device__ void load(const float* __restrict__ pr, float initial_equity)
+    {
+        size_pct = pr[12]; sl_pct = pr[13]; tp_pct = pr[14];
+        trail_pct = pr[15]; trail_act = pr[16];
+        cash = initial_equity; units = 0.f;
+        in_pos = false;
+        entry_cost = entry_price = stop = tp = peak = 0.f;
+        equity = initial_equity; max_eq = initial_equity; max_dd = 0.f;
+        n_trades = wins = gross_p = gross_l = 0.f;
+        sum_ret = sum_ret2 = 0.f;
+    }
+
+    __device__ void pstep(float close, float high, float low,
+                          bool ebit, bool xbit)
+    {
+#pragma clang fp contract(off)
+        if (in_pos) {
+            peak = fmaxf(peak, high);
+            bool trail_on = (trail_pct > 0.0f) &&
+                            (peak >= entry_price * (1.0f + trail_act));
+            if (trail_on)
+                stop = fmaxf(stop, peak * (1.0f - trail_pct));
+            bool hit_sl = low <= stop;
+            bool hit_tp = !hit_sl && high >= tp;
+            bool hit_sig = !hit_sl && !hit_tp && xbit;
+            if (hit_sl || hit_tp || hit_sig) {
+                float exit_price = hit_sl ? stop : (hit_tp ? tp : close);
+                float proceeds = units * exit_price * (1.0f - BT_FEE);
+                float pnl = proceeds - entry_cost;
+                cash += proceeds;
+                n_trades += 1.0f;
+                wins += (pnl > 0.0f) ? 1.0f : 0.0f;
+                gross_p += fmaxf(pnl, 0.0f);
+                gross_l += fmaxf(-pnl, 0.0f);
+                units = 0.0f;
+                in_pos = false;
+            }
+        } else if (ebit) {     // ebit encodes t>=WARMUP && net>=entry_v
+            float cost = fminf(size_pct * equity, cash);
+            units = cost * (1.0f - BT_FEE) / close;
+            cash -= cost;
+            entry_cost = cost;
+            entry_price = close;
+            stop = close * (1.0f - sl_pct);
+            tp = close * (1.0f + tp_pct);
+            peak = close;
+            in_pos = true;
+        }
+        // flat lanes: new_eq == cash == equity exactly -> r == 0 and
+        // every accumulator unchanged; skipping is bit-identical to
+        // engine_cpu.py (backtest.hip does the same)
+        if (units != 0.0f || cash != equity) {
+            float new_eq = cash + units * close;
+            float r = new_eq / equity - 1.0f;
+            sum_ret += r;
+            sum_ret2 += r * r;
+            equity = new_eq;
+            max_eq = fmaxf(max_eq, equity);
+            max_dd = fmaxf(max_dd, (max_eq - equity) / max_eq);
+        }
+    }
+
+    __device__ void finalize(float* __restrict__ out, int T) const
+    {
+#pragma clang fp contract(off)
+        // identical to backtest.hip BtState::finalize
+        float n = (float)max(T, 1);
+        float mean_r = sum_ret / n;
+        float var_r = fmaxf(sum_ret2 / n - mean_r * mean_r, 0.0f);
+        float sharpe = mean_r / fmaxf(sqrtf(var_r), BT_EPS) * BT_ANNUALIZE;
+        if (!(n_trades > 0.0f)) sharpe = 0.0f;
+        float win_rate = wins / fmaxf(n_trades, 1.0f);
+        float fitness = (n_trades > 0.0f)
+                            ? sharpe + win_rate - 2.0f * max_dd
+                            : -1.0f;
+        out[0] = equity; out[1] = n_trades; out[2] = wins;
+        out[3] = gross_p; out[4] = gross_l; out[5] = max_dd;
+        out[6] = sum_ret; out[7] = sum_ret2;
+        out[8] = sharpe; out[9] = fitness;
+    }
+};
+
+// ---------------------------------------------------------------------
+// Kernel 1: flags. Grid = nshards x nsym x chunks blocks of 256 lanes.
+// ---------------------------------------------------------------------
+__global__ void __launch_bounds__(BT_BLOCK) bt_flags_kernel(
+    const float* __restrict__ candles,      // (nsym, T, 4)
+    const float* __restrict__ pop,          // (P, NPARAM)
+    unsigned long long* __restrict__ eflags,  // (nsym, nwords, P)
+    unsigned long long* __restrict__ xflags,
+    int nsym, int T, int P, int chunks, int nshards, int body4, int tail)
+{
+#pragma clang fp contract(off)
+    __shared__ float chist[BT_SPAN];
+    __shared__ float hl[BT_SPAN][2];
+    __shared__ float4 sh_vote[BT_TILE];
+
+    const int bid = blockIdx.x;
+    const int shard = bid / (nsym * chunks);
+    const int rem = bid % (nsym * chunks);
+    const int sym = rem / chunks;
+    const int chunk = rem % chunks;
+    const int tid = threadIdx.x;
+    const int p = chunk * BT_BLOCK + tid;
+    const bool act = p < P;
+    const long nwords = (T + 63) >> 6;
+
+    FlagState st;
+    st.load(pop + (long)(act ? p : 0) * BT_NPARAM);
+
+    const int lo = shard * body4;
+    const int hi = (shard == nshards - 1) ? T : (shard + 1) * body4;
+    const int start = (shard == 0) ? 0 : lo - tail;
+
+    const float4* sym_candles =
+        reinterpret_cast<const float4*>(candles + (long)sym * T * 4);
+
+    float prev_close = (start > 0) ? sym_candles[start - 1].x : 0.f;
+    if (shard != 0) {
+        // warm-tail init mirrors the t==0 branch: EMA seeded at the
+        // first tail close; sig/avg_* at zero. All contract below an
+        // f32 ulp within `tail` candles (see header).
+        float c0 = sym_candles[start].x;
+        st.ema_f = c0;
+        st.ema_s = c0;
+    }
+
+    unsigned long long ew = 0ull, xw = 0ull;
+
+    for (int t0 = start; t0 < hi; t0 += BT_TILE) {
+        __syncthreads();
+        for (int i = tid; i < BT_SPAN; i += BT_BLOCK) {
+            const int t = t0 - BT_HALO + i;
+            if (t >= 0 && t < T) {
+                float4 c = sym_candles[t];
+                chist[i] = c.x;
+                hl[i][0] = c.y;
+                hl[i][1] = c.z;
+            } else {
+                chist[i] = 0.0f;
+                hl[i][0] = 0.0f;
+                hl[i][1] = 0.0f;
+            }
+        }
+        __syncthreads();
+        const int tend = min(BT_TILE, hi - t0);
+
+        if (t0 < lo) {
+            // pure warm-tail tile (tail is a multiple of BT_TILE, so
+            // tiles never straddle the body boundary): recurrences only.
+            // First tail candle: ema was seeded AT this close, so the
+            // ema update is a bitwise no-op (a*(c-c) == 0) — mirroring
+            // the t==0 init semantics without a special case.
+            for (int tt = 0; tt < tend; ++tt) {
+                const float close = chist[tt + BT_HALO];
+                st.warm_step(close, close - prev_close);
+                prev_close = close;
+            }
+            continue;
+        }
+
+        // body tile: shared vote series (exact, finite-window)
+        if (tid < tend) {
+            const int t = t0 + tid;
+            const int base = tid + BT_HALO;
+            const float cl = chist[base];
+            const int L14 = min(t + 1, 14);
+            float hmax = -1e30f, lmin = 1e30f;
+            for (int j = 0; j < L14; ++j) {
+                hmax = fmaxf(hmax, hl[base - j][0]);
+                lmin = fminf(lmin, hl[base - j][1]);
+            }
+            const int L20 = min(t + 1, 20);
+            double s20 = 0.0;
+            for (int j = 0; j < L20; ++j) s20 += (double)chist[base - j];
+            const float sma20 = (float)(s20 / (double)L20);
+            const int L50 = min(t + 1, 50);
+            double s50 = 0.0;
+            for (int j = 0; j < L50; ++j) s50 += (double)chist[base - j];
+            const float sma50 = (float)(s50 / (double)L50);
+            const float trend = (cl > sma20 && sma20 > sma50) ? 1.0f
+                                : ((cl < sma20 && sma20 < sma50) ? -1.0f
+                                                                 : 0.0f);
+            sh_vote[tid] = make_float4(
+                100.0f * (cl - lmin), -100.0f * (hmax - cl),
+                fmaxf(hmax - lmin, BT_EPS), trend);
+        }
+        __syncthreads();
+
+        // BB resnap at RESNAP-aligned tiles (all shards see the same
+        // aligned boundaries -> bit-identical to the sequential run)
+        const bool resnap_tile = (t0 > 0) && ((t0 & (BT_RESNAP - 1)) == 0);
+        if (resnap_tile)
+            st.resnap(&chist[BT_HALO]);
+
+        for (int tt = 0; tt < tend; ++tt) {
+            const int t = t0 + tt;
+            const float close = chist[tt + BT_HALO];
+            const float change = (t == 0) ? 0.0f : close - prev_close;
+            const float4 sv = sh_vote[tt];
+
+            // indicators (same op order as backtest.hip BtState::step)
+            if (t == 0) { st.ema_f = close; st.ema_s = close; }
+            else {
+                st.ema_f += st.a_f * (close - st.ema_f);
+                st.ema_s += st.a_s * (close - st.ema_s);
+            }
+            float macd = st.ema_f - st.ema_s;
+            st.sig += st.a_sig * (macd - st.sig);
+            float macd_hist = macd - st.sig;
+
+            float gain = fmaxf(change, 0.0f);
+            float loss = fmaxf(-change, 0.0f);
+            st.avg_gain += (gain - st.avg_gain) * st.inv_rsi_p;
+            st.avg_loss += (loss - st.avg_loss) * st.inv_rsi_p;
+            float rsi_num = 100.0f * st.avg_gain;
+            float rsi_den = st.avg_gain + fmaxf(st.avg_loss, BT_EPS);
+
+            if (!(resnap_tile && tt == 0)) {
+                double old = (double)chist[tt + BT_HALO - st.bb_w];
+                double c64 = (double)close;
+                st.bb_sum += c64 - old;
+                st.bb_sum2 += c64 * c64 - old * old;
+            }
+            double inv_cnt = st.inv_w;
+            if (t < BT_MAXWIN && t + 1 < st.bb_w)
+                inv_cnt = 1.0 / (t + 1.0);
+            double mean64 = st.bb_sum * inv_cnt;
+            double var64 =
+                fmax(st.bb_sum2 * inv_cnt - mean64 * mean64, 0.0);
+            float mean = (float)mean64;
+            float std_ = sqrtf((float)var64);
+            float band = st.bb_k * std_;
+            float bb_num = close - (mean - band);
+            float bb_den = fmaxf(2.0f * band, BT_EPS);
+
+            int net = 0;
+            if (t >= BT_WARMUP) {
+                int buy = (rsi_num < st.rsi_os * rsi_den) +
+                          (macd_hist > 0.0f) +
+                          (bb_num < st.bb_bth * bb_den) +
+                          (sv.x < st.stoch_os * sv.z) +
+                          (sv.y < st.will_os * sv.z) +
+                          (sv.w > 0.0f);
+                int sell = (rsi_num > st.rsi_ob * rsi_den) +
+                           (macd_hist < 0.0f) +
+                           (bb_num > st.bb_sth * bb_den) +
+                           (sv.x > st.stoch_ob * sv.z) +
+                           (sv.y > st.will_ob * sv.z) +
+                           (sv.w < 0.0f);
+                net = buy - sell;
+            }
+            const int bit = t & 63;
+            ew |= (unsigned long long)(net >= st.entry_v) << bit;
+            xw |= (unsigned long long)(net <= -st.exit_v) << bit;
+            if (bit == 63 || t == T - 1) {
+                if (act) {
+                    const long w = t >> 6;
+                    eflags[((long)sym * nwords + w) * P + p] = ew;
+                    xflags[((long)sym * nwords + w) * P + p] = xw;
+                }
+                ew = 0ull;
+                xw = 0ull;
+            }
+            prev_close = close;
+        }
+    }
+}
+
+// ---------------------------------------------------------------------
+// Kernel 2: trades. Grid = nsym x chunks blocks of 256 lanes
+// (XCD-affine mapping as in backtest.hip).
+// ---------------------------------------------------------------------
+__global__ void __launch_bounds__(BT_BLOCK) bt_trades_kernel(
+    const float* __restrict__ candles,
+    const float* __restrict__ pop,
+    const unsigned long long* __restrict__ eflags,
+    const unsigned long long* __restrict__ xflags,
+    float* __restrict__ metrics,             // (P, nsym, NMETRIC)
+    int nsym, int T, int P, int chunks_per_sym, float initial_equity)
+{
+#pragma clang fp contract(off)
+    __shared__ float sc[BT_TILE];
+    __shared__ float sh[BT_TILE];
+    __shared__ float sl[BT_TILE];
+
+    int bid = blockIdx.x;
+    int sym, chunk;
+    int nblocks = nsym * chunks_per_sym;
+    if ((nsym & 7) == 0 && (nblocks & 7) == 0) {
+        int xcd = bid & 7, j = bid >> 3;
+        sym = xcd + 8 * (j / chunks_per_sym);
+        chunk = j % chunks_per_sym;
+    } else {
+        sym = bid / chunks_per_sym;
+        chunk = bid % chunks_per_sym;
+    }
+    const int tid = threadIdx.x;
+    const int p = chunk * BT_BLOCK + tid;
+    const bool act = p < P;
+    const long nwords = (T + 63) >> 6;
+
+    TradeState st;
+    st.load(pop + (long)(act ? p : 0) * BT_NPARAM, initial_equity);
+
+    const float4* sym_candles =
+        reinterpret_cast<const float4*>(candles + (long)sym * T * 4);
+    const unsigned long long* esym = eflags + (long)sym * nwords * P;
+    const unsigned long long* xsym = xflags + (long)sym * nwords * P;
+
+    for (int t0 = 0; t0 < T; t0 += BT_TILE) {
+        __syncthreads();
+        for (int i = tid; i < BT_TILE; i += BT_BLOCK) {
+            const int t = t0 + i;
+            if (t < T) {
+                float4 c = sym_candles[t];
+                sc[i] = c.x;
+                sh[i] = c.y;
+                sl[i] = c.z;
+            }
+        }
+        __syncthreads();
+        const int tend = min(BT_TILE, T - t0);
+
+        for (int wq = 0; wq < BT_TILE / 64; ++wq) {
+            const int wbase = wq * 64;
+            if (wbase >= tend) break;
+            const int wlen = min(64, tend - wbase);
+            const long w = (t0 + wbase) >> 6;
+            const unsigned long long ewrd =
+                act ? esym[w * P + p] : 0ull;
+            // exact whole-word skip: a lane is inert for the word if it
+            // is flat, settled (cash==equity -> mark is a no-op) and no
+            // entry bit is set; when every lane of the wave is inert the
+            // 64 candles are provably no-ops (engine_cpu semantics)
+            const bool busy =
+                st.in_pos || (st.cash != st.equity) || ewrd != 0ull;
+            if (__ballot(busy) == 0ull)
+                continue;
+            const unsigned long long xwrd =
+                act ? xsym[w * P + p] : 0ull;
+            for (int k = 0; k < wlen; ++k) {
+                const int i = wbase + k;
+                st.pstep(sc[i], sh[i], sl[i],
+                         (ewrd >> k) & 1ull, (xwrd >> k) & 1ull);
+            }
+        }
+    }
+
+    if (act)
+        st.finalize(metrics + ((long)p * nsym + sym) * BT_NMETRIC, T);
+}
+
+}  // namespace
+
+extern "C" void launch_bt_flags(const float* candles, const float* pop,
+                                unsigned long long* eflags,
+                                unsigned long long* xflags,
+                                int nsym, int T, int P, int nshards,
+                                int tail, hipStream_t stream) {
+    int chunks = (P + BT_BLOCK - 1) / BT_BLOCK;
+    int body4 = nshards > 1
+                    ? (T / nshards) / BT_RESNAP * BT_RESNAP
+                    : T;
+    hipLaunchKernelGGL(bt_flags_kernel,
+                       dim3(nshards * nsym * chunks), dim3(BT_BLOCK), 0,
+                       stream, candles, pop, eflags, xflags, nsym, T, P,
+                       chunks, nshards, body4, tail);
+}
+
+extern "C" void launch_bt_trades(const float* candles, const float* pop,
+                                 const unsigned long long* eflags,
+                                 const unsigned long long* xflags,
+                                 float* metrics, int nsym, int T, int P,
+                                 float initial_equity,
+                                 hipStream_t stream) {
+    int chunks = (P + BT_BLOCK - 1) / BT_BLOCK;
+    hipLaunchKernelGGL(bt_trades_kernel, dim3(nsym * chunks),
+                       dim3(BT_BLOCK), 0, stream, candles, pop, eflags,
+                       xflags, metrics, nsym, T, P, chunks,
+                       initial_equity);
+}

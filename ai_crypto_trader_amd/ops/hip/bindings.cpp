@@ -13,6 +13,13 @@
 
 namespace py = pybind11;
 
+extern "C" void launch_bt_flags(const float*, const float*,
+                                unsigned long long*, unsigned long long*,
+                                int, int, int, int, int, hipStream_t);
+extern "C" void launch_bt_trades(const float*, const float*,
+                                 const unsigned long long*,
+                                 const unsigned long long*, float*, int,
+                                 int, int, float, hipStream_t);
 extern "C" void launch_backtest(const float*, const float*, float*, int, int,
                                 int, float, hipStream_t);
 extern "C" void launch_ga_evolve(const float*, const float*, const int*,
@@ -92,6 +99,40 @@ PYBIND11_MODULE(_hip_ops, m) {
           py::arg("candles"), py::arg("pop"), py::arg("metrics"),
           py::arg("nsym"), py::arg("T"), py::arg("P"),
           py::arg("initial_equity"), py::arg("stream"));
+
+    m.def("bt_flags",
+          [](uintptr_t candles, uintptr_t pop, uintptr_t eflags,
+             uintptr_t xflags, int nsym, int T, int P, int nshards,
+             int tail, uintptr_t stream) {
+              launch_bt_flags(
+                  reinterpret_cast<const float*>(candles),
+                  reinterpret_cast<const float*>(pop),
+                  reinterpret_cast<unsigned long long*>(eflags),
+                  reinterpret_cast<unsigned long long*>(xflags), nsym, T,
+                  P, nshards, tail, as_stream(stream));
+              check(hipGetLastError(), "bt_flags launch");
+          },
+          py::arg("candles"), py::arg("pop"), py::arg("eflags"),
+          py::arg("xflags"), py::arg("nsym"), py::arg("T"), py::arg("P"),
+          py::arg("nshards"), py::arg("tail"), py::arg("stream"));
+
+    m.def("bt_trades",
+          [](uintptr_t candles, uintptr_t pop, uintptr_t eflags,
+             uintptr_t xflags, uintptr_t metrics, int nsym, int T, int P,
+             float initial_equity, uintptr_t stream) {
+              launch_bt_trades(
+                  reinterpret_cast<const float*>(candles),
+                  reinterpret_cast<const float*>(pop),
+                  reinterpret_cast<const unsigned long long*>(eflags),
+                  reinterpret_cast<const unsigned long long*>(xflags),
+                  reinterpret_cast<float*>(metrics), nsym, T, P,
+                  initial_equity, as_stream(stream));
+              check(hipGetLastError(), "bt_trades launch");
+          },
+          py::arg("candles"), py::arg("pop"), py::arg("eflags"),
+          py::arg("xflags"), py::arg("metrics"), py::arg("nsym"),
+          py::arg("T"), py::arg("P"), py::arg("initial_equity"),
+          py::arg("stream"));
 
     m.def("ga_evolve",
           [](uintptr_t pop, uintptr_t fitness, uintptr_t order,

@@ -25,7 +25,9 @@ from __future__ import annotations
 
 import argparse
 import json
+import sys
 import time
+from pathlib import Path
 
 import numpy as np
 import torch
@@ -47,10 +49,34 @@ def parse_args():
     ap.add_argument("--seed", type=int, default=0)
     ap.add_argument("--skip-mc", action="store_true",
                     help="skip the secondary Monte-Carlo measurement")
+    ap.add_argument("--skip-train", action="store_true",
+                    help="skip the secondary LSTM/PPO train measurements")
+    ap.add_argument("--skip-continuous", action="store_true",
+                    help="skip the continuous (unsegmented) backtest loop")
     ap.add_argument("--segments", type=int, default=64,
                     help="fitness time-CV segments per symbol (64 measured "
                          "fastest: 343 vs 324 G candles/s at 16)")
     return ap.parse_args()
+
+
+def measure_train(device):
+    """Secondary metrics: BASELINE configs #2 (LSTM predictor train) and
+    #4 (PPO train) on this rank's GPU — driver-clocked via this run
+    (tools/bench_train.py holds the full-size standalone variants)."""
+    sys.path.insert(0, str(Path(__file__).resolve().parent / "tools"))
+    from bench_train import bench_lstm_train, bench_ppo
+
+    lstm = bench_lstm_train(nsym=32, T=1_000_000, steps=10, warmup=2)
+    ppo = bench_ppo(n_envs=256, horizon=128, steps=4, warmup=1,
+                    use_graph=True)
+    return {
+        "lstm_windows_per_sec": lstm["windows_per_sec"],
+        "lstm_ms_per_step": lstm["ms_per_step"],
+        "lstm_batch": lstm["config"]["batch"],
+        "ppo_env_steps_per_sec": ppo["env_steps_per_sec"],
+        "ppo_ms_per_step": ppo["ms_per_step"],
+        "ppo_n_envs": ppo["config"]["n_envs"],
+    }
 
 
 def measure_mc(device, n_paths=10_000_000, n_assets=64, n_steps=30):
@@ -127,9 +153,39 @@ def main():
     total_candle_evals = engine.candle_evals_per_step * world * args.steps
     value = total_candle_evals / elapsed
 
+    # Continuous (unsegmented) fitness: the same GA generation loop with
+    # the time-parallel kernel pair (ops/hip/backtest_tp.hip) — the
+    # multi-year-single-history case, all ranks participating.
+    continuous = {}
+    if on_gpu and not args.skip_continuous and segments > 1:
+        engine_c = GAEngine(
+            candles, pop_per_rank=pop, rank=rank, world=world,
+            device=device, seed=args.seed + 1, segments=1,
+            continuous=True,
+        )
+        for _ in range(max(args.warmup, 1)):
+            engine_c.step()
+        pdist.barrier()
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(args.steps):
+            engine_c.step()
+        torch.cuda.synchronize()
+        pdist.barrier()
+        el_c = pdist.all_reduce_max_scalar(
+            time.perf_counter() - t0, device)
+        continuous = {
+            "candles_per_sec":
+                engine_c.candle_evals_per_step * world * args.steps / el_c,
+            "ms_per_step": el_c / args.steps * 1000.0,
+            "fitness_segments": 1,
+        }
+
     secondary = {}
     if on_gpu and rank == 0 and not args.skip_mc:
         secondary = measure_mc(device)
+    if on_gpu and rank == 0 and not args.skip_train:
+        secondary.update(measure_train(device))
     pdist.barrier()      # all ranks leave together (rank 0 runs MC above)
 
     if rank == 0:
@@ -158,6 +214,7 @@ def main():
                     engine.candle_evals_per_step,
                 "fitness_segments": segments,
                 "best_fitness": best_fit,
+                "continuous": continuous,
                 "secondary": secondary,
             },
         }
