@@ -52,10 +52,13 @@ def main():
 
     ops = require_hip_ops()
     dev = torch.device("cuda:0")
+    print("[diag] ops loaded", flush=True)
     T, nsym, P = 18432, 2, 32
     candles = candles_chl_v(generate_ohlcv(T, nsym, seed=21))
     pop = random_population(P, seed=9)
+    print("[diag] running CPU reference...", flush=True)
     m_cpu, nets = run_backtest_cpu(candles, pop, record_net=True)
+    print("[diag] CPU done", flush=True)
     entry_v = pop[:, 10].astype(np.int32)[:, None, None]
     exit_v = pop[:, 11].astype(np.int32)[:, None, None]
     ef_cpu = nets >= entry_v
@@ -76,6 +79,7 @@ def main():
         ops.bt_flags(c_t.data_ptr(), p_t.data_ptr(), eflags.data_ptr(),
                      xflags.data_ptr(), nsym, T, P, ns, 2048, stream)
         torch.cuda.synchronize()
+        print(f"[diag] bt_flags ns={ns} ok", flush=True)
         ef = unpack_flags(eflags, T)
         xf = unpack_flags(xflags, T)
         ed = np.argwhere(ef != ef_cpu)
@@ -93,6 +97,7 @@ def main():
                   x_cpu_t.data_ptr(), metrics.data_ptr(), nsym, T, P,
                   1.0, stream)
     torch.cuda.synchronize()
+    print("[diag] bt_trades(cpu flags) ok", flush=True)
     m_tr = metrics.cpu().numpy()
     d = np.argwhere(m_tr != m_cpu)
     out["trades_with_cpu_flags"] = {
