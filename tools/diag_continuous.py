@@ -102,12 +102,43 @@ def main():
     d = np.argwhere(m_tr != m_cpu)
     out["trades_with_cpu_flags"] = {
         "metric_diffs": int(len(d)),
+        "diff_cols": sorted(set(int(k) for _, _, k in d)),
         "first": [
             [int(i), int(s), int(k), float(m_cpu[i, s, k]),
              float(m_tr[i, s, k])]
             for i, s, k in d[:8]
         ],
     }
+    if len(d):
+        # bit-level finalize introspection for the first differing lane:
+        # recompute sharpe from the GPU's own accumulator columns in
+        # numpy f32 and show where it lands vs CPU col8 and GPU col8
+        i, s, _ = d[0]
+        f32 = np.float32
+        acc = m_tr[i, s]
+        n = f32(T)
+        mean = f32(acc[6] / n)
+        var = np.maximum(f32(acc[7] / n) - mean * mean, f32(0.0))
+        std = np.sqrt(var)
+        den = np.maximum(std, f32(1e-9))
+        q = f32(mean / den)
+        ann = f32(np.sqrt(np.float64(525600.0)))
+        sharpe_np = f32(q * ann)
+
+        def bits(x):
+            return hex(np.float32(x).view(np.uint32))
+
+        out["finalize_introspect"] = {
+            "lane": [int(i), int(s)],
+            "acc_equal_cpu": bool(np.array_equal(acc[:8], m_cpu[i, s, :8])),
+            "sum_ret": bits(acc[6]), "sum_ret2": bits(acc[7]),
+            "mean": bits(mean), "var": bits(var), "std": bits(std),
+            "q_mean_over_std": bits(q),
+            "ann_np": bits(ann), "ann_kernel": hex(0x44353ee5),
+            "sharpe_np_from_gpu_acc": bits(sharpe_np),
+            "sharpe_cpu": bits(m_cpu[i, s, 8]),
+            "sharpe_gpu": bits(m_tr[i, s, 8]),
+        }
     print(json.dumps(out, indent=1))
 
 
