@@ -46,10 +46,11 @@ def run_backtest_gpu(
 _flag_cache: dict = {}
 
 
-def pick_nshards(nsym: int, T: int, P: int, *, target_blocks: int = 2048,
+def pick_nshards(nsym: int, T: int, P: int, *, target_blocks: int = 4096,
                  tail: int = 2048) -> int:
-    """Time shards for bt_flags: enough blocks to give every SIMD 4-8
-    resident waves (256 CUs x 8 blocks), bounded so each shard body is
+    """Time shards for bt_flags: enough blocks to oversubscribe every CU
+    (measured sweep at 1024x64x1M: 8 shards = 399, 16 = 419, 32 = 394
+    Gcandles/s -> target 16 blocks/CU), bounded so each shard body is
     >= one RESNAP period and >= 4x the warm tail."""
     chunks = (P + 255) // 256
     base = nsym * chunks

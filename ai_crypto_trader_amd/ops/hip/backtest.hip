@@ -38,7 +38,9 @@
 #define BT_RESNAP 4096         // == strategy.py RESNAP (BB sum resnap)
 #define BT_FEE 0.001f
 #define BT_EPS 1e-9f
-#define BT_ANNUALIZE 724.9827573f   // float32(sqrt(525600))
+// float32(sqrt(525600)) EXACTLY (0x44353ee6) — a shorter decimal literal
+// rounds to the neighbor 0x44353ee5 and biases every sharpe by 1 ulp
+#define BT_ANNUALIZE 0x1.6a7dccp+9f
 
 namespace {
 

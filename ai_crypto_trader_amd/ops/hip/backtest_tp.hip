@@ -59,7 +59,8 @@
 #define BT_RESNAP 4096
 #define BT_FEE 0.001f
 #define BT_EPS 1e-9f
-#define BT_ANNUALIZE 724.9827573f   // float32(sqrt(525600))
+// float32(sqrt(525600)) EXACTLY (0x44353ee6) — see backtest.hip
+#define BT_ANNUALIZE 0x1.6a7dccp+9f
 
 namespace {
 
@@ -133,13 +134,29 @@ struct FlagState {
 };
 
 // ---------------------------------------------------------------------
-// Phase 2: position machine + metrics (exact engine_cpu op order)
+// Phase 2: position machine + metrics (exact engine_cpu op order).
+//
+// Runs at 1 wave/SIMD (65,536 lanes at the flagship shape), so the
+// per-candle DEPENDENCY CHAIN is everything: the body is branchless
+// (selects, like the numpy engine itself — which computes every
+// quantity for every lane and np.where-selects) except the rare entry
+// block, whose division would otherwise be a third serial IEEE-divide
+// per candle. The two mark divisions (r and drawdown) only feed
+// accumulators, never the trading state, so with a branchless body the
+// scheduler pipelines them across unrolled iterations.
 // ---------------------------------------------------------------------
 struct TradeState {
-    float size_pct, sl_pct, tp_pct, trail_pct, trail_act;
+    float size_pct;
+    float fee_m;       // 1 - FEE
+    float sl_m;        // 1 - stop_loss_pct
+    float tp_m;        // 1 + take_profit_pct
+    float trail_m;     // 1 - trailing_stop_pct
+    float act_m;       // 1 + trailing_act_pct
+    bool trail_en;     // trailing_stop_pct > 0
     float cash, units;
     bool in_pos;
-    float entry_cost, entry_price;
+    float entry_cost;
+    float trail_arm;   // entry_price * act_m (armed threshold)
     float stop, tp, peak;
     float equity, max_eq, max_dd;
     float n_trades, wins, gross_p, gross_l;
@@ -147,11 +164,18 @@ struct TradeState {
 
     __device__ void load(const float* __restrict__ pr, float initial_equity)
     {
-        size_pct = pr[12]; sl_pct = pr[13]; tp_pct = pr[14];
-        trail_pct = pr[15]; trail_act = pr[16];
+#pragma clang fp contract(off)
+        size_pct = pr[12];
+        fee_m = 1.0f - BT_FEE;
+        sl_m = 1.0f - pr[13];
+        tp_m = 1.0f + pr[14];
+        trail_m = 1.0f - pr[15];
+        act_m = 1.0f + pr[16];
+        trail_en = pr[15] > 0.0f;
         cash = initial_equity; units = 0.f;
         in_pos = false;
-        entry_cost = entry_price = stop = tp = peak = 0.f;
+        entry_cost = trail_arm = 0.f;
+        stop = tp = peak = 0.f;
         equity = initial_equity; max_eq = initial_equity; max_dd = 0.f;
         n_trades = wins = gross_p = gross_l = 0.f;
         sum_ret = sum_ret2 = 0.f;
@@ -161,50 +185,54 @@ struct TradeState {
                           bool ebit, bool xbit)
     {
 #pragma clang fp contract(off)
-        if (in_pos) {
-            peak = fmaxf(peak, high);
-            bool trail_on = (trail_pct > 0.0f) &&
-                            (peak >= entry_price * (1.0f + trail_act));
-            if (trail_on)
-                stop = fmaxf(stop, peak * (1.0f - trail_pct));
-            bool hit_sl = low <= stop;
-            bool hit_tp = !hit_sl && high >= tp;
-            bool hit_sig = !hit_sl && !hit_tp && xbit;
-            if (hit_sl || hit_tp || hit_sig) {
-                float exit_price = hit_sl ? stop : (hit_tp ? tp : close);
-                float proceeds = units * exit_price * (1.0f - BT_FEE);
-                float pnl = proceeds - entry_cost;
-                cash += proceeds;
-                n_trades += 1.0f;
-                wins += (pnl > 0.0f) ? 1.0f : 0.0f;
-                gross_p += fmaxf(pnl, 0.0f);
-                gross_l += fmaxf(-pnl, 0.0f);
-                units = 0.0f;
-                in_pos = false;
-            }
-        } else if (ebit) {     // ebit encodes t>=WARMUP && net>=entry_v
+        const bool pos0 = in_pos;
+        // --- position management (engine_cpu section 3, branchless) --
+        float peak2 = fmaxf(peak, high);
+        peak = pos0 ? peak2 : peak;
+        bool trail_on = pos0 && trail_en && (peak >= trail_arm);
+        float stop2 = fmaxf(stop, peak * trail_m);
+        stop = trail_on ? stop2 : stop;
+
+        bool hit_sl = pos0 && (low <= stop);
+        bool hit_tp = pos0 && !hit_sl && (high >= tp);
+        bool hit_sig = pos0 && !hit_sl && !hit_tp && xbit;
+        bool exiting = hit_sl || hit_tp || hit_sig;
+        float exit_price = hit_sl ? stop : (hit_tp ? tp : close);
+        float proceeds = units * exit_price * fee_m;
+        float pnl = proceeds - entry_cost;
+        cash = exiting ? cash + proceeds : cash;
+        n_trades += exiting ? 1.0f : 0.0f;
+        wins += (exiting && pnl > 0.0f) ? 1.0f : 0.0f;
+        gross_p += exiting ? fmaxf(pnl, 0.0f) : 0.0f;
+        gross_l += exiting ? fmaxf(-pnl, 0.0f) : 0.0f;
+        units = exiting ? 0.0f : units;
+        in_pos = pos0 && !exiting;
+
+        // entry (rare; keeps the units division off the common path).
+        // ebit encodes t>=WARMUP && net>=entry_v; engine order: a lane
+        // exiting this candle cannot re-enter the same candle (pos0).
+        if (!pos0 && ebit) {
             float cost = fminf(size_pct * equity, cash);
-            units = cost * (1.0f - BT_FEE) / close;
+            units = cost * fee_m / close;
             cash -= cost;
             entry_cost = cost;
-            entry_price = close;
-            stop = close * (1.0f - sl_pct);
-            tp = close * (1.0f + tp_pct);
+            stop = close * sl_m;
+            tp = close * tp_m;
             peak = close;
+            trail_arm = close * act_m;
             in_pos = true;
         }
-        // flat lanes: new_eq == cash == equity exactly -> r == 0 and
-        // every accumulator unchanged; skipping is bit-identical to
-        // engine_cpu.py (backtest.hip does the same)
-        if (units != 0.0f || cash != equity) {
-            float new_eq = cash + units * close;
-            float r = new_eq / equity - 1.0f;
-            sum_ret += r;
-            sum_ret2 += r * r;
-            equity = new_eq;
-            max_eq = fmaxf(max_eq, equity);
-            max_dd = fmaxf(max_dd, (max_eq - equity) / max_eq);
-        }
+
+        // --- mark to market (always, like the numpy engine: flat lanes
+        // compute r = cash/cash - 1 == 0 exactly, all accumulators
+        // unchanged) ------------------------------------------------
+        float new_eq = cash + units * close;
+        float r = new_eq / equity - 1.0f;
+        sum_ret += r;
+        sum_ret2 += r * r;
+        equity = new_eq;
+        max_eq = fmaxf(max_eq, equity);
+        max_dd = fmaxf(max_dd, (max_eq - equity) / max_eq);
     }
 
     __device__ void finalize(float* __restrict__ out, int T) const
@@ -486,10 +514,19 @@ __global__ void __launch_bounds__(BT_BLOCK) bt_trades_kernel(
                 continue;
             const unsigned long long xwrd =
                 act ? xsym[w * P + p] : 0ull;
-            for (int k = 0; k < wlen; ++k) {
-                const int i = wbase + k;
-                st.pstep(sc[i], sh[i], sl[i],
-                         (ewrd >> k) & 1ull, (xwrd >> k) & 1ull);
+            if (wlen == 64) {
+#pragma unroll 4
+                for (int k = 0; k < 64; ++k) {
+                    const int i = wbase + k;
+                    st.pstep(sc[i], sh[i], sl[i],
+                             (ewrd >> k) & 1ull, (xwrd >> k) & 1ull);
+                }
+            } else {
+                for (int k = 0; k < wlen; ++k) {
+                    const int i = wbase + k;
+                    st.pstep(sc[i], sh[i], sl[i],
+                             (ewrd >> k) & 1ull, (xwrd >> k) & 1ull);
+                }
             }
         }
     }
