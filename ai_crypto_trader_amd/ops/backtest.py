@@ -59,6 +59,9 @@ def pick_nshards(nsym: int, T: int, P: int, *, target_blocks: int = 4096,
     return min(want, max_shards, 64)
 
 
+_pipe_streams: dict = {}
+
+
 def run_backtest_continuous_gpu(
     candles: torch.Tensor,     # (nsym, T, 4) f32 cuda
     population: torch.Tensor,  # (P, NPARAM) f32 cuda
@@ -66,13 +69,20 @@ def run_backtest_continuous_gpu(
     initial_equity: float = 1.0,
     nshards: int = 0,          # 0 = auto
     tail: int = 2048,
+    sym_groups: int = 0,       # 0 = auto; 1 = no pipelining
 ) -> torch.Tensor:             # (P, nsym, NMETRIC) f32 cuda
     """Continuous (unsegmented) backtest via the time-parallel kernel
     pair (ops/hip/backtest_tp.hip): indicator/vote flags computed with
     the time axis split across shards (warm-tail reconverged EMA/RSI +
     RESNAP-exact Bollinger), then the exact sequential position state
     machine over packed flag words. Same metrics contract as
-    run_backtest_gpu / engine_cpu.run_backtest_cpu."""
+    run_backtest_gpu / engine_cpu.run_backtest_cpu.
+
+    The two kernels are pipelined over symbol groups on two HIP
+    streams: trades(group g) starts as soon as flags(group g) is done,
+    so the latency-bound trades waves (1 wave/SIMD) execute co-resident
+    with the issue-bound flags waves of later groups and hide their
+    dependency stalls in the flags kernel's occupancy."""
     ops = require_hip_ops()
     assert candles.is_cuda and population.is_cuda
     assert candles.dtype == torch.float32
@@ -84,26 +94,67 @@ def run_backtest_continuous_gpu(
     if nshards <= 0:
         nshards = pick_nshards(nsym, T, P, tail=tail)
     nwords = (T + 63) // 64
-    key = (nsym, nwords, P, candles.device.index)
+    dev = candles.device
+    key = (nsym, nwords, P, dev.index)
     bufs = _flag_cache.get(key)
-    if bufs is None or bufs[0].device != candles.device:
+    if bufs is None or bufs[0].device != dev:
         eflags = torch.empty((nsym, nwords, P), dtype=torch.int64,
-                             device=candles.device)
+                             device=dev)
         xflags = torch.empty_like(eflags)
         _flag_cache.clear()      # one shape live at a time (16 GB-class)
         _flag_cache[key] = (eflags, xflags)
     else:
         eflags, xflags = bufs
     metrics = torch.empty((P, nsym, NMETRIC), dtype=torch.float32,
-                          device=candles.device)
-    stream = torch.cuda.current_stream(candles.device).cuda_stream
-    ops.bt_flags(candles.data_ptr(), population.data_ptr(),
-                 eflags.data_ptr(), xflags.data_ptr(),
-                 nsym, T, P, nshards, tail, stream)
-    ops.bt_trades(candles.data_ptr(), population.data_ptr(),
-                  eflags.data_ptr(), xflags.data_ptr(),
-                  metrics.data_ptr(), nsym, T, P,
-                  float(initial_equity), stream)
+                          device=dev)
+
+    if sym_groups <= 0:
+        # 8 groups of >= 4 symbols pipelines well at the flagship shape;
+        # degenerate shapes fall back to the serial two-launch path
+        sym_groups = max(1, min(8, nsym // 4))
+    if sym_groups == 1:
+        stream = torch.cuda.current_stream(dev).cuda_stream
+        ops.bt_flags(candles.data_ptr(), population.data_ptr(),
+                     eflags.data_ptr(), xflags.data_ptr(),
+                     nsym, T, P, nshards, tail, stream)
+        ops.bt_trades(candles.data_ptr(), population.data_ptr(),
+                      eflags.data_ptr(), xflags.data_ptr(),
+                      metrics.data_ptr(), nsym, T, P,
+                      float(initial_equity), 0, nsym, stream)
+        return metrics
+
+    ss = _pipe_streams.get(dev.index)
+    if ss is None:
+        ss = (torch.cuda.Stream(dev), torch.cuda.Stream(dev))
+        _pipe_streams[dev.index] = ss
+    sf, st = ss
+    cur = torch.cuda.current_stream(dev)
+    start_ev = torch.cuda.Event()
+    start_ev.record(cur)
+    sf.wait_event(start_ev)
+    st.wait_event(start_ev)
+    bounds = [nsym * g // sym_groups for g in range(sym_groups + 1)]
+    cand_stride = T * 4
+    flag_stride = nwords * P
+    for g in range(sym_groups):
+        s0, s1 = bounds[g], bounds[g + 1]
+        ns = s1 - s0
+        if ns == 0:
+            continue
+        c_off = candles.data_ptr() + s0 * cand_stride * 4   # f32 bytes
+        e_off = eflags.data_ptr() + s0 * flag_stride * 8
+        x_off = xflags.data_ptr() + s0 * flag_stride * 8
+        ops.bt_flags(c_off, population.data_ptr(), e_off, x_off,
+                     ns, T, P, nshards, tail, sf.cuda_stream)
+        ev = torch.cuda.Event()
+        ev.record(sf)
+        st.wait_event(ev)
+        ops.bt_trades(c_off, population.data_ptr(), e_off, x_off,
+                      metrics.data_ptr(), ns, T, P,
+                      float(initial_equity), s0, nsym, st.cuda_stream)
+    done = torch.cuda.Event()
+    done.record(st)
+    cur.wait_event(done)
     return metrics
 
 

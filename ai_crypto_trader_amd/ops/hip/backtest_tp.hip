@@ -447,12 +447,13 @@ __global__ void __launch_bounds__(BT_BLOCK) bt_flags_kernel(
 // (XCD-affine mapping as in backtest.hip).
 // ---------------------------------------------------------------------
 __global__ void __launch_bounds__(BT_BLOCK) bt_trades_kernel(
-    const float* __restrict__ candles,
+    const float* __restrict__ candles,       // offset to symbol group
     const float* __restrict__ pop,
-    const unsigned long long* __restrict__ eflags,
+    const unsigned long long* __restrict__ eflags,  // offset to group
     const unsigned long long* __restrict__ xflags,
-    float* __restrict__ metrics,             // (P, nsym, NMETRIC)
-    int nsym, int T, int P, int chunks_per_sym, float initial_equity)
+    float* __restrict__ metrics,             // (P, nsym_stride, NMETRIC)
+    int nsym, int T, int P, int chunks_per_sym, float initial_equity,
+    int sym0, int nsym_stride)               // group offset + full stride
 {
 #pragma clang fp contract(off)
     __shared__ float sc[BT_TILE];
@@ -532,7 +533,9 @@ __global__ void __launch_bounds__(BT_BLOCK) bt_trades_kernel(
     }
 
     if (act)
-        st.finalize(metrics + ((long)p * nsym + sym) * BT_NMETRIC, T);
+        st.finalize(
+            metrics + ((long)p * nsym_stride + sym0 + sym) * BT_NMETRIC,
+            T);
 }
 
 }  // namespace
@@ -556,11 +559,11 @@ extern "C" void launch_bt_trades(const float* candles, const float* pop,
                                  const unsigned long long* eflags,
                                  const unsigned long long* xflags,
                                  float* metrics, int nsym, int T, int P,
-                                 float initial_equity,
-                                 hipStream_t stream) {
+                                 float initial_equity, int sym0,
+                                 int nsym_stride, hipStream_t stream) {
     int chunks = (P + BT_BLOCK - 1) / BT_BLOCK;
     hipLaunchKernelGGL(bt_trades_kernel, dim3(nsym * chunks),
                        dim3(BT_BLOCK), 0, stream, candles, pop, eflags,
                        xflags, metrics, nsym, T, P, chunks,
-                       initial_equity);
+                       initial_equity, sym0, nsym_stride);
 }

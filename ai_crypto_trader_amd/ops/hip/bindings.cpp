@@ -19,7 +19,7 @@ extern "C" void launch_bt_flags(const float*, const float*,
 extern "C" void launch_bt_trades(const float*, const float*,
                                  const unsigned long long*,
                                  const unsigned long long*, float*, int,
-                                 int, int, float, hipStream_t);
+                                 int, int, float, int, int, hipStream_t);
 extern "C" void launch_backtest(const float*, const float*, float*, int, int,
                                 int, float, hipStream_t);
 extern "C" void launch_ga_evolve(const float*, const float*, const int*,
@@ -119,20 +119,21 @@ PYBIND11_MODULE(_hip_ops, m) {
     m.def("bt_trades",
           [](uintptr_t candles, uintptr_t pop, uintptr_t eflags,
              uintptr_t xflags, uintptr_t metrics, int nsym, int T, int P,
-             float initial_equity, uintptr_t stream) {
+             float initial_equity, int sym0, int nsym_stride,
+             uintptr_t stream) {
               launch_bt_trades(
                   reinterpret_cast<const float*>(candles),
                   reinterpret_cast<const float*>(pop),
                   reinterpret_cast<const unsigned long long*>(eflags),
                   reinterpret_cast<const unsigned long long*>(xflags),
                   reinterpret_cast<float*>(metrics), nsym, T, P,
-                  initial_equity, as_stream(stream));
+                  initial_equity, sym0, nsym_stride, as_stream(stream));
               check(hipGetLastError(), "bt_trades launch");
           },
           py::arg("candles"), py::arg("pop"), py::arg("eflags"),
           py::arg("xflags"), py::arg("metrics"), py::arg("nsym"),
           py::arg("T"), py::arg("P"), py::arg("initial_equity"),
-          py::arg("stream"));
+          py::arg("sym0"), py::arg("nsym_stride"), py::arg("stream"));
 
     m.def("ga_evolve",
           [](uintptr_t pop, uintptr_t fitness, uintptr_t order,
