@@ -64,10 +64,20 @@ def main():
     nshards = args.nshards or pick_nshards(nsym, T, P, tail=args.tail)
     out["nshards"] = nshards
 
-    # full continuous pipeline
+    # full continuous pipeline (auto symbol-group pipelining)
     dt = timeit(lambda: run_backtest_continuous_gpu(
         c_t, p_t, nshards=nshards, tail=args.tail))
     out["continuous"] = {"s": dt, "gcandles_per_s": evals / dt / 1e9}
+    # group-count sweep
+    gsweep = {}
+    for g in (1, 2, 4, 8, 16):
+        if nsym // g < 1:
+            continue
+        d = timeit(lambda g=g: run_backtest_continuous_gpu(
+            c_t, p_t, nshards=nshards, tail=args.tail, sym_groups=g),
+            iters=3)
+        gsweep[g] = round(evals / d / 1e9, 1)
+    out["group_sweep_gcps"] = gsweep
 
     # per-kernel breakdown
     from ai_crypto_trader_amd.ops.backtest import _flag_cache
@@ -80,7 +90,8 @@ def main():
         xflags.data_ptr(), nsym, T, P, nshards, args.tail, stream))
     dt_t = timeit(lambda: ops.bt_trades(
         c_t.data_ptr(), p_t.data_ptr(), eflags.data_ptr(),
-        xflags.data_ptr(), metrics.data_ptr(), nsym, T, P, 1.0, stream))
+        xflags.data_ptr(), metrics.data_ptr(), nsym, T, P, 1.0, 0, nsym,
+        stream))
     out["flags_kernel"] = {"s": dt_f, "gcandles_per_s": evals / dt_f / 1e9}
     out["trades_kernel"] = {"s": dt_t, "gcandles_per_s": evals / dt_t / 1e9}
 
