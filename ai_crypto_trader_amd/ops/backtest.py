@@ -109,9 +109,12 @@ def run_backtest_continuous_gpu(
                           device=dev)
 
     if sym_groups <= 0:
-        # 8 groups of >= 4 symbols pipelines well at the flagship shape;
-        # degenerate shapes fall back to the serial two-launch path
-        sym_groups = max(1, min(8, nsym // 4))
+        # measured: symbol-group pipelining HURTS (the trades kernel is
+        # latency-bound per wave — its wall time is the length of the
+        # serial candle chain, independent of lane count — so G groups
+        # serialize G full-length chains: 169 G/s at G=1 vs 35 at G=8).
+        # Kept only as an explicit knob for the perf sweep.
+        sym_groups = 1
     if sym_groups == 1:
         stream = torch.cuda.current_stream(dev).cuda_stream
         ops.bt_flags(candles.data_ptr(), population.data_ptr(),

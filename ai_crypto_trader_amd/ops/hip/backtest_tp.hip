@@ -516,11 +516,23 @@ __global__ void __launch_bounds__(BT_BLOCK) bt_trades_kernel(
             const unsigned long long xwrd =
                 act ? xsym[w * P + p] : 0ull;
             if (wlen == 64) {
-#pragma unroll 4
+                // software-pipelined: candle k+1's LDS loads issue
+                // before candle k's state chain, so the ~60-cycle LDS
+                // latency never sits inside the serial dependency path
+                // (1 wave/SIMD -> nothing else hides it)
+                float c0 = sc[wbase], h0 = sh[wbase], l0 = sl[wbase];
+#pragma unroll 8
                 for (int k = 0; k < 64; ++k) {
                     const int i = wbase + k;
-                    st.pstep(sc[i], sh[i], sl[i],
+                    float c1 = 0.f, h1 = 0.f, l1 = 0.f;
+                    if (k < 63) {
+                        c1 = sc[i + 1];
+                        h1 = sh[i + 1];
+                        l1 = sl[i + 1];
+                    }
+                    st.pstep(c0, h0, l0,
                              (ewrd >> k) & 1ull, (xwrd >> k) & 1ull);
+                    c0 = c1; h0 = h1; l0 = l1;
                 }
             } else {
                 for (int k = 0; k < wlen; ++k) {
