@@ -69,7 +69,7 @@ def run_backtest_continuous_gpu(
     initial_equity: float = 1.0,
     nshards: int = 0,          # 0 = auto
     tail: int = 2048,
-    sym_groups: int = 0,       # 0 = auto; 1 = no pipelining
+    time_groups: int = 0,      # 0 = auto; 1 = serial two-launch path
 ) -> torch.Tensor:             # (P, nsym, NMETRIC) f32 cuda
     """Continuous (unsegmented) backtest via the time-parallel kernel
     pair (ops/hip/backtest_tp.hip): indicator/vote flags computed with
@@ -78,15 +78,22 @@ def run_backtest_continuous_gpu(
     machine over packed flag words. Same metrics contract as
     run_backtest_gpu / engine_cpu.run_backtest_cpu.
 
-    The two kernels are pipelined over symbol groups on two HIP
-    streams: trades(group g) starts as soon as flags(group g) is done,
-    so the latency-bound trades waves (1 wave/SIMD) execute co-resident
-    with the issue-bound flags waves of later groups and hide their
-    dependency stalls in the flags kernel's occupancy."""
+    Overlap: the flags shards launch in `time_groups` groups on stream
+    F; the trades kernel is resumable over candle ranges (bitwise state
+    carry in HBM) and chunk g launches on stream T as soon as flags
+    group g completes. The latency-bound trades chain (1 wave/SIMD,
+    ~50% issue-idle) thus walks candles co-resident with the
+    issue-bound flags waves instead of after them — the chip runs both
+    at once and total time approaches max(flags, trades) + first-group
+    latency. (Symbol-group pipelining was measured and rejected: the
+    trades wall time is the length of the serial candle chain,
+    independent of lane count, so symbol groups serialize G full-length
+    chains.)"""
     ops = require_hip_ops()
     assert candles.is_cuda and population.is_cuda
     assert candles.dtype == torch.float32
     assert population.dtype == torch.float32
+    assert tail % 256 == 0
     candles = candles.contiguous()
     population = population.contiguous()
     nsym, T, _ = candles.shape
@@ -101,29 +108,27 @@ def run_backtest_continuous_gpu(
         eflags = torch.empty((nsym, nwords, P), dtype=torch.int64,
                              device=dev)
         xflags = torch.empty_like(eflags)
+        carry = torch.empty((P, nsym, 18), dtype=torch.float32,
+                            device=dev)
         _flag_cache.clear()      # one shape live at a time (16 GB-class)
-        _flag_cache[key] = (eflags, xflags)
+        _flag_cache[key] = (eflags, xflags, carry)
     else:
-        eflags, xflags = bufs
+        eflags, xflags, carry = bufs
     metrics = torch.empty((P, nsym, NMETRIC), dtype=torch.float32,
                           device=dev)
+    cptr = candles.data_ptr()
+    pptr = population.data_ptr()
 
-    if sym_groups <= 0:
-        # measured: symbol-group pipelining HURTS (the trades kernel is
-        # latency-bound per wave — its wall time is the length of the
-        # serial candle chain, independent of lane count — so G groups
-        # serialize G full-length chains: 169 G/s at G=1 vs 35 at G=8).
-        # Kept only as an explicit knob for the perf sweep.
-        sym_groups = 1
-    if sym_groups == 1:
+    if time_groups <= 0:
+        time_groups = 4 if nshards >= 8 else 1
+    time_groups = min(time_groups, nshards)
+    if time_groups == 1:
         stream = torch.cuda.current_stream(dev).cuda_stream
-        ops.bt_flags(candles.data_ptr(), population.data_ptr(),
-                     eflags.data_ptr(), xflags.data_ptr(),
-                     nsym, T, P, nshards, tail, stream)
-        ops.bt_trades(candles.data_ptr(), population.data_ptr(),
-                      eflags.data_ptr(), xflags.data_ptr(),
+        ops.bt_flags(cptr, pptr, eflags.data_ptr(), xflags.data_ptr(),
+                     nsym, T, P, nshards, tail, 0, nshards, stream)
+        ops.bt_trades(cptr, pptr, eflags.data_ptr(), xflags.data_ptr(),
                       metrics.data_ptr(), nsym, T, P,
-                      float(initial_equity), 0, nsym, stream)
+                      float(initial_equity), 0, nsym, 0, T, 0, stream)
         return metrics
 
     ss = _pipe_streams.get(dev.index)
@@ -136,25 +141,26 @@ def run_backtest_continuous_gpu(
     start_ev.record(cur)
     sf.wait_event(start_ev)
     st.wait_event(start_ev)
-    bounds = [nsym * g // sym_groups for g in range(sym_groups + 1)]
-    cand_stride = T * 4
-    flag_stride = nwords * P
-    for g in range(sym_groups):
-        s0, s1 = bounds[g], bounds[g + 1]
-        ns = s1 - s0
-        if ns == 0:
-            continue
-        c_off = candles.data_ptr() + s0 * cand_stride * 4   # f32 bytes
-        e_off = eflags.data_ptr() + s0 * flag_stride * 8
-        x_off = xflags.data_ptr() + s0 * flag_stride * 8
-        ops.bt_flags(c_off, population.data_ptr(), e_off, x_off,
-                     ns, T, P, nshards, tail, sf.cuda_stream)
+    body4 = (T // nshards) // 4096 * 4096 if nshards > 1 else T
+    sbounds = [nshards * g // time_groups for g in range(time_groups + 1)]
+    evs = []
+    for g in range(time_groups):
+        s0, s1 = sbounds[g], sbounds[g + 1]
+        ops.bt_flags(cptr, pptr, eflags.data_ptr(), xflags.data_ptr(),
+                     nsym, T, P, nshards, tail, s0, s1 - s0,
+                     sf.cuda_stream)
         ev = torch.cuda.Event()
         ev.record(sf)
-        st.wait_event(ev)
-        ops.bt_trades(c_off, population.data_ptr(), e_off, x_off,
-                      metrics.data_ptr(), ns, T, P,
-                      float(initial_equity), s0, nsym, st.cuda_stream)
+        evs.append(ev)
+    for g in range(time_groups):
+        s0, s1 = sbounds[g], sbounds[g + 1]
+        t_lo = s0 * body4
+        t_hi = T if s1 == nshards else s1 * body4
+        st.wait_event(evs[g])
+        ops.bt_trades(cptr, pptr, eflags.data_ptr(), xflags.data_ptr(),
+                      metrics.data_ptr(), nsym, T, P,
+                      float(initial_equity), 0, nsym, t_lo, t_hi,
+                      carry.data_ptr(), st.cuda_stream)
     done = torch.cuda.Event()
     done.record(st)
     cur.wait_event(done)

@@ -181,6 +181,30 @@ struct TradeState {
         sum_ret = sum_ret2 = 0.f;
     }
 
+    // bitwise state carry between chunked launches (f32 registers
+    // stored/reloaded verbatim; in_pos as 0/1)
+    __device__ void save(float* __restrict__ s) const
+    {
+        s[0] = cash; s[1] = units; s[2] = in_pos ? 1.0f : 0.0f;
+        s[3] = entry_cost; s[4] = trail_arm;
+        s[5] = stop; s[6] = tp; s[7] = peak;
+        s[8] = equity; s[9] = max_eq; s[10] = max_dd;
+        s[11] = n_trades; s[12] = wins;
+        s[13] = gross_p; s[14] = gross_l;
+        s[15] = sum_ret; s[16] = sum_ret2;
+    }
+
+    __device__ void restore(const float* __restrict__ s)
+    {
+        cash = s[0]; units = s[1]; in_pos = s[2] != 0.0f;
+        entry_cost = s[3]; trail_arm = s[4];
+        stop = s[5]; tp = s[6]; peak = s[7];
+        equity = s[8]; max_eq = s[9]; max_dd = s[10];
+        n_trades = s[11]; wins = s[12];
+        gross_p = s[13]; gross_l = s[14];
+        sum_ret = s[15]; sum_ret2 = s[16];
+    }
+
     __device__ void pstep(float close, float high, float low,
                           bool ebit, bool xbit)
     {
@@ -263,7 +287,8 @@ __global__ void __launch_bounds__(BT_BLOCK) bt_flags_kernel(
     const float* __restrict__ pop,          // (P, NPARAM)
     unsigned long long* __restrict__ eflags,  // (nsym, nwords, P)
     unsigned long long* __restrict__ xflags,
-    int nsym, int T, int P, int chunks, int nshards, int body4, int tail)
+    int nsym, int T, int P, int chunks, int nshards, int body4, int tail,
+    int shard0)                             // first shard of this launch
 {
 #pragma clang fp contract(off)
     __shared__ float chist[BT_SPAN];
@@ -271,7 +296,7 @@ __global__ void __launch_bounds__(BT_BLOCK) bt_flags_kernel(
     __shared__ float4 sh_vote[BT_TILE];
 
     const int bid = blockIdx.x;
-    const int shard = bid / (nsym * chunks);
+    const int shard = shard0 + bid / (nsym * chunks);
     const int rem = bid % (nsym * chunks);
     const int sym = rem / chunks;
     const int chunk = rem % chunks;
@@ -446,6 +471,9 @@ __global__ void __launch_bounds__(BT_BLOCK) bt_flags_kernel(
 // Kernel 2: trades. Grid = nsym x chunks blocks of 256 lanes
 // (XCD-affine mapping as in backtest.hip).
 // ---------------------------------------------------------------------
+// dynamic TradeState slots persisted between chunked launches
+#define BT_NSTATE 18
+
 __global__ void __launch_bounds__(BT_BLOCK) bt_trades_kernel(
     const float* __restrict__ candles,       // offset to symbol group
     const float* __restrict__ pop,
@@ -453,7 +481,9 @@ __global__ void __launch_bounds__(BT_BLOCK) bt_trades_kernel(
     const unsigned long long* __restrict__ xflags,
     float* __restrict__ metrics,             // (P, nsym_stride, NMETRIC)
     int nsym, int T, int P, int chunks_per_sym, float initial_equity,
-    int sym0, int nsym_stride)               // group offset + full stride
+    int sym0, int nsym_stride,               // group offset + full stride
+    int t_lo, int t_hi,                      // candle range [t_lo, t_hi)
+    float* __restrict__ state)               // (P, nsym, NSTATE) carry
 {
 #pragma clang fp contract(off)
     __shared__ float sc[BT_TILE];
@@ -478,13 +508,17 @@ __global__ void __launch_bounds__(BT_BLOCK) bt_trades_kernel(
 
     TradeState st;
     st.load(pop + (long)(act ? p : 0) * BT_NPARAM, initial_equity);
+    float* my_state =
+        state ? state + ((long)p * nsym + sym) * BT_NSTATE : nullptr;
+    if (t_lo > 0 && act)
+        st.restore(my_state);
 
     const float4* sym_candles =
         reinterpret_cast<const float4*>(candles + (long)sym * T * 4);
     const unsigned long long* esym = eflags + (long)sym * nwords * P;
     const unsigned long long* xsym = xflags + (long)sym * nwords * P;
 
-    for (int t0 = 0; t0 < T; t0 += BT_TILE) {
+    for (int t0 = t_lo; t0 < t_hi; t0 += BT_TILE) {
         __syncthreads();
         for (int i = tid; i < BT_TILE; i += BT_BLOCK) {
             const int t = t0 + i;
@@ -496,7 +530,7 @@ __global__ void __launch_bounds__(BT_BLOCK) bt_trades_kernel(
             }
         }
         __syncthreads();
-        const int tend = min(BT_TILE, T - t0);
+        const int tend = min(BT_TILE, t_hi - t0);
 
         for (int wq = 0; wq < BT_TILE / 64; ++wq) {
             const int wbase = wq * 64;
@@ -544,27 +578,36 @@ __global__ void __launch_bounds__(BT_BLOCK) bt_trades_kernel(
         }
     }
 
-    if (act)
-        st.finalize(
-            metrics + ((long)p * nsym_stride + sym0 + sym) * BT_NMETRIC,
-            T);
+    if (act) {
+        if (t_hi < T)
+            st.save(my_state);
+        else
+            st.finalize(
+                metrics +
+                    ((long)p * nsym_stride + sym0 + sym) * BT_NMETRIC,
+                T);
+    }
 }
 
 }  // namespace
 
+// shard-group launch: shards [shard0, shard0+nlaunch) of nshards total
 extern "C" void launch_bt_flags(const float* candles, const float* pop,
                                 unsigned long long* eflags,
                                 unsigned long long* xflags,
                                 int nsym, int T, int P, int nshards,
-                                int tail, hipStream_t stream) {
+                                int tail, int shard0, int nlaunch,
+                                hipStream_t stream) {
     int chunks = (P + BT_BLOCK - 1) / BT_BLOCK;
     int body4 = nshards > 1
                     ? (T / nshards) / BT_RESNAP * BT_RESNAP
                     : T;
+    if (nlaunch <= 0)
+        nlaunch = nshards - shard0;
     hipLaunchKernelGGL(bt_flags_kernel,
-                       dim3(nshards * nsym * chunks), dim3(BT_BLOCK), 0,
+                       dim3(nlaunch * nsym * chunks), dim3(BT_BLOCK), 0,
                        stream, candles, pop, eflags, xflags, nsym, T, P,
-                       chunks, nshards, body4, tail);
+                       chunks, nshards, body4, tail, shard0);
 }
 
 extern "C" void launch_bt_trades(const float* candles, const float* pop,
@@ -572,10 +615,12 @@ extern "C" void launch_bt_trades(const float* candles, const float* pop,
                                  const unsigned long long* xflags,
                                  float* metrics, int nsym, int T, int P,
                                  float initial_equity, int sym0,
-                                 int nsym_stride, hipStream_t stream) {
+                                 int nsym_stride, int t_lo, int t_hi,
+                                 float* state, hipStream_t stream) {
     int chunks = (P + BT_BLOCK - 1) / BT_BLOCK;
     hipLaunchKernelGGL(bt_trades_kernel, dim3(nsym * chunks),
                        dim3(BT_BLOCK), 0, stream, candles, pop, eflags,
                        xflags, metrics, nsym, T, P, chunks,
-                       initial_equity, sym0, nsym_stride);
+                       initial_equity, sym0, nsym_stride, t_lo, t_hi,
+                       state);
 }

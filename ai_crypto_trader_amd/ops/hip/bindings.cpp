@@ -15,11 +15,13 @@ namespace py = pybind11;
 
 extern "C" void launch_bt_flags(const float*, const float*,
                                 unsigned long long*, unsigned long long*,
-                                int, int, int, int, int, hipStream_t);
+                                int, int, int, int, int, int, int,
+                                hipStream_t);
 extern "C" void launch_bt_trades(const float*, const float*,
                                  const unsigned long long*,
                                  const unsigned long long*, float*, int,
-                                 int, int, float, int, int, hipStream_t);
+                                 int, int, float, int, int, int, int,
+                                 float*, hipStream_t);
 extern "C" void launch_backtest(const float*, const float*, float*, int, int,
                                 int, float, hipStream_t);
 extern "C" void launch_ga_evolve(const float*, const float*, const int*,
@@ -103,37 +105,40 @@ PYBIND11_MODULE(_hip_ops, m) {
     m.def("bt_flags",
           [](uintptr_t candles, uintptr_t pop, uintptr_t eflags,
              uintptr_t xflags, int nsym, int T, int P, int nshards,
-             int tail, uintptr_t stream) {
+             int tail, int shard0, int nlaunch, uintptr_t stream) {
               launch_bt_flags(
                   reinterpret_cast<const float*>(candles),
                   reinterpret_cast<const float*>(pop),
                   reinterpret_cast<unsigned long long*>(eflags),
                   reinterpret_cast<unsigned long long*>(xflags), nsym, T,
-                  P, nshards, tail, as_stream(stream));
+                  P, nshards, tail, shard0, nlaunch, as_stream(stream));
               check(hipGetLastError(), "bt_flags launch");
           },
           py::arg("candles"), py::arg("pop"), py::arg("eflags"),
           py::arg("xflags"), py::arg("nsym"), py::arg("T"), py::arg("P"),
-          py::arg("nshards"), py::arg("tail"), py::arg("stream"));
+          py::arg("nshards"), py::arg("tail"), py::arg("shard0") = 0,
+          py::arg("nlaunch") = 0, py::arg("stream") = 0);
 
     m.def("bt_trades",
           [](uintptr_t candles, uintptr_t pop, uintptr_t eflags,
              uintptr_t xflags, uintptr_t metrics, int nsym, int T, int P,
-             float initial_equity, int sym0, int nsym_stride,
-             uintptr_t stream) {
+             float initial_equity, int sym0, int nsym_stride, int t_lo,
+             int t_hi, uintptr_t state, uintptr_t stream) {
               launch_bt_trades(
                   reinterpret_cast<const float*>(candles),
                   reinterpret_cast<const float*>(pop),
                   reinterpret_cast<const unsigned long long*>(eflags),
                   reinterpret_cast<const unsigned long long*>(xflags),
                   reinterpret_cast<float*>(metrics), nsym, T, P,
-                  initial_equity, sym0, nsym_stride, as_stream(stream));
+                  initial_equity, sym0, nsym_stride, t_lo, t_hi,
+                  reinterpret_cast<float*>(state), as_stream(stream));
               check(hipGetLastError(), "bt_trades launch");
           },
           py::arg("candles"), py::arg("pop"), py::arg("eflags"),
           py::arg("xflags"), py::arg("metrics"), py::arg("nsym"),
           py::arg("T"), py::arg("P"), py::arg("initial_equity"),
-          py::arg("sym0"), py::arg("nsym_stride"), py::arg("stream"));
+          py::arg("sym0"), py::arg("nsym_stride"), py::arg("t_lo"),
+          py::arg("t_hi"), py::arg("state"), py::arg("stream"));
 
     m.def("ga_evolve",
           [](uintptr_t pop, uintptr_t fitness, uintptr_t order,

@@ -108,3 +108,26 @@ def test_continuous_in_ga_engine(dev):
     f1 = e1.eval_fitness().cpu().numpy()
     f2 = e2.eval_fitness().cpu().numpy()
     assert np.array_equal(f1, f2)
+
+
+def test_continuous_time_chunked_resume(dev):
+    """Chunked trades launches (bitwise state carry through HBM between
+    time groups) vs the single-launch path and the CPU engine."""
+    from ai_crypto_trader_amd.backtesting.engine_cpu import run_backtest_cpu
+    from ai_crypto_trader_amd.backtesting.strategy import random_population
+    from ai_crypto_trader_amd.ops.backtest import run_backtest_continuous_gpu
+
+    candles = _market(36864, 2, seed=31)     # 9 RESNAP periods
+    pop = random_population(32, seed=12)
+    m_cpu = run_backtest_cpu(candles, pop)
+    c_t = torch.from_numpy(candles).to(dev)
+    p_t = torch.from_numpy(pop).to(dev)
+    m1 = run_backtest_continuous_gpu(c_t, p_t, nshards=8,
+                                     time_groups=1).cpu().numpy()
+    m3 = run_backtest_continuous_gpu(c_t, p_t, nshards=8,
+                                     time_groups=3).cpu().numpy()
+    m8 = run_backtest_continuous_gpu(c_t, p_t, nshards=8,
+                                     time_groups=8).cpu().numpy()
+    assert np.array_equal(m_cpu, m1)
+    assert np.array_equal(m_cpu, m3)
+    assert np.array_equal(m_cpu, m8)

@@ -64,34 +64,35 @@ def main():
     nshards = args.nshards or pick_nshards(nsym, T, P, tail=args.tail)
     out["nshards"] = nshards
 
-    # full continuous pipeline (auto symbol-group pipelining)
+    # full continuous pipeline (auto time-group overlap)
     dt = timeit(lambda: run_backtest_continuous_gpu(
         c_t, p_t, nshards=nshards, tail=args.tail))
     out["continuous"] = {"s": dt, "gcandles_per_s": evals / dt / 1e9}
-    # group-count sweep
+    # time-group sweep (flags/trades overlap granularity)
     gsweep = {}
     for g in (1, 2, 4, 8, 16):
-        if nsym // g < 1:
+        if g > nshards:
             continue
         d = timeit(lambda g=g: run_backtest_continuous_gpu(
-            c_t, p_t, nshards=nshards, tail=args.tail, sym_groups=g),
+            c_t, p_t, nshards=nshards, tail=args.tail, time_groups=g),
             iters=3)
         gsweep[g] = round(evals / d / 1e9, 1)
-    out["group_sweep_gcps"] = gsweep
+    out["time_group_sweep_gcps"] = gsweep
 
     # per-kernel breakdown
     from ai_crypto_trader_amd.ops.backtest import _flag_cache
-    (eflags, xflags), = [v for v in _flag_cache.values()]
+    (eflags, xflags, carry), = [v for v in _flag_cache.values()]
     metrics = torch.empty((P, nsym, 10), dtype=torch.float32, device=dev)
     stream = torch.cuda.current_stream(dev).cuda_stream
 
     dt_f = timeit(lambda: ops.bt_flags(
         c_t.data_ptr(), p_t.data_ptr(), eflags.data_ptr(),
-        xflags.data_ptr(), nsym, T, P, nshards, args.tail, stream))
+        xflags.data_ptr(), nsym, T, P, nshards, args.tail, 0, nshards,
+        stream))
     dt_t = timeit(lambda: ops.bt_trades(
         c_t.data_ptr(), p_t.data_ptr(), eflags.data_ptr(),
         xflags.data_ptr(), metrics.data_ptr(), nsym, T, P, 1.0, 0, nsym,
-        stream))
+        0, T, 0, stream))
     out["flags_kernel"] = {"s": dt_f, "gcandles_per_s": evals / dt_f / 1e9}
     out["trades_kernel"] = {"s": dt_t, "gcandles_per_s": evals / dt_t / 1e9}
 
@@ -102,7 +103,7 @@ def main():
             continue
         d = timeit(lambda s=s: ops.bt_flags(
             c_t.data_ptr(), p_t.data_ptr(), eflags.data_ptr(),
-            xflags.data_ptr(), nsym, T, P, s, args.tail, stream),
+            xflags.data_ptr(), nsym, T, P, s, args.tail, 0, s, stream),
             iters=3)
         sweep[s] = round(evals / d / 1e9, 1)
     out["flags_shard_sweep_gcps"] = sweep

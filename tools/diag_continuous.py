@@ -77,7 +77,8 @@ def main():
     for ns in (1, 4):
         eflags.zero_(); xflags.zero_()
         ops.bt_flags(c_t.data_ptr(), p_t.data_ptr(), eflags.data_ptr(),
-                     xflags.data_ptr(), nsym, T, P, ns, 2048, stream)
+                     xflags.data_ptr(), nsym, T, P, ns, 2048, 0, ns,
+                     stream)
         torch.cuda.synchronize()
         print(f"[diag] bt_flags ns={ns} ok", flush=True)
         ef = unpack_flags(eflags, T)
@@ -95,7 +96,7 @@ def main():
     x_cpu_t = pack_flags(xf_cpu).to(dev)
     ops.bt_trades(c_t.data_ptr(), p_t.data_ptr(), e_cpu_t.data_ptr(),
                   x_cpu_t.data_ptr(), metrics.data_ptr(), nsym, T, P,
-                  1.0, stream)
+                  1.0, 0, nsym, 0, T, 0, stream)
     torch.cuda.synchronize()
     print("[diag] bt_trades(cpu flags) ok", flush=True)
     m_tr = metrics.cpu().numpy()
