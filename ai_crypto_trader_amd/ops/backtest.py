@@ -120,7 +120,7 @@ def run_backtest_continuous_gpu(
     pptr = population.data_ptr()
 
     if time_groups <= 0:
-        time_groups = 4 if nshards >= 8 else 1
+        time_groups = 8 if nshards >= 8 else 1
     time_groups = min(time_groups, nshards)
     if time_groups == 1:
         stream = torch.cuda.current_stream(dev).cuda_stream
@@ -142,7 +142,22 @@ def run_backtest_continuous_gpu(
     sf.wait_event(start_ev)
     st.wait_event(start_ev)
     body4 = (T // nshards) // 4096 * 4096 if nshards > 1 else T
-    sbounds = [nshards * g // time_groups for g in range(time_groups + 1)]
+    # geometric group sizes (1,1,2,4,...): the first trades chunk starts
+    # after one shard of flags (~1/nshards of the flags time) while the
+    # later, bigger flags launches keep the chip oversubscribed
+    sizes = []
+    left, s = nshards, 1
+    while left > 0 and len(sizes) < time_groups - 1:
+        s = min(s, left)
+        sizes.append(s)
+        left -= s
+        s *= 2
+    if left > 0:
+        sizes.append(left)
+    sbounds = [0]
+    for s in sizes:
+        sbounds.append(sbounds[-1] + s)
+    time_groups = len(sizes)
     evs = []
     for g in range(time_groups):
         s0, s1 = sbounds[g], sbounds[g + 1]

@@ -256,7 +256,14 @@ struct TradeState {
         sum_ret2 += r * r;
         equity = new_eq;
         max_eq = fmaxf(max_eq, equity);
-        max_dd = fmaxf(max_dd, (max_eq - equity) / max_eq);
+        // drawdown with the IEEE division only on candidate records:
+        // fl(x/m) > d  ⟹  x > fl(fl(d*m)*(1-5e-7))  (conservative
+        // over-trigger; the exact fmax(fl(x/m)) then decides, which is
+        // idempotent when the candidate is not a new record) — removes
+        // an 8-instruction serial divide from the common mark path
+        float x = max_eq - equity;
+        if (x > max_dd * max_eq * 0.9999995f)
+            max_dd = fmaxf(max_dd, x / max_eq);
     }
 
     __device__ void finalize(float* __restrict__ out, int T) const
@@ -292,6 +299,8 @@ __global__ void __launch_bounds__(BT_BLOCK) bt_flags_kernel(
 {
 #pragma clang fp contract(off)
     __shared__ float chist[BT_SPAN];
+    __shared__ double csq[BT_SPAN];   // (double)close^2, shared: saves
+                                      // two f64 muls per candle per lane
     __shared__ float hl[BT_SPAN][2];
     __shared__ float4 sh_vote[BT_TILE];
 
@@ -334,10 +343,12 @@ __global__ void __launch_bounds__(BT_BLOCK) bt_flags_kernel(
             if (t >= 0 && t < T) {
                 float4 c = sym_candles[t];
                 chist[i] = c.x;
+                csq[i] = (double)c.x * (double)c.x;
                 hl[i][0] = c.y;
                 hl[i][1] = c.z;
             } else {
                 chist[i] = 0.0f;
+                csq[i] = 0.0;
                 hl[i][0] = 0.0f;
                 hl[i][1] = 0.0f;
             }
@@ -420,10 +431,13 @@ __global__ void __launch_bounds__(BT_BLOCK) bt_flags_kernel(
                 double old = (double)chist[tt + BT_HALO - st.bb_w];
                 double c64 = (double)close;
                 st.bb_sum += c64 - old;
-                st.bb_sum2 += c64 * c64 - old * old;
+                st.bb_sum2 += csq[tt + BT_HALO] -
+                              csq[tt + BT_HALO - st.bb_w];
             }
             double inv_cnt = st.inv_w;
-            if (t < BT_MAXWIN && t + 1 < st.bb_w)
+            // the early-window divisor only exists in the very first
+            // tile (t < MAX_WIN <= 32 < BT_TILE): scalar-uniform guard
+            if (t0 == 0 && t < BT_MAXWIN && t + 1 < st.bb_w)
                 inv_cnt = 1.0 / (t + 1.0);
             double mean64 = st.bb_sum * inv_cnt;
             double var64 =
