@@ -256,14 +256,10 @@ struct TradeState {
         sum_ret2 += r * r;
         equity = new_eq;
         max_eq = fmaxf(max_eq, equity);
-        // drawdown with the IEEE division only on candidate records:
-        // fl(x/m) > d  ⟹  x > fl(fl(d*m)*(1-5e-7))  (conservative
-        // over-trigger; the exact fmax(fl(x/m)) then decides, which is
-        // idempotent when the candidate is not a new record) — removes
-        // an 8-instruction serial divide from the common mark path
-        float x = max_eq - equity;
-        if (x > max_dd * max_eq * 0.9999995f)
-            max_dd = fmaxf(max_dd, x / max_eq);
+        // (a division-free record-trigger variant was measured SLOWER:
+        // the extra branch defeats cross-iteration pipelining — the
+        // straight-line divide wins at 1 wave/SIMD)
+        max_dd = fmaxf(max_dd, (max_eq - equity) / max_eq);
     }
 
     __device__ void finalize(float* __restrict__ out, int T) const
