@@ -205,12 +205,15 @@ struct TradeState {
         sum_ret = s[15]; sum_ret2 = s[16];
     }
 
-    __device__ void pstep(float close, float high, float low,
-                          bool ebit, bool xbit)
+    // producer half of the per-candle step: position management + the
+    // equity mark (engine_cpu sections 3 + the new_eq assignment of 4).
+    // Returns new_eq for the consumer wave, which owns the r/drawdown
+    // divisions and the return accumulators.
+    __device__ float pstep_pos(float close, float high, float low,
+                               bool ebit, bool xbit)
     {
 #pragma clang fp contract(off)
         const bool pos0 = in_pos;
-        // --- position management (engine_cpu section 3, branchless) --
         float peak2 = fmaxf(peak, high);
         peak = pos0 ? peak2 : peak;
         bool trail_on = pos0 && trail_en && (peak >= trail_arm);
@@ -247,38 +250,9 @@ struct TradeState {
             in_pos = true;
         }
 
-        // --- mark to market (always, like the numpy engine: flat lanes
-        // compute r = cash/cash - 1 == 0 exactly, all accumulators
-        // unchanged) ------------------------------------------------
         float new_eq = cash + units * close;
-        float r = new_eq / equity - 1.0f;
-        sum_ret += r;
-        sum_ret2 += r * r;
         equity = new_eq;
-        max_eq = fmaxf(max_eq, equity);
-        // (a division-free record-trigger variant was measured SLOWER:
-        // the extra branch defeats cross-iteration pipelining — the
-        // straight-line divide wins at 1 wave/SIMD)
-        max_dd = fmaxf(max_dd, (max_eq - equity) / max_eq);
-    }
-
-    __device__ void finalize(float* __restrict__ out, int T) const
-    {
-#pragma clang fp contract(off)
-        // identical to backtest.hip BtState::finalize
-        float n = (float)max(T, 1);
-        float mean_r = sum_ret / n;
-        float var_r = fmaxf(sum_ret2 / n - mean_r * mean_r, 0.0f);
-        float sharpe = mean_r / fmaxf(sqrtf(var_r), BT_EPS) * BT_ANNUALIZE;
-        if (!(n_trades > 0.0f)) sharpe = 0.0f;
-        float win_rate = wins / fmaxf(n_trades, 1.0f);
-        float fitness = (n_trades > 0.0f)
-                            ? sharpe + win_rate - 2.0f * max_dd
-                            : -1.0f;
-        out[0] = equity; out[1] = n_trades; out[2] = wins;
-        out[3] = gross_p; out[4] = gross_l; out[5] = max_dd;
-        out[6] = sum_ret; out[7] = sum_ret2;
-        out[8] = sharpe; out[9] = fitness;
+        return new_eq;
     }
 };
 
@@ -478,13 +452,27 @@ __global__ void __launch_bounds__(BT_BLOCK) bt_flags_kernel(
 }
 
 // ---------------------------------------------------------------------
-// Kernel 2: trades. Grid = nsym x chunks blocks of 256 lanes
+// Kernel 2: trades. Grid = nsym x chunks blocks of 512 threads
 // (XCD-affine mapping as in backtest.hip).
+//
+// Producer-consumer split: waves 0-3 (tid 0-255, one lane per param)
+// run the position state machine; waves 4-7 (tid 256-511, same params
+// mirrored) run the mark accumulation — the two serial IEEE divisions
+// (per-candle return r and drawdown) plus the sum_ret/sum_ret2/max_eq/
+// max_dd chains. The equity series crosses waves through a
+// double-buffered LDS tile (one 64-candle word per phase), so each
+// SIMD hosts one producer and one consumer wave and the ~50% of issue
+// slots the single-wave version lost to dependency stalls are filled
+// by the other wave. All f32 values and op order are bit-identical to
+// engine_cpu (the consumer reads the exact f32 equity values the
+// producer assigned).
 // ---------------------------------------------------------------------
-// dynamic TradeState slots persisted between chunked launches
+// dynamic TradeState slots persisted between chunked launches:
+// producer owns [0..12], consumer owns [13..16]
 #define BT_NSTATE 18
+#define BT_TBLOCK 512
 
-__global__ void __launch_bounds__(BT_BLOCK) bt_trades_kernel(
+__global__ void __launch_bounds__(BT_TBLOCK) bt_trades_kernel(
     const float* __restrict__ candles,       // offset to symbol group
     const float* __restrict__ pop,
     const unsigned long long* __restrict__ eflags,  // offset to group
@@ -499,6 +487,8 @@ __global__ void __launch_bounds__(BT_BLOCK) bt_trades_kernel(
     __shared__ float sc[BT_TILE];
     __shared__ float sh[BT_TILE];
     __shared__ float sl[BT_TILE];
+    __shared__ float eqbuf[2][64][256];      // equity handoff, 128 KB
+    __shared__ int skipw[2][4];              // per-producer-wave skip
 
     int bid = blockIdx.x;
     int sym, chunk;
@@ -512,7 +502,10 @@ __global__ void __launch_bounds__(BT_BLOCK) bt_trades_kernel(
         chunk = bid % chunks_per_sym;
     }
     const int tid = threadIdx.x;
-    const int p = chunk * BT_BLOCK + tid;
+    const bool is_prod = tid < 256;
+    const int lane = tid & 255;
+    const int wv = (lane >> 6);              // wave index within half
+    const int p = chunk * 256 + lane;
     const bool act = p < P;
     const long nwords = (T + 63) >> 6;
 
@@ -520,82 +513,147 @@ __global__ void __launch_bounds__(BT_BLOCK) bt_trades_kernel(
     st.load(pop + (long)(act ? p : 0) * BT_NPARAM, initial_equity);
     float* my_state =
         state ? state + ((long)p * nsym + sym) * BT_NSTATE : nullptr;
-    if (t_lo > 0 && act)
-        st.restore(my_state);
+
+    // consumer-side accumulators
+    float eq_prev = initial_equity;
+    float max_eq = initial_equity, max_dd = 0.f;
+    float sum_ret = 0.f, sum_ret2 = 0.f;
+    if (t_lo > 0 && act) {
+        if (is_prod) {
+            st.restore(my_state);
+        } else {
+            eq_prev = my_state[8];
+            max_eq = my_state[13];
+            max_dd = my_state[14];
+            sum_ret = my_state[15];
+            sum_ret2 = my_state[16];
+        }
+    }
 
     const float4* sym_candles =
         reinterpret_cast<const float4*>(candles + (long)sym * T * 4);
     const unsigned long long* esym = eflags + (long)sym * nwords * P;
     const unsigned long long* xsym = xflags + (long)sym * nwords * P;
 
-    for (int t0 = t_lo; t0 < t_hi; t0 += BT_TILE) {
-        __syncthreads();
-        for (int i = tid; i < BT_TILE; i += BT_BLOCK) {
-            const int t = t0 + i;
-            if (t < T) {
-                float4 c = sym_candles[t];
-                sc[i] = c.x;
-                sh[i] = c.y;
-                sl[i] = c.z;
-            }
-        }
-        __syncthreads();
-        const int tend = min(BT_TILE, t_hi - t0);
+    const int w_lo = t_lo >> 6;              // t_lo is tile-aligned
+    const int w_hi = (t_hi + 63) >> 6;
+    const int nw = w_hi - w_lo;
 
-        for (int wq = 0; wq < BT_TILE / 64; ++wq) {
-            const int wbase = wq * 64;
-            if (wbase >= tend) break;
-            const int wlen = min(64, tend - wbase);
-            const long w = (t0 + wbase) >> 6;
-            const unsigned long long ewrd =
-                act ? esym[w * P + p] : 0ull;
-            // exact whole-word skip: a lane is inert for the word if it
-            // is flat, settled (cash==equity -> mark is a no-op) and no
-            // entry bit is set; when every lane of the wave is inert the
-            // 64 candles are provably no-ops (engine_cpu semantics)
-            const bool busy =
-                st.in_pos || (st.cash != st.equity) || ewrd != 0ull;
-            if (__ballot(busy) == 0ull)
-                continue;
-            const unsigned long long xwrd =
-                act ? xsym[w * P + p] : 0ull;
-            if (wlen == 64) {
-                // software-pipelined: candle k+1's LDS loads issue
-                // before candle k's state chain, so the ~60-cycle LDS
-                // latency never sits inside the serial dependency path
-                // (1 wave/SIMD -> nothing else hides it)
-                float c0 = sc[wbase], h0 = sh[wbase], l0 = sl[wbase];
-#pragma unroll 8
-                for (int k = 0; k < 64; ++k) {
-                    const int i = wbase + k;
-                    float c1 = 0.f, h1 = 0.f, l1 = 0.f;
-                    if (k < 63) {
-                        c1 = sc[i + 1];
-                        h1 = sh[i + 1];
-                        l1 = sl[i + 1];
-                    }
-                    st.pstep(c0, h0, l0,
-                             (ewrd >> k) & 1ull, (xwrd >> k) & 1ull);
-                    c0 = c1; h0 = h1; l0 = l1;
+    for (int wi = 0; wi <= nw; ++wi) {
+        const int wabs = w_lo + wi;
+        if (wi < nw && (wabs & 3) == 0) {
+            // stage the 256-candle tile this word opens (producers only
+            // read sc/sh/sl; consumers only read eqbuf — no conflict)
+            const int t0 = wabs << 6;
+            for (int i = tid; i < BT_TILE; i += BT_TBLOCK) {
+                const int t = t0 + i;
+                if (t < T) {
+                    float4 c = sym_candles[t];
+                    sc[i] = c.x;
+                    sh[i] = c.y;
+                    sl[i] = c.z;
                 }
-            } else {
-                for (int k = 0; k < wlen; ++k) {
-                    const int i = wbase + k;
-                    st.pstep(sc[i], sh[i], sl[i],
-                             (ewrd >> k) & 1ull, (xwrd >> k) & 1ull);
+            }
+            __syncthreads();
+        }
+        if (is_prod) {
+            if (wi < nw) {
+                const int buf = wi & 1;
+                const int base = (wabs & 3) << 6;
+                const int wlen = min(64, t_hi - (wabs << 6));
+                const unsigned long long ewrd =
+                    act ? esym[(long)wabs * P + p] : 0ull;
+                // exact whole-word skip: flat + settled + no entry bit
+                // for every lane of the wave -> the 64 candles are
+                // provably no-ops (engine_cpu semantics); the consumer
+                // skips via skipw (equity series constant)
+                const bool busy = st.in_pos ||
+                                  (st.cash != st.equity) || ewrd != 0ull;
+                const bool skip = __ballot(busy) == 0ull;
+                if ((lane & 63) == 0)
+                    skipw[buf][wv] = skip ? 1 : 0;
+                if (!skip) {
+                    const unsigned long long xwrd =
+                        act ? xsym[(long)wabs * P + p] : 0ull;
+#pragma unroll 4
+                    for (int k = 0; k < 64; ++k) {
+                        if (k >= wlen) break;
+                        const int i = base + k;
+                        float eq = st.pstep_pos(
+                            sc[i], sh[i], sl[i],
+                            (ewrd >> k) & 1ull, (xwrd >> k) & 1ull);
+                        eqbuf[buf][k][lane] = eq;
+                    }
+                }
+            }
+        } else if (wi > 0) {
+            const int buf = (wi - 1) & 1;
+            if (!skipw[buf][wv]) {
+                const int wprev = wabs - 1;
+                const int wlen = min(64, t_hi - (wprev << 6));
+#pragma unroll 4
+                for (int k = 0; k < 64; ++k) {
+                    if (k >= wlen) break;
+                    float eq = eqbuf[buf][k][lane];
+                    float r = eq / eq_prev - 1.0f;
+                    sum_ret += r;
+                    sum_ret2 += r * r;
+                    eq_prev = eq;
+                    max_eq = fmaxf(max_eq, eq);
+                    max_dd = fmaxf(max_dd, (max_eq - eq) / max_eq);
                 }
             }
         }
+        __syncthreads();
     }
 
-    if (act) {
-        if (t_hi < T)
-            st.save(my_state);
-        else
-            st.finalize(
-                metrics +
-                    ((long)p * nsym_stride + sym0 + sym) * BT_NMETRIC,
-                T);
+    if (t_hi < T) {
+        if (act) {
+            if (is_prod) {
+                st.save(my_state);
+            } else {
+                my_state[13] = max_eq;
+                my_state[14] = max_dd;
+                my_state[15] = sum_ret;
+                my_state[16] = sum_ret2;
+            }
+        }
+        return;
+    }
+
+    // finalize: producer publishes its five trade metrics through LDS;
+    // the consumer computes sharpe/fitness and writes all ten
+    // (identical formulas to backtest.hip BtState::finalize)
+    if (is_prod) {
+        eqbuf[0][0][lane] = st.equity;
+        eqbuf[0][1][lane] = st.n_trades;
+        eqbuf[0][2][lane] = st.wins;
+        eqbuf[0][3][lane] = st.gross_p;
+        eqbuf[0][4][lane] = st.gross_l;
+    }
+    __syncthreads();
+    if (!is_prod && act) {
+        float equity = eqbuf[0][0][lane];
+        float n_trades = eqbuf[0][1][lane];
+        float wins = eqbuf[0][2][lane];
+        float gross_p = eqbuf[0][3][lane];
+        float gross_l = eqbuf[0][4][lane];
+        float n = (float)max(T, 1);
+        float mean_r = sum_ret / n;
+        float var_r = fmaxf(sum_ret2 / n - mean_r * mean_r, 0.0f);
+        float sharpe =
+            mean_r / fmaxf(sqrtf(var_r), BT_EPS) * BT_ANNUALIZE;
+        if (!(n_trades > 0.0f)) sharpe = 0.0f;
+        float win_rate = wins / fmaxf(n_trades, 1.0f);
+        float fitness = (n_trades > 0.0f)
+                            ? sharpe + win_rate - 2.0f * max_dd
+                            : -1.0f;
+        float* out =
+            metrics + ((long)p * nsym_stride + sym0 + sym) * BT_NMETRIC;
+        out[0] = equity; out[1] = n_trades; out[2] = wins;
+        out[3] = gross_p; out[4] = gross_l; out[5] = max_dd;
+        out[6] = sum_ret; out[7] = sum_ret2;
+        out[8] = sharpe; out[9] = fitness;
     }
 }
 
@@ -627,9 +685,9 @@ extern "C" void launch_bt_trades(const float* candles, const float* pop,
                                  float initial_equity, int sym0,
                                  int nsym_stride, int t_lo, int t_hi,
                                  float* state, hipStream_t stream) {
-    int chunks = (P + BT_BLOCK - 1) / BT_BLOCK;
+    int chunks = (P + 255) / 256;
     hipLaunchKernelGGL(bt_trades_kernel, dim3(nsym * chunks),
-                       dim3(BT_BLOCK), 0, stream, candles, pop, eflags,
+                       dim3(BT_TBLOCK), 0, stream, candles, pop, eflags,
                        xflags, metrics, nsym, T, P, chunks,
                        initial_equity, sym0, nsym_stride, t_lo, t_hi,
                        state);
