@@ -158,9 +158,8 @@ struct TradeState {
     float entry_cost;
     float trail_arm;   // entry_price * act_m (armed threshold)
     float stop, tp, peak;
-    float equity, max_eq, max_dd;
+    float equity;
     float n_trades, wins, gross_p, gross_l;
-    float sum_ret, sum_ret2;
 
     __device__ void load(const float* __restrict__ pr, float initial_equity)
     {
@@ -176,22 +175,22 @@ struct TradeState {
         in_pos = false;
         entry_cost = trail_arm = 0.f;
         stop = tp = peak = 0.f;
-        equity = initial_equity; max_eq = initial_equity; max_dd = 0.f;
+        equity = initial_equity;
         n_trades = wins = gross_p = gross_l = 0.f;
-        sum_ret = sum_ret2 = 0.f;
     }
 
     // bitwise state carry between chunked launches (f32 registers
-    // stored/reloaded verbatim; in_pos as 0/1)
+    // stored/reloaded verbatim; in_pos as 0/1). Producer owns slots
+    // [0..12]; the consumer wave owns [13..16] (max_eq, max_dd,
+    // sum_ret, sum_ret2) — disjoint, saved by different threads.
     __device__ void save(float* __restrict__ s) const
     {
         s[0] = cash; s[1] = units; s[2] = in_pos ? 1.0f : 0.0f;
         s[3] = entry_cost; s[4] = trail_arm;
         s[5] = stop; s[6] = tp; s[7] = peak;
-        s[8] = equity; s[9] = max_eq; s[10] = max_dd;
-        s[11] = n_trades; s[12] = wins;
-        s[13] = gross_p; s[14] = gross_l;
-        s[15] = sum_ret; s[16] = sum_ret2;
+        s[8] = equity;
+        s[9] = n_trades; s[10] = wins;
+        s[11] = gross_p; s[12] = gross_l;
     }
 
     __device__ void restore(const float* __restrict__ s)
@@ -199,10 +198,9 @@ struct TradeState {
         cash = s[0]; units = s[1]; in_pos = s[2] != 0.0f;
         entry_cost = s[3]; trail_arm = s[4];
         stop = s[5]; tp = s[6]; peak = s[7];
-        equity = s[8]; max_eq = s[9]; max_dd = s[10];
-        n_trades = s[11]; wins = s[12];
-        gross_p = s[13]; gross_l = s[14];
-        sum_ret = s[15]; sum_ret2 = s[16];
+        equity = s[8];
+        n_trades = s[9]; wins = s[10];
+        gross_p = s[11]; gross_l = s[12];
     }
 
     // producer half of the per-candle step: position management + the
