@@ -485,7 +485,9 @@ __global__ void __launch_bounds__(BT_TBLOCK) bt_trades_kernel(
     __shared__ float sc[BT_TILE];
     __shared__ float sh[BT_TILE];
     __shared__ float sl[BT_TILE];
-    __shared__ float eqbuf[2][64][256];      // equity handoff, 128 KB
+    // equity handoff in 32-candle half-words: 64 KB (not 128) so flags
+    // blocks can co-reside on the CU during the overlap schedule
+    __shared__ float eqbuf[2][32][256];
     __shared__ int skipw[2][4];              // per-producer-wave skip
 
     int bid = blockIdx.x;
@@ -533,16 +535,17 @@ __global__ void __launch_bounds__(BT_TBLOCK) bt_trades_kernel(
     const unsigned long long* esym = eflags + (long)sym * nwords * P;
     const unsigned long long* xsym = xflags + (long)sym * nwords * P;
 
-    const int w_lo = t_lo >> 6;              // t_lo is tile-aligned
-    const int w_hi = (t_hi + 63) >> 6;
-    const int nw = w_hi - w_lo;
+    const int h_lo = t_lo >> 5;              // t_lo is tile-aligned
+    const int h_hi = (t_hi + 31) >> 5;       // 32-candle half-words
+    const int nh = h_hi - h_lo;
+    unsigned long long ewrd = 0ull, xwrd = 0ull;
 
-    for (int wi = 0; wi <= nw; ++wi) {
-        const int wabs = w_lo + wi;
-        if (wi < nw && (wabs & 3) == 0) {
-            // stage the 256-candle tile this word opens (producers only
+    for (int hi_ = 0; hi_ <= nh; ++hi_) {
+        const int habs = h_lo + hi_;
+        if (hi_ < nh && (habs & 7) == 0) {
+            // stage the 256-candle tile this half opens (producers only
             // read sc/sh/sl; consumers only read eqbuf — no conflict)
-            const int t0 = wabs << 6;
+            const int t0 = habs << 5;
             for (int i = tid; i < BT_TILE; i += BT_TBLOCK) {
                 const int t = t0 + i;
                 if (t < T) {
@@ -555,43 +558,48 @@ __global__ void __launch_bounds__(BT_TBLOCK) bt_trades_kernel(
             __syncthreads();
         }
         if (is_prod) {
-            if (wi < nw) {
-                const int buf = wi & 1;
-                const int base = (wabs & 3) << 6;
-                const int wlen = min(64, t_hi - (wabs << 6));
-                const unsigned long long ewrd =
-                    act ? esym[(long)wabs * P + p] : 0ull;
-                // exact whole-word skip: flat + settled + no entry bit
-                // for every lane of the wave -> the 64 candles are
+            if (hi_ < nh) {
+                const int buf = hi_ & 1;
+                const int base = (habs & 7) << 5;
+                const int hlen = min(32, t_hi - (habs << 5));
+                if ((habs & 1) == 0) {
+                    const long w = habs >> 1;
+                    ewrd = act ? esym[w * P + p] : 0ull;
+                    xwrd = act ? xsym[w * P + p] : 0ull;
+                }
+                const int kbit = (habs & 1) << 5;
+                // exact half-word skip: flat + settled + no entry bit
+                // for every lane of the wave -> the 32 candles are
                 // provably no-ops (engine_cpu semantics); the consumer
                 // skips via skipw (equity series constant)
+                const unsigned long long ehalf =
+                    (ewrd >> kbit) & 0xffffffffull;
                 const bool busy = st.in_pos ||
-                                  (st.cash != st.equity) || ewrd != 0ull;
+                                  (st.cash != st.equity) || ehalf != 0ull;
                 const bool skip = __ballot(busy) == 0ull;
                 if ((lane & 63) == 0)
                     skipw[buf][wv] = skip ? 1 : 0;
                 if (!skip) {
-                    const unsigned long long xwrd =
-                        act ? xsym[(long)wabs * P + p] : 0ull;
 #pragma unroll 4
-                    for (int k = 0; k < 64; ++k) {
-                        if (k >= wlen) break;
+                    for (int k = 0; k < 32; ++k) {
+                        if (k >= hlen) break;
                         const int i = base + k;
                         float eq = st.pstep_pos(
                             sc[i], sh[i], sl[i],
-                            (ewrd >> k) & 1ull, (xwrd >> k) & 1ull);
+                            (ewrd >> (kbit + k)) & 1ull,
+                            (xwrd >> (kbit + k)) & 1ull);
                         eqbuf[buf][k][lane] = eq;
                     }
                 }
             }
-        } else if (wi > 0) {
-            const int buf = (wi - 1) & 1;
+        } else if (hi_ > 0) {
+            const int buf = (hi_ - 1) & 1;
             if (!skipw[buf][wv]) {
-                const int wprev = wabs - 1;
-                const int wlen = min(64, t_hi - (wprev << 6));
+                const int hprev = habs - 1;
+                const int hlen = min(32, t_hi - (hprev << 5));
 #pragma unroll 4
-                for (int k = 0; k < 64; ++k) {
-                    if (k >= wlen) break;
+                for (int k = 0; k < 32; ++k) {
+                    if (k >= hlen) break;
                     float eq = eqbuf[buf][k][lane];
                     float r = eq / eq_prev - 1.0f;
                     sum_ret += r;
