@@ -40,9 +40,26 @@ class HistoricalDataManager:
         d.mkdir(parents=True, exist_ok=True)
         return d / f"{symbol}_social.csv"
 
-    # --- fetch (offline: seeded synthetic; seam for a live fetcher) ------
+    # --- fetch: seeded synthetic (default, no egress here) or the live
+    # paginated Binance klines client (reference
+    # backtesting/data_manager.py:47-114) ---------------------------------
     def fetch_market_data(self, symbol: str, interval: str = "1m",
-                          n_candles: int = 10_000) -> pd.DataFrame:
+                          n_candles: int = 10_000,
+                          source: str = "synthetic",
+                          transport=None,
+                          start_ms: int | None = None,
+                          end_ms: int | None = None) -> pd.DataFrame:
+        if source == "binance":
+            from ..live.binance import fetch_klines
+
+            df = fetch_klines(symbol, interval, start_ms=start_ms,
+                              end_ms=end_ms, limit=n_candles,
+                              transport=transport)
+            # open column is not part of the store schema's value set
+            # used downstream; keep the venue's columns as returned
+            df = df[MARKET_COLS]
+            df.to_csv(self.market_path(symbol, interval), index=False)
+            return df
         seed = int(hashlib.sha1(
             f"{symbol}:{interval}".encode()).hexdigest()[:8], 16)
         ohlcv = generate_ohlcv(n_candles, 1, seed=seed)[0]

@@ -18,6 +18,9 @@ from pydantic import BaseModel, Field
 class TradingConfig(BaseModel):
     symbols: list[str] = Field(default_factory=lambda: ["BTCUSDC", "ETHUSDC"])
     quote_asset: str = "USDC"
+    # venue adapter: "fake" (offline sim) or "binance" (live REST via
+    # ai_crypto_trader_amd.live) — the only switch callers need
+    exchange: str = "fake"
     ai_analysis_interval: float = 60.0        # config.json:10
     min_price_change_pct: float = 0.5
     min_volume_usdc: float = 10_000.0

@@ -36,6 +36,13 @@ def build_service(name: str, bus, cfg: AppConfig, args):
         return SyntheticFeed(market, symbols, start=0, speed=args.speed)
 
     def exchange():
+        name = getattr(cfg.trading, "exchange", "fake")
+        if name == "binance":
+            import os
+            return ExchangeFactory.create_exchange(
+                "binance", quote=cfg.trading.quote_asset,
+                api_key=os.environ.get("BINANCE_API_KEY", ""),
+                api_secret=os.environ.get("BINANCE_API_SECRET", ""))
         return ExchangeFactory.create_exchange(
             "fake", fee_rate=cfg.trading.fee_rate,
             quote=cfg.trading.quote_asset)

@@ -192,6 +192,8 @@ class ExchangeFactory:
 
     @classmethod
     def create_exchange(cls, name: str = "fake", **kw) -> ExchangeInterface:
+        if name not in cls._registry and name == "binance":
+            from .. import live  # noqa: F401  registers the adapter
         if name not in cls._registry:
             raise ValueError(
                 f"unknown exchange '{name}' (known: {list(cls._registry)})")

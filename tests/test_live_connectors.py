@@ -1,0 +1,284 @@
+"""Live-connector tests: the Binance REST adapter, paginated klines
+client, miniTicker WS feed and social/news fetchers, all exercised
+offline through recorded-fixture replay (live/transport.py) and the
+schema-faithful BinanceFixtureServer — the integration style the
+reference never had (its tests required live credentials,
+SURVEY.md §4)."""
+
+import asyncio
+import json
+from pathlib import Path
+
+import pytest
+
+from ai_crypto_trader_amd.config import AppConfig
+from ai_crypto_trader_amd.live import (
+    BinanceExchange, BinanceFixtureServer, BinanceWSFeed,
+    LiveNewsHeadlines, LunarCrushClient, LunarCrushSocialSource,
+    RecordingTransport, ReplayTransport, fetch_klines,
+)
+from ai_crypto_trader_amd.utils.exchange import ExchangeFactory
+
+FIX = Path(__file__).parent / "fixtures"
+
+
+def make_exchange(server=None, **kw):
+    server = server or BinanceFixtureServer(
+        prices={"BTCUSDC": 50_000.0}, balances={"USDC": 10_000.0})
+    ex = BinanceExchange(api_key="test-key", api_secret="test-secret",
+                         transport=server, time_fn=lambda: 1.7e9, **kw)
+    return ex, server
+
+
+def test_factory_registers_binance():
+    ex = ExchangeFactory.create_exchange(
+        "binance", transport=BinanceFixtureServer(
+            prices={"BTCUSDC": 100.0}))
+    assert isinstance(ex, BinanceExchange)
+
+
+def test_signing_is_deterministic_hmac():
+    ex, server = make_exchange()
+    ex.get_balances()
+    method, path, params = server.requests[-1]
+    assert path == "/api/v3/account"
+    assert params["timestamp"] == 1.7e9 * 1000
+    # recompute the signature the venue would verify
+    import hashlib
+    import hmac as hm
+    import urllib.parse
+    qs = urllib.parse.urlencode(
+        {k: v for k, v in params.items() if k != "signature"})
+    want = hm.new(b"test-secret", qs.encode(),
+                  hashlib.sha256).hexdigest()
+    assert params["signature"] == want
+
+
+def test_ticker_depth_filters_fees():
+    ex, _ = make_exchange()
+    tk = ex.get_ticker("BTCUSDC")
+    assert tk["bid"] < tk["price"] < tk["ask"]
+    book = ex.get_order_book("BTCUSDC", limit=10)
+    assert len(book["bids"]) == 10 and book["bids"][0][0] > \
+        book["bids"][1][0]
+    f = ex.get_symbol_filters("BTCUSDC")
+    assert f["tick_size"] > 0 and f["step_size"] > 0
+    fees = ex.get_trading_fees("BTCUSDC")
+    assert fees["taker"] == pytest.approx(0.001)
+
+
+def test_order_lifecycle_market_stop_cancel():
+    """Reference executor order flow (trade_executor_service.py:909-992):
+    MARKET buy -> protective STOP_LOSS_LIMIT -> query -> cancel."""
+    ex, server = make_exchange()
+    o = ex.create_order("BTCUSDC", "BUY", "MARKET", 0.1)
+    assert o.status == "FILLED"
+    assert o.filled_price == pytest.approx(50_000.0)
+    bal = ex.get_balances()
+    assert bal["BTC"] == pytest.approx(0.1 * 0.999)
+    assert bal["USDC"] == pytest.approx(10_000 - 5_000)
+
+    stop = ex.create_order("BTCUSDC", "SELL", "STOP_LOSS_LIMIT",
+                           bal["BTC"], price=49_000 * 0.99,
+                           stop_price=49_000)
+    assert stop.status == "NEW"
+    q = ex.get_order("BTCUSDC", stop.order_id)
+    assert q is not None and q.status == "NEW" and \
+        q.stop_price == pytest.approx(49_000)
+    assert ex.cancel_order("BTCUSDC", stop.order_id)
+    q2 = ex.get_order("BTCUSDC", stop.order_id)
+    assert q2.status == "CANCELED"
+
+
+def test_stop_order_triggers_on_price_drop():
+    ex, server = make_exchange()
+    ex.create_order("BTCUSDC", "BUY", "MARKET", 0.1)
+    qty = ex.get_balances()["BTC"]
+    stop = ex.create_order("BTCUSDC", "SELL", "STOP_LOSS_LIMIT", qty,
+                           price=48_500, stop_price=49_000)
+    server.set_price("BTCUSDC", 48_900.0)    # through the stop
+    q = ex.get_order("BTCUSDC", stop.order_id)
+    assert q.status == "FILLED"
+    assert ex.get_balances().get("BTC", 0.0) == 0.0
+
+
+def test_executor_service_trades_through_binance_adapter():
+    """Full TradeExecutorService buy path through the REAL BinanceExchange
+    adapter against the wire-format fixture server (VERDICT item 2's
+    'done' criterion)."""
+    from ai_crypto_trader_amd.bus.message_bus import InProcessBus
+    from ai_crypto_trader_amd.services.trade_executor import (
+        TradeExecutorService,
+    )
+
+    ex, server = make_exchange()
+    cfg = AppConfig()
+    cfg.trading.min_confidence = 0.3
+    bus = InProcessBus()
+    svc = TradeExecutorService(bus, ex, cfg)
+
+    async def run():
+        ok, why = await svc.check_trading_conditions(
+            {"symbol": "BTCUSDC"})
+        assert ok, why
+        await svc.execute_buy({"symbol": "BTCUSDC", "decision": "BUY",
+                               "confidence": 0.9})
+        assert "BTCUSDC" in svc.active
+        trade = svc.active["BTCUSDC"]
+        # protective stop rests on the venue
+        q = ex.get_order("BTCUSDC", trade["stop_order_id"])
+        assert q is not None and q.status == "NEW"
+        # stop triggers on the venue; executor reconciles as stop_loss
+        server.set_price("BTCUSDC", trade["stop_price"] * 0.995)
+        await svc.execute_sell("BTCUSDC", "manual")
+        assert "BTCUSDC" not in svc.active
+        assert ex.get_balances().get("BTC", 0.0) == 0.0
+
+    asyncio.run(run())
+
+
+def test_klines_pagination_stitches_pages():
+    server = BinanceFixtureServer(prices={"BTCUSDC": 100.0})
+    server.kline_end_ms = 60_000 * 1500       # 1500 candles available
+    df = fetch_klines("BTCUSDC", "1m", start_ms=0, transport=server)
+    assert len(df) == 1500
+    # two pages requested: 1000 then 500
+    kreqs = [r for r in server.requests if r[1] == "/api/v3/klines"]
+    assert len(kreqs) == 2
+    assert int(kreqs[1][2]["startTime"]) == 1000 * 60_000
+    assert list(df.columns) == ["timestamp", "open", "high", "low",
+                                "close", "volume"]
+    assert (df["timestamp"].diff().dropna() == 60_000).all()
+
+
+def test_data_manager_binance_source():
+    import tempfile
+
+    from ai_crypto_trader_amd.backtesting.data_manager import (
+        HistoricalDataManager,
+    )
+
+    server = BinanceFixtureServer(prices={"BTCUSDC": 100.0})
+    with tempfile.TemporaryDirectory() as d:
+        dm = HistoricalDataManager(d)
+        df = dm.fetch_market_data("BTCUSDC", "1m", n_candles=250,
+                                  source="binance", transport=server,
+                                  start_ms=0)
+        assert len(df) == 250
+        again = dm.load_market_data("BTCUSDC", "1m")
+        assert len(again) == 250
+
+
+def test_ws_feed_replays_miniticker_frames():
+    from ai_crypto_trader_amd.live.binance import jsonl_frames
+
+    async def run():
+        feed = BinanceWSFeed(
+            jsonl_frames(FIX / "ws_miniticker.jsonl"),
+            symbols=["BTCUSDC", "ETHUSDC"])
+        out = []
+        async for c in feed:
+            out.append(c)
+        return out
+
+    candles = asyncio.run(run())
+    # 40 frames x 2 USDC symbols (BNBBTC filtered by quote)
+    assert len(candles) == 80
+    assert {c.symbol for c in candles} == {"BTCUSDC", "ETHUSDC"}
+    assert all(c.low <= c.close <= c.high for c in candles)
+    assert candles[-1].t == 39
+
+
+def test_market_monitor_consumes_ws_feed():
+    """The live WS feed drives the same market monitor the synthetic
+    feed does — market_updates come out with the reference schema."""
+    from ai_crypto_trader_amd.bus.message_bus import InProcessBus
+    from ai_crypto_trader_amd.bus.schema import Channels
+    from ai_crypto_trader_amd.live.binance import jsonl_frames
+    from ai_crypto_trader_amd.services.market_monitor import (
+        MarketMonitorService,
+    )
+
+    cfg = AppConfig()
+    bus = InProcessBus()
+    got = []
+
+    async def run():
+        sub = bus.subscribe(Channels.MARKET_UPDATES)
+        feed = BinanceWSFeed(jsonl_frames(FIX / "ws_miniticker.jsonl"),
+                             symbols=["BTCUSDC", "ETHUSDC"])
+        svc = MarketMonitorService(bus, feed, cfg)
+        task = asyncio.create_task(svc.start())
+        try:
+            while len(got) < 2:
+                chan, msg = await asyncio.wait_for(sub.get(), timeout=5)
+                got.append(msg)
+        finally:
+            await svc.stop()
+            task.cancel()
+
+    asyncio.run(run())
+    m = got[0]
+    assert m["symbol"] in ("BTCUSDC", "ETHUSDC")
+    assert "current_price" in m and "rsi" in m
+
+
+def test_lunarcrush_client_replay():
+    lc = LunarCrushClient(
+        transport=ReplayTransport(FIX / "lunarcrush_tape.jsonl"))
+    m = lc.asset_metrics("BTCUSDC")
+    assert m["social_volume"] == 182345
+    assert 0.0 <= m["sentiment"] <= 1.0
+    feeds = lc.feeds()
+    assert len(feeds) == 2 and feeds[0]["source"] == "lunarcrush"
+    # cache: second call serves without a new tape entry
+    assert lc.asset_metrics("BTCUSDC")["sentiment"] == m["sentiment"]
+
+
+def test_social_source_adapter_schema():
+    lc = LunarCrushClient(
+        transport=ReplayTransport(FIX / "lunarcrush_tape.jsonl"))
+    src = LunarCrushSocialSource(client=lc)
+    blk = src.metrics("BTCUSDC", 0)
+    assert blk.social_sentiment == pytest.approx(0.78)
+    assert blk.social_volume == 182345
+    assert blk.news_volume == 310
+
+
+def test_live_news_headlines_replay():
+    src = LiveNewsHeadlines(
+        transport=ReplayTransport(FIX / "news_tape.jsonl", strict=False))
+    heads = src.headlines("BTCUSDC", n=3)
+    assert len(heads) == 3
+    assert any("BTC" in h for h in heads)
+
+
+def test_news_service_scores_live_headlines():
+    from ai_crypto_trader_amd.services.news import NewsAnalyzer
+
+    src = LiveNewsHeadlines(
+        transport=ReplayTransport(FIX / "news_tape.jsonl", strict=False))
+    analyzer = NewsAnalyzer()
+    for h in src.headlines("BTCUSDC", n=3):
+        s = analyzer.sentiment(h)
+        assert 0.0 <= s <= 1.0
+
+
+def test_record_then_replay_roundtrip(tmp_path):
+    """RecordingTransport tape of a fixture-server session replays
+    identically through ReplayTransport — the mechanism that produces
+    checked-in fixtures from a real venue session."""
+    server = BinanceFixtureServer(prices={"BTCUSDC": 50_000.0})
+    tape = tmp_path / "tape.jsonl"
+    ex1 = BinanceExchange(api_key="k", api_secret="s",
+                          transport=RecordingTransport(server, tape),
+                          time_fn=lambda: 1.7e9)
+    t1 = ex1.get_ticker("BTCUSDC")
+    b1 = ex1.get_balances()
+    assert tape.exists() and len(tape.read_text().splitlines()) == 2
+
+    ex2 = BinanceExchange(api_key="k", api_secret="s",
+                          transport=ReplayTransport(tape),
+                          time_fn=lambda: 9.9e9)   # different timestamp
+    assert ex2.get_ticker("BTCUSDC") == t1
+    assert ex2.get_balances() == b1
