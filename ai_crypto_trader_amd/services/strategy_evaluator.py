@@ -158,6 +158,262 @@ class AIStrategyEvaluator:
         return jp
 
 
+STRATEGY_CODE_TEMPLATE = '''\
+# auto-generated strategy (proposer: {proposer})
+rsi_oversold = {rsi_oversold}
+rsi_overbought = {rsi_overbought}
+stop_loss_pct = {stop_loss_pct}
+take_profit_pct = {take_profit_pct}
+position_size_pct = {position_size_pct}
+
+def decide(ctx):
+    """Return BUY / SELL / HOLD from a per-candle context."""
+    rsi = ctx["rsi14"]
+    macd_hist = ctx["macd"] - ctx["macd_signal"]
+    if not ctx["in_position"]:
+        if rsi < rsi_oversold:
+            return "BUY"
+    else:
+        if rsi > rsi_overbought or macd_hist < 0:
+            return "SELL"
+    return "HOLD"
+'''
+
+
+class CodeStrategyCycle:
+    """Evaluate-improve cycle over strategy *code* (reference
+    ai_strategy_evaluator.py:732-1359 + strategy_evaluation_system.py
+    :358-432: GPT generates strategy code, statically evaluates it, runs
+    it through cross-validated simulation, applies improvement
+    suggestions to the code, iterates, and emits a comparison report).
+
+    Here the code artifact is a python `decide(ctx)` snippet; the
+    offline proposer is a deterministic template (the LocalAnalyst
+    stand-in — an LLM plugs into `propose_fn`/`suggest_fn`), the
+    simulator is backtesting/strategy_tester.py (per-candle context ->
+    BUY/SELL/HOLD callable), scoring is the evaluation system's metrics
+    over time folds and synthetic market conditions, and improvements
+    are applied by rewriting parameter assignments in the code text."""
+
+    def __init__(self, propose_fn=None, suggest_fn=None,
+                 report_dir: str = "evaluation_reports", seed: int = 0):
+        from ..backtesting.strategy_tester import StrategyTester
+
+        self.propose_fn = propose_fn or self._default_proposal
+        self.suggest_fn = suggest_fn or self._default_suggestions
+        self.tester = StrategyTester()
+        self.report_dir = Path(report_dir)
+        self.seed = seed
+
+    # --- generation (reference :100) ------------------------------------
+    @staticmethod
+    def _default_proposal(spec: dict) -> str:
+        base = {"rsi_oversold": 30.0, "rsi_overbought": 70.0,
+                "stop_loss_pct": 0.03, "take_profit_pct": 0.06,
+                "position_size_pct": 0.5, "proposer": "local"}
+        base.update(spec or {})
+        return STRATEGY_CODE_TEMPLATE.format(**base)
+
+    # --- static code evaluation (reference :148) ------------------------
+    @staticmethod
+    def static_eval(code: str) -> tuple[bool, list[str]]:
+        issues = []
+        if "def decide" not in code:
+            issues.append("no decide(ctx) entrypoint")
+        if "BUY" not in code or "SELL" not in code:
+            issues.append("code never trades")
+        params = params_from_code(code)
+        ok, val_issues = validate_strategy(params)
+        issues += val_issues
+        try:
+            compile(code, "<strategy>", "exec")
+        except SyntaxError as e:
+            issues.append(f"syntax error: {e}")
+        return (not issues), issues
+
+    @staticmethod
+    def compile_code(code: str):
+        """Compile the code artifact into the tester's decision
+        callable in a minimal namespace (no builtins beyond math-safe
+        ones — the reference 'deploys' to an isolated worker,
+        strategy_evolution_service.py:1465-1510)."""
+        ns: dict = {"__builtins__": {"abs": abs, "min": min, "max": max,
+                                     "float": float}}
+        exec(code, ns)                       # noqa: S102 (sandboxed ns)
+        fn = ns.get("decide")
+        if not callable(fn):
+            raise ValueError("strategy code defines no decide(ctx)")
+        return fn
+
+    # --- simulation scoring (reference :261 run CV) ---------------------
+    def run_code(self, code: str, candles: np.ndarray,
+                 k: int = 3) -> dict:
+        fn = self.compile_code(code)
+        T = len(candles)
+        fold_stats = []
+        for i in range(k):
+            fold = candles[i * T // k:(i + 1) * T // k]
+            res = self.tester.backtest_strategy(fold, fn)
+            fold_stats.append(res.stats)
+        sharpes = [f["sharpe"] for f in fold_stats]
+        mean_sharpe = float(np.mean(sharpes))
+        consistency = float(np.mean([s > 0 for s in sharpes]))
+        n_trades = int(sum(f["n_trades"] for f in fold_stats))
+        score = float(np.clip(mean_sharpe / 3.0, -1, 1) * 40 + 40
+                      + consistency * 20)
+        if n_trades == 0:
+            score = 10.0        # a strategy that never trades is inert
+        return {
+            "folds": fold_stats,
+            "mean_sharpe": mean_sharpe,
+            "consistency": consistency,
+            "n_trades": int(sum(f["n_trades"] for f in fold_stats)),
+            "win_rate": float(np.mean(
+                [f["win_rate"] for f in fold_stats])),
+            "quality": score,
+        }
+
+    # --- improvement (reference :535-731) -------------------------------
+    @staticmethod
+    def _default_suggestions(code: str, ev: dict) -> list[dict]:
+        params = params_from_code(code)
+        sugs = []
+        if ev["n_trades"] < 5:
+            sugs.append({
+                "reason": "too few trades: loosen RSI entry",
+                "changes": {"rsi_oversold":
+                            min(params.get("rsi_oversold", 30) + 8, 48)},
+            })
+        if ev["win_rate"] < 0.5 and ev["n_trades"] >= 5:
+            sugs.append({
+                "reason": "low win rate: tighten entry, wider TP",
+                "changes": {
+                    "rsi_oversold":
+                        max(params.get("rsi_oversold", 30) - 5, 10),
+                    "take_profit_pct":
+                        min(params.get("take_profit_pct", 0.06) * 1.5,
+                            0.3),
+                },
+            })
+        if ev["mean_sharpe"] < 0:
+            sugs.append({
+                "reason": "negative sharpe: cut risk",
+                "changes": {
+                    "position_size_pct":
+                        max(params.get("position_size_pct", 0.5) / 2,
+                            0.1),
+                    "stop_loss_pct":
+                        max(params.get("stop_loss_pct", 0.03) * 0.75,
+                            0.01),
+                },
+            })
+        if not sugs:
+            sugs.append({
+                "reason": "explore: slightly wider take profit",
+                "changes": {"take_profit_pct":
+                            min(params.get("take_profit_pct", 0.06)
+                                * 1.25, 0.3)},
+            })
+        return sugs
+
+    @staticmethod
+    def apply_to_code(code: str, changes: dict) -> str:
+        """Rewrite parameter assignments in the code text (the
+        improvement lands in the ARTIFACT, as the reference edits its
+        generated JS)."""
+        import re
+
+        for name, val in changes.items():
+            pat = re.compile(rf"^{name}\s*=\s*[-0-9.e]+",
+                             flags=re.MULTILINE)
+            if pat.search(code):
+                code = pat.sub(f"{name} = {val}", code)
+            else:
+                code = f"{name} = {val}\n" + code
+        return code
+
+    # --- the cycle (reference :732) -------------------------------------
+    def cycle(self, candles: np.ndarray, spec: dict | None = None,
+              rounds: int = 3) -> dict:
+        code = self.propose_fn(spec or {})
+        ok, issues = self.static_eval(code)
+        iterations = []
+        ev = self.run_code(code, candles)
+        iterations.append({"code": code, "eval": ev,
+                           "static_ok": ok, "static_issues": issues,
+                           "applied": "initial proposal"})
+        best = iterations[0]
+        frontier = iterations[0]
+        seen = {code}
+        for _ in range(rounds):
+            # suggestions explore from the FRONTIER (latest evaluated
+            # candidate) so non-improving rounds still move — the
+            # reference's cycle likewise iterates on the latest code,
+            # keeping the best separately (:732-900)
+            sugs = self.suggest_fn(frontier["code"], frontier["eval"])
+            any_new = False
+            for sug in sugs:
+                cand_code = self.apply_to_code(frontier["code"],
+                                               sug["changes"])
+                if cand_code in seen:
+                    continue        # converged on this edit already
+                seen.add(cand_code)
+                any_new = True
+                s_ok, s_issues = self.static_eval(cand_code)
+                if not s_ok:
+                    iterations.append({
+                        "code": cand_code, "eval": None,
+                        "static_ok": False, "static_issues": s_issues,
+                        "applied": sug["reason"]})
+                    continue
+                cand_ev = self.run_code(cand_code, candles)
+                iterations.append({"code": cand_code, "eval": cand_ev,
+                                   "static_ok": True,
+                                   "static_issues": [],
+                                   "applied": sug["reason"]})
+                frontier = iterations[-1]
+                if cand_ev["quality"] > best["eval"]["quality"]:
+                    best = iterations[-1]
+            if not any_new:
+                break               # every suggested edit already tried
+        return {
+            "best_code": best["code"],
+            "best_eval": best["eval"],
+            "iterations": len(iterations),
+            "initial_quality": iterations[0]["eval"]["quality"],
+            "final_quality": best["eval"]["quality"],
+            "improved":
+                best["eval"]["quality"]
+                > iterations[0]["eval"]["quality"],
+            "trail": [{"applied": it["applied"],
+                       "quality": (it["eval"] or {}).get("quality"),
+                       "static_ok": it["static_ok"]}
+                      for it in iterations],
+        }
+
+    # --- comparison report (reference :910-1359) ------------------------
+    def report(self, result: dict, name: str = "code-cycle") -> Path:
+        self.report_dir.mkdir(parents=True, exist_ok=True)
+        ts = int(time.time())
+        jp = self.report_dir / f"{name}-{ts}.json"
+        jp.write_text(json.dumps(
+            json.loads(json.dumps(result, default=float)), indent=2))
+        md = [f"# Strategy-code evaluate-improve cycle — {name}",
+              f"- iterations: {result['iterations']}",
+              f"- quality: {result['initial_quality']:.1f} -> "
+              f"{result['final_quality']:.1f} "
+              f"({'improved' if result['improved'] else 'no gain'})",
+              "", "## Trail"]
+        for it in result["trail"]:
+            q = "static-reject" if it["quality"] is None \
+                else f"{it['quality']:.1f}"
+            md.append(f"- {it['applied']}: {q}")
+        md += ["", "## Best strategy code", "```python",
+               result["best_code"], "```"]
+        (self.report_dir / f"{name}-{ts}.md").write_text("\n".join(md))
+        return jp
+
+
 PARAM_ALIASES = {
     # common spellings in generated strategy code -> canonical param name
     "rsi_period": "rsi_period", "rsiperiod": "rsi_period",

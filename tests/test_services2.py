@@ -617,3 +617,73 @@ def test_llm_evolution_seam():
         assert d["rsi_period"] == 9.0
 
     asyncio.run(go())
+
+
+def test_code_strategy_cycle_improves_and_registers(tmp_path):
+    """>=2 improvement iterations over strategy CODE
+    (generate -> static eval -> simulate via strategy_tester -> suggest
+    -> apply-to-code -> re-evaluate), report emitted, result round-trips
+    through the model registry (reference
+    ai_strategy_evaluator.py:732-1359)."""
+    import asyncio
+    import json as _json
+
+    import numpy as np
+
+    from ai_crypto_trader_amd.bus.message_bus import InProcessBus
+    from ai_crypto_trader_amd.data.synthetic import (
+        candles_chl_v, generate_ohlcv,
+    )
+    from ai_crypto_trader_amd.services.registry import ModelRegistryService
+    from ai_crypto_trader_amd.services.strategy_evaluator import (
+        CodeStrategyCycle,
+    )
+
+    candles = candles_chl_v(generate_ohlcv(6000, 1, seed=4))[0]
+    cyc = CodeStrategyCycle(report_dir=str(tmp_path), seed=1)
+
+    # deliberately weak initial spec so suggestions have room to act
+    res = cyc.cycle(candles, spec={"rsi_oversold": 12.0,
+                                   "take_profit_pct": 0.02},
+                    rounds=3)
+    assert res["iterations"] >= 3          # initial + >=2 improvements
+    assert res["best_eval"]["n_trades"] >= 0
+    assert "def decide" in res["best_code"]
+    # the improvement trail shows applied suggestions
+    assert len(res["trail"]) == res["iterations"]
+    assert res["trail"][1]["applied"] != "initial proposal"
+
+    rp = cyc.report(res, name="cycle-test")
+    assert rp.exists()
+    md = list(tmp_path.glob("cycle-test-*.md"))
+    assert md and "evaluate-improve" in md[0].read_text()
+
+    # registry round-trip
+    async def roundtrip():
+        bus = InProcessBus()
+        reg = ModelRegistryService(bus)
+        await reg.register(
+            "code-cycle-best", "strategy_code",
+            params={"quality": res["final_quality"],
+                    "code": res["best_code"]})
+        stored = await bus.get_json(
+            __import__("ai_crypto_trader_amd.bus.schema",
+                       fromlist=["Keys"]).Keys.MODEL_REGISTRY)
+        return stored
+
+    stored = asyncio.run(roundtrip())
+    s = _json.dumps(stored)
+    assert "code-cycle-best" in s and "def decide" in s
+
+
+def test_code_static_eval_rejects_bad_code():
+    from ai_crypto_trader_amd.services.strategy_evaluator import (
+        CodeStrategyCycle,
+    )
+
+    ok, issues = CodeStrategyCycle.static_eval("x = 1")
+    assert not ok and any("decide" in i for i in issues)
+    # no stop loss -> risk-management gate
+    code = CodeStrategyCycle._default_proposal({"stop_loss_pct": 0.0})
+    ok2, issues2 = CodeStrategyCycle.static_eval(code)
+    assert not ok2 and any("stop loss" in i for i in issues2)
