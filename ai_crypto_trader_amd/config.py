@@ -67,6 +67,9 @@ class NNConfig(BaseModel):
     lr: float = 1e-3
     retrain_interval_s: float = 86_400.0      # model_checkpoint_interval
     model_dir: str = "models_store"
+    # NN feature attribution (SHAP stand-in): grad_input |
+    # integrated_gradients (neural_network_service.py:957-1003)
+    attribution: str = "integrated_gradients"
 
 
 class EvolutionConfig(BaseModel):
@@ -86,7 +89,7 @@ class EvolutionConfig(BaseModel):
 
 
 class RegimeConfig(BaseModel):
-    method: str = "kmeans"                    # kmeans | gmm | hmm | rule
+    method: str = "kmeans"                    # kmeans | gmm | hmm | rf | rule
     n_regimes: int = 4
     lookback: int = 500
     retrain_interval_s: float = 3600.0

@@ -215,6 +215,47 @@ def label_clusters(X: np.ndarray, assign: np.ndarray, k: int) -> dict:
     return labels
 
 
+class RandomForestRegime:
+    """Supervised RandomForest regime classifier — the reference's 4th
+    detector mode (market_regime_detector.py:156): trained on labeled
+    historical windows, predicts the regime of the latest one.
+
+    Labels come from `label_windows` (rule-based labeling of each
+    historical window — the offline stand-in for the reference's
+    collected outcome labels, market_regime_data_collector.py:44-241);
+    a caller with real labeled data passes them straight to fit()."""
+
+    def __init__(self, n_estimators: int = 100, seed: int = 0):
+        from sklearn.ensemble import RandomForestClassifier
+
+        self.clf = RandomForestClassifier(
+            n_estimators=n_estimators, random_state=seed, n_jobs=1)
+        self.classes_: list[str] = []
+
+    @staticmethod
+    def label_windows(closes: np.ndarray, win: int = 32) -> np.ndarray:
+        """Rule-label each feature window (bull/bear/ranging/volatile)."""
+        n = len(closes) - win + 1
+        labels = np.empty(n, dtype=object)
+        for i in range(n):
+            labels[i] = rule_based_regime(closes[i:i + win])[0]
+        return labels
+
+    def fit(self, X: np.ndarray, y: np.ndarray):
+        self.clf.fit(X, y)
+        self.classes_ = list(self.clf.classes_)
+        return self
+
+    def predict(self, X: np.ndarray) -> np.ndarray:
+        return self.clf.predict(X)
+
+    def predict_proba(self, X: np.ndarray) -> np.ndarray:
+        return self.clf.predict_proba(X)
+
+    def feature_importances(self) -> np.ndarray:
+        return self.clf.feature_importances_
+
+
 def rule_based_regime(closes: np.ndarray) -> tuple[str, float]:
     """Rule-based thresholds (market_regime_service.py:503-604)."""
     if len(closes) < 30:
@@ -274,6 +315,18 @@ class MarketRegimeService(Service):
         X_np = extract_features(closes)
         if len(X_np) < 4 * self.config.regime.n_regimes:
             return rule_based_regime(closes)
+        if method == "rf":
+            # supervised mode (market_regime_detector.py:156): rule
+            # labels over historical windows -> RandomForest -> predict
+            # the latest window with a class-probability confidence
+            y = RandomForestRegime.label_windows(closes)
+            n = min(len(X_np), len(y))
+            rf = RandomForestRegime(seed=self.config.seed)
+            rf.fit(X_np[:n - 1], y[:n - 1])
+            proba = rf.predict_proba(X_np[n - 1:n])[0]
+            i = int(np.argmax(proba))
+            self.model = rf
+            return str(rf.classes_[i]), float(proba[i])
         X = torch.from_numpy(X_np)
         if self.device != "cpu":
             X = X.to(self.device)

@@ -10,6 +10,7 @@ keys. Feature scaling is min-max over the training window (reference
 
 from __future__ import annotations
 
+import time
 from pathlib import Path
 
 import numpy as np
@@ -205,10 +206,16 @@ class NeuralNetworkService(Service):
                      "macd_hist", "bb_position", "return_1", "atr"]
 
     def feature_importance(self, sym: str, candles: np.ndarray) -> dict | None:
-        """Gradient x input saliency per input feature — the offline
-        counterpart of the reference's SHAP DeepExplainer importances
-        (neural_network_service.py:957-1003): |d prediction / d feature *
-        feature| averaged over the window and batch, normalized to sum 1."""
+        """Per-input-feature attribution on the fused-LSTM predictor —
+        the offline counterpart of the reference's SHAP DeepExplainer
+        (neural_network_service.py:957-1003). Two methods behind
+        config.neural_network.attribution:
+          grad_input            |d pred / d feature * feature|
+          integrated_gradients  (x - x0) * mean_k grad(x0 + k/m (x-x0)),
+                                baseline x0 = batch-mean window; IG's
+                                completeness axiom makes attributions sum
+                                to f(x) - f(x0), the SHAP-style property
+        Both averaged over window steps and batch, normalized to sum 1."""
         model = self.models.get(sym)
         scaler = self.scalers.get(sym)
         if model is None or scaler is None:
@@ -219,12 +226,57 @@ class NeuralNetworkService(Service):
                             np.zeros(len(feats), np.float32), cfg.seq_len)
         if X is None:
             return None
-        x = torch.from_numpy(X[-64:]).to(self.device).requires_grad_(True)
-        model(x).sum().backward()
-        sal = (x.grad * x).abs().mean(dim=(0, 1))          # (F,)
+        x = torch.from_numpy(X[-64:]).to(self.device)
+        if getattr(cfg, "attribution", "grad_input") == \
+                "integrated_gradients":
+            sal = self._integrated_gradients(model, x)
+        else:
+            x = x.requires_grad_(True)
+            model(x).sum().backward()
+            sal = (x.grad * x).abs().mean(dim=(0, 1))       # (F,)
         sal = (sal / sal.sum().clamp_min(1e-12)).detach().cpu().numpy()
         return {name: float(v)
                 for name, v in zip(self.FEATURE_NAMES, sal)}
+
+    @staticmethod
+    def _integrated_gradients(model, x: torch.Tensor,
+                              steps: int = 24) -> torch.Tensor:
+        """Integrated gradients (Sundararajan et al.) along the straight
+        path from the batch-mean baseline to each window."""
+        x0 = x.mean(dim=0, keepdim=True).expand_as(x)
+        acc = torch.zeros_like(x[0, 0])
+        total = torch.zeros_like(x)
+        for k in range(1, steps + 1):
+            xi = (x0 + (k / steps) * (x - x0)).detach() \
+                .requires_grad_(True)
+            model(xi).sum().backward()
+            total = total + xi.grad
+        ig = (x - x0) * total / steps
+        acc = ig.abs().mean(dim=(0, 1))                     # (F,)
+        return acc
+
+    def importance_report(self, sym: str,
+                          candles: np.ndarray) -> dict | None:
+        """The `feature_importance` report shape the reference publishes
+        (README.md:420-468 / feature_importance_analyzer.py:610),
+        sourced from the NN attribution instead of the signal RF."""
+        imp = self.feature_importance(sym, candles)
+        if imp is None:
+            return None
+        ranked = sorted(imp.items(), key=lambda kv: -kv[1])
+        cfg = self.config.neural_network
+        return {
+            "symbol": sym,
+            "model": "lstm_price_predictor",
+            "method": getattr(cfg, "attribution", "grad_input"),
+            "feature_importance": imp,
+            "top_features": [k for k, _ in ranked[:5]],
+            "recommendations": [
+                f"feature '{ranked[-1][0]}' contributes "
+                f"{ranked[-1][1]:.1%} — candidate to prune"
+            ],
+            "at": time.time(),
+        }
 
     async def _train_loop(self):
         while self.running:

@@ -687,3 +687,70 @@ def test_code_static_eval_rejects_bad_code():
     code = CodeStrategyCycle._default_proposal({"stop_loss_pct": 0.0})
     ok2, issues2 = CodeStrategyCycle.static_eval(code)
     assert not ok2 and any("stop loss" in i for i in issues2)
+
+
+def test_rf_supervised_regime_mode():
+    """the supervised RandomForest regime classifier
+    (reference market_regime_detector.py:156) behind config
+    regime.method='rf'."""
+    import numpy as np
+
+    from ai_crypto_trader_amd.bus.message_bus import InProcessBus
+    from ai_crypto_trader_amd.config import AppConfig
+    from ai_crypto_trader_amd.services.market_regime import (
+        MarketRegimeService, RandomForestRegime,
+    )
+
+    rng = np.random.default_rng(3)
+    # clearly-trending market -> rule labels mostly 'bull'
+    closes = 100 * np.cumprod(1 + 0.002 + 0.002 * rng.standard_normal(600))
+    cfg = AppConfig()
+    cfg.regime.method = "rf"
+    svc = MarketRegimeService(InProcessBus(), cfg)
+    regime, conf = svc.detect(closes)
+    assert regime in ("bull", "bear", "ranging", "volatile")
+    assert 0.0 <= conf <= 1.0
+    assert isinstance(svc.model, RandomForestRegime)
+    # the trained forest exposes feature importances over the 6 features
+    fi = svc.model.feature_importances()
+    assert len(fi) == 6 and abs(fi.sum() - 1.0) < 1e-6
+    # trending data should not be labeled bear
+    assert regime != "bear"
+
+
+def test_nn_integrated_gradients_attribution():
+    """integrated-gradients attribution on the fused
+    LSTM predictor (SHAP DeepExplainer stand-in,
+    neural_network_service.py:957-1003), published in the
+    feature_importance report shape."""
+    import numpy as np
+
+    from ai_crypto_trader_amd.bus.message_bus import InProcessBus
+    from ai_crypto_trader_amd.config import AppConfig
+    from ai_crypto_trader_amd.data.synthetic import (
+        candles_chl_v, generate_ohlcv,
+    )
+    from ai_crypto_trader_amd.services.neural_network import (
+        NeuralNetworkService,
+    )
+
+    cfg = AppConfig()
+    cfg.neural_network.seq_len = 16
+    cfg.neural_network.attribution = "integrated_gradients"
+    svc = NeuralNetworkService(InProcessBus(), cfg)
+    candles = candles_chl_v(generate_ohlcv(600, 1, seed=7))[0]
+    svc.train("BTCUSDC", candles, epochs=1)
+    rep = svc.importance_report("BTCUSDC", candles)
+    assert rep is not None
+    assert rep["method"] == "integrated_gradients"
+    imp = rep["feature_importance"]
+    assert set(imp) == set(svc.FEATURE_NAMES)
+    assert abs(sum(imp.values()) - 1.0) < 1e-5
+    assert all(v >= 0 for v in imp.values())
+    assert len(rep["top_features"]) == 5 and rep["recommendations"]
+
+    # both methods behind config produce valid, differing attributions
+    cfg.neural_network.attribution = "grad_input"
+    rep2 = svc.importance_report("BTCUSDC", candles)
+    assert rep2["method"] == "grad_input"
+    assert abs(sum(rep2["feature_importance"].values()) - 1.0) < 1e-5
