@@ -754,3 +754,91 @@ def test_nn_integrated_gradients_attribution():
     rep2 = svc.importance_report("BTCUSDC", candles)
     assert rep2["method"] == "grad_input"
     assert abs(sum(rep2["feature_importance"].values()) - 1.0) < 1e-5
+
+
+def test_regime_transition_drives_strategy_switch_with_hysteresis():
+    """a regime transition (bull -> bear) propagates
+    regime service -> selection service -> strategy_switch publication +
+    strategy_params hot-swap, and the selection hysteresis blocks
+    marginal switches (reference market_regime_service.py:637-1113,
+    strategy_selection_service.py:884-935)."""
+    import asyncio
+
+    import numpy as np
+
+    from ai_crypto_trader_amd.bus.message_bus import InProcessBus
+    from ai_crypto_trader_amd.bus.schema import Channels, Keys
+    from ai_crypto_trader_amd.config import AppConfig
+    from ai_crypto_trader_amd.services.market_regime import (
+        MarketRegimeService,
+    )
+    from ai_crypto_trader_amd.services.strategy_selection import (
+        STRATEGY_PRESETS, StrategySelectionService,
+    )
+
+    cfg = AppConfig()
+    cfg.regime.method = "rule"
+    bus = InProcessBus()
+    regime_svc = MarketRegimeService(bus, cfg)
+    sel = StrategySelectionService(bus, cfg)
+
+    rng = np.random.default_rng(0)
+
+    def market(drift, n=400):
+        return 100 * np.cumprod(
+            1 + drift + 0.001 * rng.standard_normal(n))
+
+    async def run():
+        switches = []
+        sub = bus.subscribe(Channels.STRATEGY_SWITCH)
+
+        async def pump_regime(closes):
+            regime, conf = regime_svc.detect(closes)
+            vol = float(np.diff(np.log(closes[-100:])).std()
+                        * np.sqrt(525_600))
+            await bus.set(Keys.CURRENT_MARKET_REGIME,
+                          {"regime": regime, "confidence": conf,
+                           "volatility": min(vol, 2.0)})
+            return regime
+
+        # establish bull
+        r1 = await pump_regime(market(+0.003))
+        assert r1 == "bull"
+        sel.running = True
+        task = asyncio.create_task(sel.run())
+        try:
+            await asyncio.sleep(0.1)
+            base_strategy = sel.current
+            # transition to bear
+            r2 = await pump_regime(market(-0.003))
+            assert r2 == "bear"
+            for _ in range(50):
+                try:
+                    chan, msg = await asyncio.wait_for(sub.get(), 0.2)
+                    switches.append(msg)
+                    break
+                except asyncio.TimeoutError:
+                    continue
+        finally:
+            sel.running = False
+            task.cancel()
+        return base_strategy, switches
+
+    base_strategy, switches = asyncio.run(run())
+    assert switches, "regime transition produced no strategy_switch"
+    sw = switches[-1]
+    assert sw["market_regime"] == "bear"
+    assert sw["new_strategy_id"] != base_strategy
+    assert sel.current == sw["new_strategy_id"]
+    assert sel.current in STRATEGY_PRESETS
+
+    # hysteresis: equal/marginal scores must NOT switch
+    cur = sel.current
+    scores = {s: 1.0 for s in STRATEGY_PRESETS}
+    best = [s for s in STRATEGY_PRESETS if s != cur][0]
+    scores[best] = 1.0 + sel.min_improvement * 0.5     # below threshold
+    assert not sel.should_switch(best, scores)
+    scores[best] = 1.0 + sel.min_improvement * 2.0     # above threshold
+    assert sel.should_switch(best, scores)
+    # staying on the same strategy never switches
+    assert not sel.should_switch(cur, scores)
