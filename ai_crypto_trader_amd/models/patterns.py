@@ -138,6 +138,39 @@ class PatternRecognitionModel:
         self.trained = True
         return acc
 
+    # typical pattern lengths in candles — the reference's table
+    # (pattern_recognition.py:494-510)
+    TYPICAL_LENGTH = {
+        "head_and_shoulders": 30, "inverse_head_and_shoulders": 30,
+        "double_top": 25, "double_bottom": 25,
+        "ascending_triangle": 20, "descending_triangle": 20,
+        "symmetric_triangle": 20, "rectangle": 20,
+        "flag_bullish": 10, "flag_bearish": 10, "pennant": 15,
+        "cup_and_handle": 40, "rising_wedge": 25, "falling_wedge": 25,
+        "no_pattern": 1,
+    }
+    CONFIRMATION_PATTERNS = {
+        "double_top", "double_bottom", "head_and_shoulders",
+        "inverse_head_and_shoulders",
+    }
+
+    @classmethod
+    def estimate_completion(cls, closes: np.ndarray,
+                            pattern: str) -> float:
+        """Completion %% with the reference estimator's exact semantics
+        (pattern_recognition.py:476-530): data length over the typical
+        pattern length, a confirmation-move boost for reversal patterns
+        (mean |5-candle %%-change| / 10), rounded to the nearest 5%%."""
+        if pattern in ("no_pattern", "none"):
+            return 0.0
+        typical = cls.TYPICAL_LENGTH.get(pattern, 20)
+        completion = min(100.0, len(closes) / typical * 100.0)
+        if pattern in cls.CONFIRMATION_PATTERNS and len(closes) >= 6:
+            c = np.asarray(closes, np.float64)
+            recent = np.abs(np.diff(c[-6:]) / c[-6:-1]).mean() * 100.0
+            completion = min(100.0, completion * (1.0 + recent / 10.0))
+        return round(completion / 5.0) * 5.0
+
     @torch.no_grad()
     def detect(self, closes: np.ndarray) -> dict:
         """Windowed detection + completion estimate
@@ -154,8 +187,6 @@ class PatternRecognitionModel:
         conf = float(probs[ci])
         signal = ("bullish" if name in BULLISH else
                   "bearish" if name in BEARISH else "neutral")
-        # completion: how far along the window the pattern's extremum sits
-        extremum = int(np.argmax(w) if name in BEARISH else np.argmin(w))
-        completion = min(1.0, extremum / (WIN * 0.8))
+        completion = self.estimate_completion(closes, name)
         return {"pattern": name, "confidence": conf, "signal": signal,
                 "completion": completion}
