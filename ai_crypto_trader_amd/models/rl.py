@@ -340,6 +340,16 @@ class PPOAgent:
         self._ugraph = None
         self._ugraph_key = None
         self._uloss = None
+        # DDP gradient path: the eager update uses bucketed async
+        # all-reduce overlapped with backward (parallel/dist.py
+        # GradBucketer); the hip-graphed update keeps the capturable
+        # synchronous flat collective (_allreduce_grads) because hook-
+        # launched async ops cannot be graph-captured. Hooks register
+        # permanently, so the bucketer exists only in eager+dist mode.
+        self._bucketer = None
+        if pdist.is_dist() and not (use_graph and
+                                    self.device.type == "cuda"):
+            self._bucketer = pdist.GradBucketer(self.net.parameters())
 
     @torch.no_grad()
     def policy(self, obs):
@@ -455,8 +465,11 @@ class PPOAgent:
                 ent = dist_.entropy().mean()
                 loss = pi_loss + self.vf_coef * v_loss - self.ent_coef * ent
                 self.opt.zero_grad()
-                loss.backward()
-                self._allreduce_grads()
+                loss.backward()          # bucketer hooks overlap comm
+                if self._bucketer is not None:
+                    self._bucketer.finalize()
+                else:
+                    self._allreduce_grads()
                 self.opt.step()
                 stats["pi_loss"] += float(pi_loss.detach())
                 stats["v_loss"] += float(v_loss.detach())
