@@ -114,8 +114,12 @@ struct BtState {
         bb_sum2 = s2;
     }
 
+    // SKIP_BB is a compile-time flag (a resnap() just replaced the
+    // sums for this candle) so the steady-state loop carries no
+    // per-candle branch — the resnap candle is peeled at the call site.
+    template <bool SKIP_BB>
     __device__ void step(int t, float close, float high, float low,
-                         float oldc, float change, float4 sv, bool skip_bb)
+                         float oldc, float change, float4 sv)
     {
 #pragma clang fp contract(off)           // match the numpy f32 reference
         // --- 1. indicators -------------------------------------------
@@ -138,9 +142,8 @@ struct BtState {
         float rsi_den = avg_gain + fmaxf(avg_loss, BT_EPS);
 
         // Bollinger: close[t - W] comes from the shared halo tile
-        // (== the zero-initialized per-lane ring of engine_cpu.py).
-        // skip_bb: a resnap() just replaced the sums for this candle.
-        if (!skip_bb) {
+        // (== the zero-initialized per-lane ring of engine_cpu.py)
+        if (!SKIP_BB) {
             double old = (double)oldc;
             double c64 = (double)close;
             bb_sum += c64 - old;
@@ -348,7 +351,23 @@ __global__ void __launch_bounds__(BT_BLOCK) backtest_kernel(
                 fmaxf(hmax - lmin, BT_EPS), trend);
         }
         __syncthreads();
-        for (int tt = 0; tt < tend; ++tt) {
+        int tt0 = 0;
+        if (resnap_tile) {
+            // peeled resnap candle: BB increment skipped (template
+            // flag -> zero cost in the steady-state loop below)
+            const float close = chist[BT_HALO];
+            const float change = close - prev_close;
+            const float4 sv = sh_vote[0];
+#pragma unroll
+            for (int i = 0; i < ILP; ++i)
+                st[i].step<true>(t0, close, hl[BT_HALO][0],
+                                 hl[BT_HALO][1],
+                                 chist[BT_HALO - st[i].q.bb_w],
+                                 change, sv);
+            prev_close = close;
+            tt0 = 1;
+        }
+        for (int tt = tt0; tt < tend; ++tt) {
             const int t = t0 + tt;
             const float close = chist[tt + BT_HALO];
             const float high = hl[tt + BT_HALO][0];
@@ -357,9 +376,9 @@ __global__ void __launch_bounds__(BT_BLOCK) backtest_kernel(
             const float4 sv = sh_vote[tt];   // b128 broadcast
 #pragma unroll
             for (int i = 0; i < ILP; ++i)
-                st[i].step(t, close, high, low,
-                           chist[tt + BT_HALO - st[i].q.bb_w], change, sv,
-                           resnap_tile && tt == 0);
+                st[i].step<false>(t, close, high, low,
+                                  chist[tt + BT_HALO - st[i].q.bb_w],
+                                  change, sv);
             prev_close = close;
         }
     }

@@ -129,8 +129,68 @@ struct FlagState {
         bb_sum2 = s2;
     }
 
-    // (the full body step — indicators + votes + flag bits — is inlined
-    // in bt_flags_kernel so the packed-word bookkeeping stays local)
+    // Full body candle: indicators + votes -> (entry, exit) bits.
+    // SKIP_BB is compile-time (resnap peeled at the call site) so the
+    // steady-state loop carries no per-candle branch.
+    template <bool SKIP_BB>
+    __device__ uint2 candle(int t, float close, float change,
+                            float oldc, double csq_new, double csq_old,
+                            float4 sv)
+    {
+#pragma clang fp contract(off)
+        if (t == 0) { ema_f = close; ema_s = close; }
+        else {
+            ema_f += a_f * (close - ema_f);
+            ema_s += a_s * (close - ema_s);
+        }
+        float macd = ema_f - ema_s;
+        sig += a_sig * (macd - sig);
+        float macd_hist = macd - sig;
+
+        float gain = fmaxf(change, 0.0f);
+        float loss = fmaxf(-change, 0.0f);
+        avg_gain += (gain - avg_gain) * inv_rsi_p;
+        avg_loss += (loss - avg_loss) * inv_rsi_p;
+        float rsi_num = 100.0f * avg_gain;
+        float rsi_den = avg_gain + fmaxf(avg_loss, BT_EPS);
+
+        if (!SKIP_BB) {
+            double old = (double)oldc;
+            double c64 = (double)close;
+            bb_sum += c64 - old;
+            bb_sum2 += csq_new - csq_old;
+        }
+        double inv_cnt = inv_w;
+        // the early-window divisor only exists in the very first tile
+        // (t < MAX_WIN <= 32 < BT_TILE)
+        if (t < BT_MAXWIN && t + 1 < bb_w)
+            inv_cnt = 1.0 / (t + 1.0);
+        double mean64 = bb_sum * inv_cnt;
+        double var64 = fmax(bb_sum2 * inv_cnt - mean64 * mean64, 0.0);
+        float mean = (float)mean64;
+        float std_ = sqrtf((float)var64);
+        float band = bb_k * std_;
+        float bb_num = close - (mean - band);
+        float bb_den = fmaxf(2.0f * band, BT_EPS);
+
+        int net = 0;
+        if (t >= BT_WARMUP) {
+            int buy = (rsi_num < rsi_os * rsi_den) +
+                      (macd_hist > 0.0f) +
+                      (bb_num < bb_bth * bb_den) +
+                      (sv.x < stoch_os * sv.z) +
+                      (sv.y < will_os * sv.z) +
+                      (sv.w > 0.0f);
+            int sell = (rsi_num > rsi_ob * rsi_den) +
+                       (macd_hist < 0.0f) +
+                       (bb_num > bb_sth * bb_den) +
+                       (sv.x > stoch_ob * sv.z) +
+                       (sv.y > will_ob * sv.z) +
+                       (sv.w < 0.0f);
+            net = buy - sell;
+        }
+        return make_uint2(net >= entry_v, net <= -exit_v);
+    }
 };
 
 // ---------------------------------------------------------------------
@@ -372,72 +432,31 @@ __global__ void __launch_bounds__(BT_BLOCK) bt_flags_kernel(
         // BB resnap at RESNAP-aligned tiles (all shards see the same
         // aligned boundaries -> bit-identical to the sequential run)
         const bool resnap_tile = (t0 > 0) && ((t0 & (BT_RESNAP - 1)) == 0);
-        if (resnap_tile)
+        int tt0 = 0;
+        if (resnap_tile) {
             st.resnap(&chist[BT_HALO]);
-
-        for (int tt = 0; tt < tend; ++tt) {
+            // peeled resnap candle (compile-time SKIP_BB)
+            const float close = chist[BT_HALO];
+            uint2 b = st.candle<true>(
+                t0, close, close - prev_close,
+                chist[BT_HALO - st.bb_w], csq[BT_HALO],
+                csq[BT_HALO - st.bb_w], sh_vote[0]);
+            ew |= (unsigned long long)b.x << (t0 & 63);
+            xw |= (unsigned long long)b.y << (t0 & 63);
+            prev_close = close;
+            tt0 = 1;
+        }
+        for (int tt = tt0; tt < tend; ++tt) {
             const int t = t0 + tt;
             const float close = chist[tt + BT_HALO];
             const float change = (t == 0) ? 0.0f : close - prev_close;
-            const float4 sv = sh_vote[tt];
-
-            // indicators (same op order as backtest.hip BtState::step)
-            if (t == 0) { st.ema_f = close; st.ema_s = close; }
-            else {
-                st.ema_f += st.a_f * (close - st.ema_f);
-                st.ema_s += st.a_s * (close - st.ema_s);
-            }
-            float macd = st.ema_f - st.ema_s;
-            st.sig += st.a_sig * (macd - st.sig);
-            float macd_hist = macd - st.sig;
-
-            float gain = fmaxf(change, 0.0f);
-            float loss = fmaxf(-change, 0.0f);
-            st.avg_gain += (gain - st.avg_gain) * st.inv_rsi_p;
-            st.avg_loss += (loss - st.avg_loss) * st.inv_rsi_p;
-            float rsi_num = 100.0f * st.avg_gain;
-            float rsi_den = st.avg_gain + fmaxf(st.avg_loss, BT_EPS);
-
-            if (!(resnap_tile && tt == 0)) {
-                double old = (double)chist[tt + BT_HALO - st.bb_w];
-                double c64 = (double)close;
-                st.bb_sum += c64 - old;
-                st.bb_sum2 += csq[tt + BT_HALO] -
-                              csq[tt + BT_HALO - st.bb_w];
-            }
-            double inv_cnt = st.inv_w;
-            // the early-window divisor only exists in the very first
-            // tile (t < MAX_WIN <= 32 < BT_TILE): scalar-uniform guard
-            if (t0 == 0 && t < BT_MAXWIN && t + 1 < st.bb_w)
-                inv_cnt = 1.0 / (t + 1.0);
-            double mean64 = st.bb_sum * inv_cnt;
-            double var64 =
-                fmax(st.bb_sum2 * inv_cnt - mean64 * mean64, 0.0);
-            float mean = (float)mean64;
-            float std_ = sqrtf((float)var64);
-            float band = st.bb_k * std_;
-            float bb_num = close - (mean - band);
-            float bb_den = fmaxf(2.0f * band, BT_EPS);
-
-            int net = 0;
-            if (t >= BT_WARMUP) {
-                int buy = (rsi_num < st.rsi_os * rsi_den) +
-                          (macd_hist > 0.0f) +
-                          (bb_num < st.bb_bth * bb_den) +
-                          (sv.x < st.stoch_os * sv.z) +
-                          (sv.y < st.will_os * sv.z) +
-                          (sv.w > 0.0f);
-                int sell = (rsi_num > st.rsi_ob * rsi_den) +
-                           (macd_hist < 0.0f) +
-                           (bb_num > st.bb_sth * bb_den) +
-                           (sv.x > st.stoch_ob * sv.z) +
-                           (sv.y > st.will_ob * sv.z) +
-                           (sv.w < 0.0f);
-                net = buy - sell;
-            }
+            uint2 b = st.candle<false>(
+                t, close, change, chist[tt + BT_HALO - st.bb_w],
+                csq[tt + BT_HALO], csq[tt + BT_HALO - st.bb_w],
+                sh_vote[tt]);
             const int bit = t & 63;
-            ew |= (unsigned long long)(net >= st.entry_v) << bit;
-            xw |= (unsigned long long)(net <= -st.exit_v) << bit;
+            ew |= (unsigned long long)b.x << bit;
+            xw |= (unsigned long long)b.y << bit;
             if (bit == 63 || t == T - 1) {
                 if (act) {
                     const long w = t >> 6;
