@@ -360,7 +360,7 @@ __global__ void __launch_bounds__(BT_BLOCK) backtest_kernel(
             const float4 sv = sh_vote[0];
 #pragma unroll
             for (int i = 0; i < ILP; ++i)
-                st[i].step<true>(t0, close, hl[BT_HALO][0],
+                st[i].template step<true>(t0, close, hl[BT_HALO][0],
                                  hl[BT_HALO][1],
                                  chist[BT_HALO - st[i].q.bb_w],
                                  change, sv);
@@ -376,7 +376,7 @@ __global__ void __launch_bounds__(BT_BLOCK) backtest_kernel(
             const float4 sv = sh_vote[tt];   // b128 broadcast
 #pragma unroll
             for (int i = 0; i < ILP; ++i)
-                st[i].step<false>(t, close, high, low,
+                st[i].template step<false>(t, close, high, low,
                                   chist[tt + BT_HALO - st[i].q.bb_w],
                                   change, sv);
             prev_close = close;
