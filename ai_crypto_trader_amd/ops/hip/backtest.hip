@@ -249,7 +249,11 @@ struct BtState {
 };
 
 template <int ILP>
-__global__ void __launch_bounds__(BT_BLOCK) backtest_kernel(
+// second arg: min resident blocks/CU — pins the allocation at <=102
+// VGPR so 5 waves/SIMD stay resident (the resnap addition had silently
+// inflated the allocation to 160 VGPR = 3 waves/SIMD, a 20% headline
+// regression; measured: occupancy beats spill-free codegen here)
+__global__ void __launch_bounds__(BT_BLOCK, ILP == 1 ? 5 : 3) backtest_kernel(
     const float* __restrict__ candles,   // (nsym, T, 4)
     const float* __restrict__ pop,       // (P, NPARAM)
     float* __restrict__ metrics,         // (P, nsym, NMETRIC)
