@@ -98,7 +98,8 @@ struct BtState {
     // Drift-free Bollinger resnap (strategy.py RESNAP): recompute the
     // window sums directly, oldest->newest, from the close history.
     // `hist` points at close[t] (the resnap candle itself, j = 0 term).
-    __device__ void resnap(const float* __restrict__ hist)
+    __device__ __attribute__((noinline)) void resnap(
+        const float* __restrict__ hist)
     {
 #pragma clang fp contract(off)
         double s = 0.0, s2 = 0.0;
