@@ -98,8 +98,9 @@ struct BtState {
     // Drift-free Bollinger resnap (strategy.py RESNAP): recompute the
     // window sums directly, oldest->newest, from the close history.
     // `hist` points at close[t] (the resnap candle itself, j = 0 term).
-    __device__ __attribute__((noinline)) void resnap(
-        const float* __restrict__ hist)
+    // (noinline was tried here: 86 VGPR/no spills but the call
+    // inside the tile loop cost 25% — inlined + launch-bounds wins)
+    __device__ void resnap(const float* __restrict__ hist)
     {
 #pragma clang fp contract(off)
         double s = 0.0, s2 = 0.0;
