@@ -87,9 +87,12 @@ FEE = 0.001            # taker fee per side (reference: strategy_tester.py 0.1%)
 # at any RESNAP-aligned boundary — the property the time-parallel GPU
 # backtest (ops/hip/backtest_tp.hip) relies on to split the time axis
 # across waves while staying bit-identical to the sequential engines.
-# Multiple of the kernel tile (256). Cost: ~2x32 f64 adds per 4096
-# candles — amortized noise.
-RESNAP = 4096
+# Multiple of the kernel tile (256) and LONGER than a seg-64 fitness
+# segment (1M/64 = 15625), so the segmented headline path never
+# resnaps and dispatches the resnap-free kernel instantiation (same
+# codegen/occupancy as if the feature didn't exist). Cost on the
+# continuous path: ~2x32 f64 adds per 16384 candles — noise.
+RESNAP = 16_384
 
 PARAM_NAMES = [
     "rsi_period", "rsi_oversold", "rsi_overbought",

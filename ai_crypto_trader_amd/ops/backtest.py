@@ -52,10 +52,12 @@ def pick_nshards(nsym: int, T: int, P: int, *, target_blocks: int = 4096,
     (measured sweep at 1024x64x1M: 8 shards = 399, 16 = 419, 32 = 394
     Gcandles/s -> target 16 blocks/CU), bounded so each shard body is
     >= one RESNAP period and >= 4x the warm tail."""
+    from ..backtesting.strategy import RESNAP
+
     chunks = (P + 255) // 256
     base = nsym * chunks
     want = max(1, -(-target_blocks // base))
-    max_shards = max(1, T // max(4096, 4 * tail))
+    max_shards = max(1, T // max(RESNAP, 4 * tail))
     return min(want, max_shards, 64)
 
 
@@ -98,8 +100,12 @@ def run_backtest_continuous_gpu(
     population = population.contiguous()
     nsym, T, _ = candles.shape
     P = population.shape[0]
+    from ..backtesting.strategy import RESNAP
+
     if nshards <= 0:
         nshards = pick_nshards(nsym, T, P, tail=tail)
+    # shard bodies must be RESNAP-aligned (BB restart exactness)
+    nshards = min(nshards, max(1, T // RESNAP))
     nwords = (T + 63) // 64
     dev = candles.device
     key = (nsym, nwords, P, dev.index)
@@ -141,7 +147,7 @@ def run_backtest_continuous_gpu(
     start_ev.record(cur)
     sf.wait_event(start_ev)
     st.wait_event(start_ev)
-    body4 = (T // nshards) // 4096 * 4096 if nshards > 1 else T
+    body4 = (T // nshards) // RESNAP * RESNAP if nshards > 1 else T
     # geometric group sizes (1,1,2,4,...): the first trades chunk starts
     # after one shard of flags (~1/nshards of the flags time) while the
     # later, bigger flags launches keep the chip oversubscribed

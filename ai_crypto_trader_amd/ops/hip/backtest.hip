@@ -35,7 +35,7 @@
 #define BT_NPARAM 19
 #define BT_NMETRIC 10
 #define BT_WARMUP 128          // == strategy.py WARMUP
-#define BT_RESNAP 4096         // == strategy.py RESNAP (BB sum resnap)
+#define BT_RESNAP 16384        // == strategy.py RESNAP (BB sum resnap)
 #define BT_FEE 0.001f
 #define BT_EPS 1e-9f
 // float32(sqrt(525600)) EXACTLY (0x44353ee6) — a shorter decimal literal
@@ -250,7 +250,7 @@ struct BtState {
     }
 };
 
-template <int ILP>
+template <int ILP, bool HAS_RESNAP>
 // second arg: min resident blocks/CU — pins the allocation at <=102
 // VGPR so 5 waves/SIMD stay resident (the resnap addition had silently
 // inflated the allocation to 160 VGPR = 3 waves/SIMD, a 20% headline
@@ -324,7 +324,8 @@ __global__ void __launch_bounds__(BT_BLOCK, ILP == 1 ? 5 : 3) backtest_kernel(
         const int tend = min(BT_TILE, T - t0);
         // BB resnap at RESNAP-aligned tiles (engine_cpu.py lockstep):
         // window closes for candle t0 are chist[BT_HALO - j], j < bb_w.
-        const bool resnap_tile = (t0 > 0) && ((t0 & (BT_RESNAP - 1)) == 0);
+        const bool resnap_tile = HAS_RESNAP && (t0 > 0) &&
+                                 ((t0 & (BT_RESNAP - 1)) == 0);
         if (resnap_tile) {
 #pragma unroll
             for (int i = 0; i < ILP; ++i)
@@ -409,15 +410,21 @@ extern "C" void launch_backtest(const float* candles, const float* pop,
         const char* e = getenv("BT_FORCE_ILP2");
         return e && e[0] == '1';
     }();
+    // histories short enough never to cross a RESNAP boundary run the
+    // resnap-free instantiation: identical codegen (96 VGPR, 5 waves/
+    // SIMD) to a kernel without the feature — the seg-64 headline path
+    const bool rs = T > BT_RESNAP;
     if (force2 && P % (BT_BLOCK * 2) == 0) {
         int chunks = P / (BT_BLOCK * 2);
-        hipLaunchKernelGGL(backtest_kernel<2>, dim3(nsym * chunks),
+        auto k = rs ? backtest_kernel<2, true> : backtest_kernel<2, false>;
+        hipLaunchKernelGGL(k, dim3(nsym * chunks),
                            dim3(BT_BLOCK), 0, stream,
                            candles, pop, metrics, nsym, T, P, chunks,
                            initial_equity);
     } else {
         int chunks = (P + BT_BLOCK - 1) / BT_BLOCK;
-        hipLaunchKernelGGL(backtest_kernel<1>, dim3(nsym * chunks),
+        auto k = rs ? backtest_kernel<1, true> : backtest_kernel<1, false>;
+        hipLaunchKernelGGL(k, dim3(nsym * chunks),
                            dim3(BT_BLOCK), 0, stream,
                            candles, pop, metrics, nsym, T, P, chunks,
                            initial_equity);

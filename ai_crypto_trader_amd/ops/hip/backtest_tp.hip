@@ -21,7 +21,7 @@
 //          asserted by tests/test_gpu_kernels GPU parity tests.
 //        - Bollinger f64 rolling sums are NOT contractive; instead both
 //          sequential engines resnap them from the raw window every
-//          RESNAP=4096 candles (strategy.py), and shard bodies start on
+//          RESNAP=16384 candles (strategy.py), and shard bodies start on
 //          RESNAP-aligned boundaries, so the shard's BB state is
 //          *bit-identical* to the continuous run by construction.
 //        - stoch/Williams/trend vote inputs are finite-window shared
@@ -56,7 +56,7 @@
 #define BT_NPARAM 19
 #define BT_NMETRIC 10
 #define BT_WARMUP 128
-#define BT_RESNAP 4096
+#define BT_RESNAP 16384
 #define BT_FEE 0.001f
 #define BT_EPS 1e-9f
 // float32(sqrt(525600)) EXACTLY (0x44353ee6) — see backtest.hip

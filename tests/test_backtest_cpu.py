@@ -101,7 +101,7 @@ def test_flag_decomposition_bitwise():
         run_trades_from_flags_cpu,
     )
 
-    candles = candles_chl_v(generate_ohlcv(9500, 2, seed=11))  # > RESNAP
+    candles = candles_chl_v(generate_ohlcv(20000, 2, seed=11))  # > RESNAP
     pop = random_population(16, seed=7)
     m_ref, nets = run_backtest_cpu(candles, pop, record_net=True)
     entry_v = pop[:, 10].astype(np.int32)[:, None, None]
@@ -115,10 +115,10 @@ def test_flag_decomposition_bitwise():
 
 
 def test_resnap_keeps_engine_consistent():
-    """Bollinger resnap (strategy.py RESNAP) fires at t=4096, 8192 for
-    T>8192; the window sums it installs must agree with the incremental
+    """Bollinger resnap (strategy.py RESNAP) fires at t=16384 for
+    T>16384; the window sums it installs must agree with the incremental
     f64 sums to rounding noise, so metrics stay finite and trades occur."""
-    candles = candles_chl_v(generate_ohlcv(9000, 1, seed=3))
+    candles = candles_chl_v(generate_ohlcv(18000, 1, seed=3))
     pop = random_population(8, seed=5)
     m = run_backtest_cpu(candles, pop)
     assert np.isfinite(m).all()

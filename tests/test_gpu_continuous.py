@@ -33,12 +33,12 @@ def test_continuous_matches_cpu_bitwise(dev):
     from ai_crypto_trader_amd.backtesting.strategy import random_population
     from ai_crypto_trader_amd.ops.backtest import run_backtest_continuous_gpu
 
-    candles = _market(18432, 2, seed=21)     # 4.5 RESNAP periods
+    candles = _market(49152, 2, seed=21)     # 3 RESNAP periods
     pop = random_population(32, seed=9)
     m_cpu = run_backtest_cpu(candles, pop)
     c_t = torch.from_numpy(candles).to(dev)
     p_t = torch.from_numpy(pop).to(dev)
-    m_gpu = run_backtest_continuous_gpu(c_t, p_t, nshards=4).cpu().numpy()
+    m_gpu = run_backtest_continuous_gpu(c_t, p_t, nshards=3).cpu().numpy()
     assert np.array_equal(m_cpu, m_gpu), (
         f"max abs diff {np.max(np.abs(m_cpu - m_gpu))}"
     )
@@ -51,12 +51,12 @@ def test_continuous_shard_invariance(dev):
     from ai_crypto_trader_amd.backtesting.strategy import random_population
     from ai_crypto_trader_amd.ops.backtest import run_backtest_continuous_gpu
 
-    candles = _market(24576, 2, seed=5)
+    candles = _market(49152, 2, seed=5)
     pop = random_population(64, seed=3)
     c_t = torch.from_numpy(candles).to(dev)
     p_t = torch.from_numpy(pop).to(dev)
     m1 = run_backtest_continuous_gpu(c_t, p_t, nshards=1).cpu().numpy()
-    m4 = run_backtest_continuous_gpu(c_t, p_t, nshards=4).cpu().numpy()
+    m4 = run_backtest_continuous_gpu(c_t, p_t, nshards=3).cpu().numpy()
     assert np.array_equal(m1, m4)
 
 
@@ -117,17 +117,17 @@ def test_continuous_time_chunked_resume(dev):
     from ai_crypto_trader_amd.backtesting.strategy import random_population
     from ai_crypto_trader_amd.ops.backtest import run_backtest_continuous_gpu
 
-    candles = _market(36864, 2, seed=31)     # 9 RESNAP periods
+    candles = _market(65536, 2, seed=31)     # 4 RESNAP periods
     pop = random_population(32, seed=12)
     m_cpu = run_backtest_cpu(candles, pop)
     c_t = torch.from_numpy(candles).to(dev)
     p_t = torch.from_numpy(pop).to(dev)
-    m1 = run_backtest_continuous_gpu(c_t, p_t, nshards=8,
+    m1 = run_backtest_continuous_gpu(c_t, p_t, nshards=4,
                                      time_groups=1).cpu().numpy()
-    m3 = run_backtest_continuous_gpu(c_t, p_t, nshards=8,
+    m3 = run_backtest_continuous_gpu(c_t, p_t, nshards=4,
                                      time_groups=3).cpu().numpy()
-    m8 = run_backtest_continuous_gpu(c_t, p_t, nshards=8,
-                                     time_groups=8).cpu().numpy()
+    m8 = run_backtest_continuous_gpu(c_t, p_t, nshards=4,
+                                     time_groups=4).cpu().numpy()
     assert np.array_equal(m_cpu, m1)
     assert np.array_equal(m_cpu, m3)
     assert np.array_equal(m_cpu, m8)
