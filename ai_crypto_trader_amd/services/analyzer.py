@@ -82,9 +82,20 @@ class LocalAnalyst:
         agreement = sum(
             1 for k in self.weights
             if f[k] * score > 0 and abs(f[k]) > 0.05)
-        confidence = min(0.99, abs(score) * 1.6 + 0.08 * agreement)
         decision = "BUY" if score > 0.12 else (
             "SELL" if score < -0.12 else "HOLD")
+        # confidence on the REFERENCE analyst's scale: GPT confidences
+        # cluster in [0.5, 0.9] for actionable calls (which is what the
+        # reference's 0.7 default gate was tuned against,
+        # config.json min_confidence + ai_trader.py:368-387) — an
+        # actionable decision starts at ~0.5 and climbs with score
+        # magnitude and factor agreement; HOLD stays low-confidence
+        if decision == "HOLD":
+            confidence = min(0.49, abs(score) * 1.6
+                             + 0.08 * agreement)
+        else:
+            confidence = min(0.99, 0.45 + abs(score) * 0.9
+                             + 0.04 * agreement)
         risk = "low" if abs(score) > 0.5 else (
             "high" if abs(f["oscillators"]) > 0.8 else "medium")
         key = sorted(self.weights, key=lambda k: -abs(f[k] * self.weights[k]))

@@ -50,6 +50,9 @@ def parse_args():
     ap.add_argument("--speed", type=float, default=0.0,
                     help="0 = replay as fast as possible")
     ap.add_argument("--seed", type=int, default=0)
+    ap.add_argument("--market", choices=["calm", "regime"], default="calm",
+                    help="synthetic market: calm GBM or regime-switching "
+                         "episodes (bull/calm/bear/volatile)")
     ap.add_argument("--gpu", action="store_true",
                     help="use cuda:0 for the numeric services")
     ap.add_argument("--status-interval", type=float, default=5.0)
@@ -146,8 +149,15 @@ async def main():
 
     device = "cuda:0" if args.gpu else "cpu"
     bus = InProcessBus()
-    market = candles_chl_v(
-        generate_ohlcv(args.candles, len(symbols), seed=args.seed))
+    if args.market == "regime":
+        from ai_crypto_trader_amd.data.synthetic import (
+            generate_regime_ohlcv,
+        )
+        market = candles_chl_v(generate_regime_ohlcv(
+            args.candles, len(symbols), seed=args.seed))
+    else:
+        market = candles_chl_v(
+            generate_ohlcv(args.candles, len(symbols), seed=args.seed))
     feed = SyntheticFeed(market, symbols, start=0, speed=args.speed)
     exchange = ExchangeFactory.create_exchange(
         "fake", fee_rate=cfg.trading.fee_rate,

@@ -93,7 +93,7 @@ def bench_lstm_train(nsym=32, T=1_000_000, seq=60, batch=16_384,
         "ms_per_step": dt * 1e3,
         "windows_per_sec": batch / dt,
         "cells_per_sec": batch * seq / dt,
-        "final_loss": float(loss),
+        "final_loss": float(loss.detach()),
         "config": {"nsym": nsym, "T": T, "seq": seq, "batch": batch,
                    "dtype": "bf16 (recurrent cells) + f32 master",
                    "resident_bytes": int(feats.numel() * 4 +
