@@ -298,4 +298,4 @@ def test_order_book_analyzer_on_live_depth():
     assert out["mid"] == pytest.approx(50_000.0, rel=1e-3)
     assert out["spread_bps"] > 0
     assert len(out["price_impact"]) == 5
-    assert out["signal"]["signal"] in ("buy", "sell", "neutral")
+    assert out["signal"]["direction"] in ("bullish", "bearish", "neutral")
