@@ -156,6 +156,9 @@ class NewsConfig(BaseModel):
 class PatternConfig(BaseModel):
     min_confidence: float = 0.5               # pattern_detection_threshold
     interval_s: float = 5.0
+    # classifier architecture (reference model types,
+    # pattern_recognition.py:94-196): cnn | lstm | cnn_lstm
+    model_type: str = "cnn"
 
 
 class VolumeProfileConfig(BaseModel):

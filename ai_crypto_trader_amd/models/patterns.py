@@ -101,11 +101,54 @@ class PatternCNN(nn.Module):
         return self.net(x[:, None, :])
 
 
+class PatternLSTM(nn.Module):
+    """LSTM(64)->LSTM(32)->dense classifier — the reference's LSTM
+    pattern-model type (pattern_recognition.py:128-158)."""
+
+    def __init__(self, n_classes: int = len(PATTERNS), win: int = WIN):
+        super().__init__()
+        self.l1 = nn.LSTM(1, 64, batch_first=True)
+        self.l2 = nn.LSTM(64, 32, batch_first=True)
+        self.head = nn.Sequential(nn.Linear(32, 64), nn.ReLU(),
+                                  nn.Linear(64, n_classes))
+
+    def forward(self, x):        # (B, WIN)
+        h, _ = self.l1(x[:, :, None])
+        h, _ = self.l2(h)
+        return self.head(h[:, -1])
+
+
+class PatternCNNLSTM(nn.Module):
+    """Conv64/Conv32 front-end -> LSTM(32) -> dense — the reference's
+    hybrid type (pattern_recognition.py:160-196)."""
+
+    def __init__(self, n_classes: int = len(PATTERNS), win: int = WIN):
+        super().__init__()
+        self.conv = nn.Sequential(
+            nn.Conv1d(1, 64, 5, padding=2), nn.ReLU(), nn.MaxPool1d(2),
+            nn.Conv1d(64, 32, 3, padding=1), nn.ReLU(), nn.MaxPool1d(2),
+        )
+        self.lstm = nn.LSTM(32, 32, batch_first=True)
+        self.head = nn.Linear(32, n_classes)
+
+    def forward(self, x):        # (B, WIN)
+        h = self.conv(x[:, None, :])            # (B, 32, WIN/4)
+        h, _ = self.lstm(h.transpose(1, 2))     # (B, WIN/4, 32)
+        return self.head(h[:, -1])
+
+
+PATTERN_MODELS = {
+    "cnn": PatternCNN, "lstm": PatternLSTM, "cnn_lstm": PatternCNNLSTM,
+}
+
+
 class PatternRecognitionModel:
-    def __init__(self, device="cpu", seed: int = 0):
+    def __init__(self, device="cpu", seed: int = 0,
+                 model_type: str = "cnn"):
         self.device = torch.device(device)
         torch.manual_seed(seed)
-        self.model = PatternCNN().to(self.device)
+        self.model_type = model_type
+        self.model = PATTERN_MODELS[model_type]().to(self.device)
         self.trained = False
 
     def make_dataset(self, n_per_class: int = 64, seed: int = 0):

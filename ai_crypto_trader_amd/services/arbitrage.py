@@ -105,3 +105,46 @@ class ArbitrageDetectionService(Service):
                     await self.bus.publish(
                         Channels.ARBITRAGE_NOTIFICATIONS, op)
             await self.sleep(2.0)
+
+
+class CrossExchangeDetector:
+    """Cross-exchange spread detection (reference
+    arbitrage_detection_service.py:434-522): compares the same symbol's
+    quotes across two ExchangeInterface venues and reports spreads that
+    survive both venues' taker fees."""
+
+    def __init__(self, venues: dict, min_profit_pct: float = 0.05):
+        self.venues = dict(venues)            # name -> ExchangeInterface
+        self.min_profit_pct = min_profit_pct
+
+    def scan(self, symbols: list[str]) -> list[dict]:
+        out = []
+        names = list(self.venues)
+        for sym in symbols:
+            quotes = {}
+            for name in names:
+                try:
+                    tk = self.venues[name].get_ticker(sym)
+                    fee = self.venues[name].get_trading_fees(sym) \
+                        .get("taker", 0.001)
+                except Exception:
+                    continue
+                if tk.get("ask", 0) > 0 and tk.get("bid", 0) > 0:
+                    quotes[name] = (tk["bid"], tk["ask"], fee)
+            for buy_v in quotes:
+                for sell_v in quotes:
+                    if buy_v == sell_v:
+                        continue
+                    bid_s, _, fee_s = quotes[sell_v]
+                    _, ask_b, fee_b = quotes[buy_v]
+                    # buy at buy_v's ask, sell at sell_v's bid, both fees
+                    net = bid_s * (1 - fee_s) / (ask_b * (1 + fee_b)) - 1
+                    if net * 100 >= self.min_profit_pct:
+                        out.append({
+                            "symbol": sym, "buy_on": buy_v,
+                            "sell_on": sell_v,
+                            "buy_ask": ask_b, "sell_bid": bid_s,
+                            "net_profit_pct": round(net * 100, 4),
+                            "at": time.time(),
+                        })
+        return sorted(out, key=lambda o: -o["net_profit_pct"])

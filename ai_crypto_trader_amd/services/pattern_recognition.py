@@ -18,7 +18,9 @@ class PatternRecognitionService(Service):
     def __init__(self, bus, config=None, device="cpu",
                  min_confidence: float | None = None):
         super().__init__(bus, config)
-        self.model = PatternRecognitionModel(device, seed=self.config.seed)
+        self.model = PatternRecognitionModel(
+            device, seed=self.config.seed,
+            model_type=getattr(self.config.patterns, 'model_type', 'cnn'))
         self.min_confidence = (min_confidence if min_confidence is not None
                                else self.config.patterns.min_confidence)
         self.closes: dict[str, list[float]] = {}

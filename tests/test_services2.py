@@ -842,3 +842,57 @@ def test_regime_transition_drives_strategy_switch_with_hysteresis():
     assert sel.should_switch(best, scores)
     # staying on the same strategy never switches
     assert not sel.should_switch(cur, scores)
+
+
+def test_pattern_model_types_cnn_lstm_hybrid():
+    """All three reference pattern-classifier architectures (cnn, lstm,
+    cnn_lstm — pattern_recognition.py:94-196) train and detect behind
+    patterns.model_type."""
+    import numpy as np
+
+    from ai_crypto_trader_amd.models.patterns import (
+        PATTERN_MODELS, PatternRecognitionModel, generate_pattern,
+    )
+
+    assert set(PATTERN_MODELS) == {"cnn", "lstm", "cnn_lstm"}
+    rng = np.random.default_rng(0)
+    probe = generate_pattern("double_top", rng)
+    # recurrent types are slow starters from scratch: give them a
+    # higher lr + more epochs; the bar is "clearly learning" (random
+    # chance over 15 classes is 0.067)
+    plans = {"cnn": (6, 2e-3, 0.6), "lstm": (25, 5e-3, 0.25),
+             "cnn_lstm": (25, 5e-3, 0.6)}
+    for mt in PATTERN_MODELS:
+        epochs, lr, bar = plans[mt]
+        m = PatternRecognitionModel(seed=0, model_type=mt)
+        acc = m.train(epochs=epochs, n_per_class=24, lr=lr)
+        assert acc > bar, (mt, acc)
+        det = m.detect(probe * 100.0)
+        assert det["pattern"] in __import__(
+            "ai_crypto_trader_amd.models.patterns",
+            fromlist=["PATTERNS"]).PATTERNS
+        assert 0.0 <= det["confidence"] <= 1.0
+
+
+def test_cross_exchange_arbitrage_detection():
+    """Cross-exchange spread scan (reference
+    arbitrage_detection_service.py:434-522): a fee-surviving spread
+    between two venues is detected; a sub-fee spread is not."""
+    from ai_crypto_trader_amd.services.arbitrage import (
+        CrossExchangeDetector,
+    )
+    from ai_crypto_trader_amd.utils.exchange import FakeExchange
+
+    a = FakeExchange()
+    b = FakeExchange()
+    a.set_price("BTCUSDC", 100.0)
+    b.set_price("BTCUSDC", 101.0)      # 1% spread >> 2x0.1% fees
+    det = CrossExchangeDetector({"alpha": a, "beta": b},
+                                min_profit_pct=0.05)
+    opps = det.scan(["BTCUSDC"])
+    assert opps and opps[0]["buy_on"] == "alpha" \
+        and opps[0]["sell_on"] == "beta"
+    assert opps[0]["net_profit_pct"] > 0.5
+
+    b.set_price("BTCUSDC", 100.05)     # 0.05% spread < fees
+    assert det.scan(["BTCUSDC"]) == []
