@@ -299,3 +299,24 @@ def test_order_book_analyzer_on_live_depth():
     assert out["spread_bps"] > 0
     assert len(out["price_impact"]) == 5
     assert out["signal"]["direction"] in ("bullish", "bearish", "neutral")
+
+
+def test_binance_fetch_to_backtest_engine_end_to_end(tmp_path):
+    """Venue klines (via the live paginated client against the
+    wire-format server) flow through the data manager's CSV store into
+    BacktestEngine.run_backtest — the reference's fetch-then-backtest
+    pipeline (backtest_engine.py:64-125 + data_manager.py:47-114) with
+    `source='binance'` as the only switch."""
+    from ai_crypto_trader_amd.backtesting.engine import BacktestEngine
+
+    server = BinanceFixtureServer(prices={"BTCUSDC": 100.0})
+    server.kline_end_ms = 60_000 * 3000
+    eng = BacktestEngine(data_dir=str(tmp_path), device="cpu")
+    eng.dm.fetch_market_data("BTCUSDC", "1m", n_candles=3000,
+                             source="binance", transport=server,
+                             start_ms=0)
+    res = eng.run_backtest("BTCUSDC", strategy="momentum",
+                           n_candles=3000)
+    assert res["n_candles"] == 3000 if "n_candles" in res else True
+    assert "final_equity" in res and res["final_equity"] > 0
+    assert "sharpe" in res
