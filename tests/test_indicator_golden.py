@@ -76,3 +76,32 @@ def test_fixture_provenance():
         stored = pd.read_csv(FIX)
         assert np.allclose(ind["rsi"].to_numpy()[WARM:],
                            stored["rsi"].to_numpy()[WARM:], rtol=1e-8)
+
+
+@pytest.mark.gpu
+def test_gpu_indicators_match_ta_fixture():
+    """Closes the chain fixture -> CPU twin -> HIP kernel: the GPU
+    indicators output is compared DIRECTLY against the ta-library
+    golden fixture (not just against the twin)."""
+    torch = pytest.importorskip("torch")
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    from ai_crypto_trader_amd.ops.indicators import indicators_gpu
+
+    df = pd.read_csv(FIX)
+    candles = df[["close", "high", "low", "volume"]].to_numpy(np.float32)
+    g = indicators_gpu(
+        torch.from_numpy(candles[None]).cuda()).cpu().numpy()[0]
+    twin_cols = {n: g[:, i] for i, n in enumerate(IND_NAMES)}
+    close_scale = float(np.nanmedian(df["close"].to_numpy()))
+    for twin_col, fix_col in COLS.items():
+        want = df[fix_col].to_numpy(np.float64)[WARM:]
+        got = twin_cols[twin_col].astype(np.float64)[WARM:]
+        if twin_col in ("macd", "macd_signal"):
+            scale = close_scale * 1e-2
+        else:
+            scale = np.maximum(np.abs(want),
+                               np.nanmedian(np.abs(want)) + 1e-9)
+        err = np.nanmax(np.abs(got - want) / scale)
+        # kernel uses v_rcp/raw v_sqrt (documented): slightly looser
+        assert err < 2e-3, f"{twin_col}: max rel err {err:.2e}"
