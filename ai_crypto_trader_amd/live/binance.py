@@ -230,7 +230,9 @@ def fetch_klines(symbol: str, interval: str = "1m",
     step = _INTERVAL_MS.get(interval, 60_000)
     rows: list[list] = []
     cursor = start_ms
-    while True:
+    max_pages = 5000                    # safety: ~5M rows / request run
+    while max_pages > 0:
+        max_pages -= 1
         params = {"symbol": symbol, "interval": interval, "limit": 1000}
         if cursor is not None:
             params["startTime"] = int(cursor)

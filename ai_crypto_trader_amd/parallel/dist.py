@@ -128,6 +128,11 @@ class GradBucketer:
         loss.backward()          # hooks fire during backward
         bucketer.finalize()      # wait + scatter averaged grads
         opt.step()
+
+    One finalize() per backward: gradient ACCUMULATION across several
+    backwards before a single finalize is not supported (buckets launch
+    mid-accumulation) — call finalize per micro-batch or use the flat
+    synchronous collective for that pattern.
     """
 
     def __init__(self, params, bucket_bytes: int = 1 << 20):
