@@ -267,6 +267,34 @@ class TradeExecutorService(Service):
         self.metrics.portfolio_value.set(total)
         self.metrics.active_trades.set(len(self.active))
 
+    async def cleanup_positions(self) -> int:
+        """Startup liquidation (reference trade_executor_service.py:
+        488-547): sell any non-quote balances to the quote asset before
+        trading begins, so the executor starts from a known-flat book.
+        Returns the number of positions liquidated."""
+        balances = self.exchange.get_balances()
+        quote = self.config.trading.quote_asset
+        n = 0
+        for asset, qty in balances.items():
+            if asset == quote or qty <= 0:
+                continue
+            sym = asset + quote
+            price = self.exchange.get_ticker(sym)["price"]
+            if price <= 0 or qty * price < self.config.trading.min_trade_usd:
+                continue           # dust stays (reference skips dust too)
+            filters = self.exchange.get_symbol_filters(sym)
+            sell_qty, _, ok = round_to_filters(qty, price, filters)
+            if not ok or sell_qty <= 0:
+                continue
+            o = self.exchange.create_order(sym, "SELL", "MARKET", sell_qty)
+            if o.status == "FILLED":
+                n += 1
+                self.log.info("startup liquidation: sold %s %s @ %s",
+                              sell_qty, asset, o.filled_price)
+        if n:
+            await self._write_state()
+        return n
+
     # --- loops -----------------------------------------------------------
     async def _consume_signals(self):
         sub = self.bus.subscribe(Channels.TRADING_SIGNALS,

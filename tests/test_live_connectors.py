@@ -320,3 +320,26 @@ def test_binance_fetch_to_backtest_engine_end_to_end(tmp_path):
     assert res["n_candles"] == 3000 if "n_candles" in res else True
     assert "final_equity" in res and res["final_equity"] > 0
     assert "sharpe" in res
+
+
+def test_executor_startup_liquidation():
+    """cleanup_positions (reference trade_executor_service.py:488-547):
+    pre-existing base balances are liquidated to quote at startup,
+    dust is left alone."""
+    from ai_crypto_trader_amd.bus.message_bus import InProcessBus
+    from ai_crypto_trader_amd.services.trade_executor import (
+        TradeExecutorService,
+    )
+
+    ex, server = make_exchange(
+        server=BinanceFixtureServer(
+            prices={"BTCUSDC": 50_000.0, "ETHUSDC": 3_000.0},
+            balances={"USDC": 100.0, "BTC": 0.5,
+                      "ETH": 0.001}))   # ETH position = $3 -> dust
+    svc = TradeExecutorService(InProcessBus(), ex, AppConfig())
+    n = asyncio.run(svc.cleanup_positions())
+    assert n == 1
+    bal = ex.get_balances()
+    assert bal.get("BTC", 0.0) < 1e-5          # liquidated (step residue)
+    assert bal["ETH"] == pytest.approx(0.001)  # dust untouched
+    assert bal["USDC"] > 100.0
