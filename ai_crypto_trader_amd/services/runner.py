@@ -140,7 +140,30 @@ def parse_args():
     ap.add_argument("--speed", type=float, default=1.0)
     ap.add_argument("--seed", type=int, default=0)
     ap.add_argument("--device", default="cpu")
+    ap.add_argument("--health-port", type=int, default=0,
+                    help="serve a TCP health endpoint on this port "
+                         "(reference: per-service health_check_server, "
+                         "ports 8001-8016, probed with `nc -z` in "
+                         "docker-compose healthchecks); 0 = off")
     return ap.parse_args()
+
+
+async def serve_health(svc, port: int):
+    """One-line JSON health over raw TCP — `nc -z` (connect probe) and
+    `curl`-style reads both work (reference market_monitor_service.py:
+    635-656 health_check_server)."""
+    import json as _json
+
+    async def handle(reader, writer):
+        h = svc.health()
+        writer.write((_json.dumps(h) + "\n").encode())
+        try:
+            await writer.drain()
+        finally:
+            writer.close()
+
+    server = await asyncio.start_server(handle, "0.0.0.0", port)
+    return server
 
 
 async def amain():
@@ -155,6 +178,9 @@ async def amain():
         bus = make_bus()
     svc = build_service(args.service, bus, cfg, args)
     await svc.start()
+    health_server = None
+    if args.health_port:
+        health_server = await serve_health(svc, args.health_port)
     try:
         if args.minutes > 0:
             await asyncio.sleep(args.minutes * 60)
@@ -162,6 +188,8 @@ async def amain():
             while True:
                 await asyncio.sleep(3600)
     finally:
+        if health_server is not None:
+            health_server.close()
         await svc.stop()
         await bus.close()
 

@@ -896,3 +896,33 @@ def test_cross_exchange_arbitrage_detection():
 
     b.set_price("BTCUSDC", 100.05)     # 0.05% spread < fees
     assert det.scan(["BTCUSDC"]) == []
+
+
+def test_runner_tcp_health_endpoint():
+    """The per-service TCP health endpoint (reference
+    health_check_server + nc -z compose probes) responds with the
+    service's health JSON."""
+    import asyncio
+    import json as _json
+
+    from ai_crypto_trader_amd.bus.message_bus import InProcessBus
+    from ai_crypto_trader_amd.config import AppConfig
+    from ai_crypto_trader_amd.services.registry import ModelRegistryService
+    from ai_crypto_trader_amd.services.runner import serve_health
+
+    async def run():
+        svc = ModelRegistryService(InProcessBus(), AppConfig())
+        await svc.start()
+        server = await serve_health(svc, 18943)
+        try:
+            reader, writer = await asyncio.open_connection(
+                "127.0.0.1", 18943)
+            line = await asyncio.wait_for(reader.readline(), 5)
+            writer.close()
+            return _json.loads(line)
+        finally:
+            server.close()
+            await svc.stop()
+
+    h = asyncio.run(run())
+    assert h["service"] == "model_registry" and h["healthy"]
