@@ -68,6 +68,17 @@ def main():
     dt = timeit(lambda: run_backtest_continuous_gpu(
         c_t, p_t, nshards=nshards, tail=args.tail))
     out["continuous"] = {"s": dt, "gcandles_per_s": evals / dt / 1e9}
+    # activity-sorted lanes: permute the population by measured trade
+    # count (what GAEngine does from generation 2 on) and re-measure
+    m0 = run_backtest_continuous_gpu(c_t, p_t, nshards=nshards,
+                                     tail=args.tail)
+    order = torch.argsort(m0[..., 1].sum(dim=1))
+    p_sorted = p_t[order].contiguous()
+    dts = timeit(lambda: run_backtest_continuous_gpu(
+        c_t, p_sorted, nshards=nshards, tail=args.tail))
+    out["continuous_activity_sorted"] = {
+        "s": dts, "gcandles_per_s": evals / dts / 1e9}
+
     # time-group sweep (flags/trades overlap granularity)
     gsweep = {}
     for g in (1, 2, 4, 8, 16):
