@@ -99,7 +99,6 @@ class GAEngine:
             self.candles_np = np.ascontiguousarray(candles)
         self.last_metrics = None          # local shard (P, nsym, NMETRIC)
         self.last_fitness_global = None   # (world*P,) torch tensor
-        self._activity = None             # per-individual n_trades (sort key)
 
     def _my_slice(self):
         p = self.pop_per_rank
@@ -112,25 +111,14 @@ class GAEngine:
             if self.continuous:
                 from ..ops.backtest import run_backtest_continuous_gpu
 
-                # activity-sorted lanes: order the shard by last
-                # generation's per-individual trade count so waves hold
-                # similarly-active lanes — the trades kernel's exact
-                # whole-word skip then fires for quiet waves (unsorted
-                # populations at ~50% mean duty leave ~100% of waves
-                # busy: one active lane pins all 64). Pure permutation:
-                # fitness is unsorted back, evolution order-independent.
-                if self._activity is not None and \
-                        self._activity.shape[0] == shard.shape[0]:
-                    order = torch.argsort(self._activity)
-                    inv = torch.empty_like(order)
-                    inv[order] = torch.arange(order.shape[0],
-                                              device=order.device)
-                    metrics = run_backtest_continuous_gpu(
-                        self.candles_t, shard[order].contiguous())[inv]
-                else:
-                    metrics = run_backtest_continuous_gpu(
-                        self.candles_t, shard)
-                self._activity = metrics[..., 1].sum(dim=1)
+                # (activity-sorted lanes were tried here — permuting
+                # the shard by trade count to make the trades kernel's
+                # wave-word skip fire — and measured a wash, 204 vs 208
+                # G/s: total trade count doesn't align the TIMING of
+                # activity, so whole-wave-flat 64-candle words barely
+                # increase. See profiles/backtest_continuous_pmc.json.)
+                metrics = run_backtest_continuous_gpu(
+                    self.candles_t, shard)
             else:
                 from ..ops.backtest import run_backtest_gpu
                 metrics = run_backtest_gpu(self.candles_t, shard)
