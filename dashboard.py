@@ -236,6 +236,9 @@ def build_app(bus, store: DataStore):
     async def lifespan(_app):
         await store.start()
         yield
+        for t in (store._task, store._poll_task):
+            if t is not None:
+                t.cancel()
 
     app = FastAPI(title="ai-crypto-trader-amd dashboard",
                   lifespan=lifespan)
