@@ -131,3 +131,37 @@ def test_continuous_time_chunked_resume(dev):
     assert np.array_equal(m_cpu, m1)
     assert np.array_equal(m_cpu, m3)
     assert np.array_equal(m_cpu, m8)
+
+
+def test_continuous_random_shape_fuzz(dev):
+    """Property fuzz: random (T, nsym, P, nshards, time_groups, tail)
+    configurations must ALL be bitwise-equal to the sequential CPU
+    engine — shard alignment, partial words, padding lanes, chunked
+    resume and warm tails under one sweep."""
+    from ai_crypto_trader_amd.backtesting.engine_cpu import run_backtest_cpu
+    from ai_crypto_trader_amd.backtesting.strategy import (
+        RESNAP, random_population,
+    )
+    from ai_crypto_trader_amd.ops.backtest import run_backtest_continuous_gpu
+
+    rng = np.random.default_rng(77)
+    for case in range(6):
+        T = int(rng.integers(2_000, 60_000))
+        nsym = int(rng.integers(1, 4))
+        P = int(rng.integers(3, 70))
+        nshards = int(rng.integers(1, 5))
+        tgroups = int(rng.integers(1, 5))
+        tail = 256 * int(rng.integers(4, 9))
+        candles = _market(T, nsym, seed=1000 + case)
+        pop = random_population(P, seed=2000 + case)
+        m_cpu = run_backtest_cpu(candles, pop)
+        m_gpu = run_backtest_continuous_gpu(
+            torch.from_numpy(candles).to(dev),
+            torch.from_numpy(pop).to(dev),
+            nshards=nshards, time_groups=tgroups, tail=tail,
+        ).cpu().numpy()
+        assert np.array_equal(m_cpu, m_gpu), (
+            f"case {case}: T={T} nsym={nsym} P={P} nshards={nshards} "
+            f"tgroups={tgroups} tail={tail} RESNAP={RESNAP} "
+            f"maxdiff={np.max(np.abs(m_cpu - m_gpu))}"
+        )
