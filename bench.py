@@ -55,7 +55,8 @@ def parse_args():
                     help="skip the continuous (unsegmented) backtest loop")
     ap.add_argument("--segments", type=int, default=64,
                     help="fitness time-CV segments per symbol (64 measured "
-                         "fastest: 343 vs 324 G candles/s at 16)")
+                         "fastest; the continuous seg=1 case is measured "
+                         "separately via the time-parallel kernel pair)")
     return ap.parse_args()
 
 

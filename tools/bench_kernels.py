@@ -50,6 +50,14 @@ def bench_backtest(reps):
         print(json.dumps({"kernel": f"backtest_seg{seg}", "ms": dts * 1e3,
                           "candles_per_sec": P * nsym * T / dts}),
               flush=True)
+    # continuous (unsegmented) time-parallel pair (backtest_tp.hip)
+    from ai_crypto_trader_amd.ops.backtest import (
+        run_backtest_continuous_gpu,
+    )
+    dtc = timed(lambda: run_backtest_continuous_gpu(candles, pop), reps)
+    print(json.dumps({"kernel": "backtest_continuous(tp)",
+                      "ms": dtc * 1e3,
+                      "candles_per_sec": P * nsym * T / dtc}), flush=True)
     return {"kernel": "backtest", "ms": dt * 1e3,
             "candles_per_sec": P * nsym * T / dt,
             "config": {"nsym": nsym, "T": T, "P": P}}
