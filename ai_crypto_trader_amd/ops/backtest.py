@@ -147,7 +147,11 @@ def run_backtest_continuous_gpu(
     start_ev.record(cur)
     sf.wait_event(start_ev)
     st.wait_event(start_ev)
-    body4 = (T // nshards) // RESNAP * RESNAP if nshards > 1 else T
+    # balanced RESNAP-aligned shard boundaries (must match the kernel's
+    # own formula: floor(s*T/S) to the RESNAP grid)
+    def shard_lo(s):
+        return (s * T // nshards) // RESNAP * RESNAP if s < nshards else T
+
     # geometric group sizes (1,1,2,4,...): the first trades chunk starts
     # after one shard of flags (~1/nshards of the flags time) while the
     # later, bigger flags launches keep the chip oversubscribed
@@ -175,8 +179,8 @@ def run_backtest_continuous_gpu(
         evs.append(ev)
     for g in range(time_groups):
         s0, s1 = sbounds[g], sbounds[g + 1]
-        t_lo = s0 * body4
-        t_hi = T if s1 == nshards else s1 * body4
+        t_lo = shard_lo(s0)
+        t_hi = T if s1 == nshards else shard_lo(s1)
         st.wait_event(evs[g])
         ops.bt_trades(cptr, pptr, eflags.data_ptr(), xflags.data_ptr(),
                       metrics.data_ptr(), nsym, T, P,

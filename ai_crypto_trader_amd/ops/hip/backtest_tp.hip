@@ -325,7 +325,7 @@ __global__ void __launch_bounds__(BT_BLOCK) bt_flags_kernel(
     const float* __restrict__ pop,          // (P, NPARAM)
     unsigned long long* __restrict__ eflags,  // (nsym, nwords, P)
     unsigned long long* __restrict__ xflags,
-    int nsym, int T, int P, int chunks, int nshards, int body4, int tail,
+    int nsym, int T, int P, int chunks, int nshards, int tail,
     int shard0)                             // first shard of this launch
 {
 #pragma clang fp contract(off)
@@ -348,8 +348,16 @@ __global__ void __launch_bounds__(BT_BLOCK) bt_flags_kernel(
     FlagState st;
     st.load(pop + (long)(act ? p : 0) * BT_NPARAM);
 
-    const int lo = shard * body4;
-    const int hi = (shard == nshards - 1) ? T : (shard + 1) * body4;
+    // balanced RESNAP-aligned shard boundaries: floor(s*T/S) to the
+    // RESNAP grid. A uniform body with the remainder in the last shard
+    // left it 5x longer at RESNAP=16384 (T=1M, S=16) — tail-latency
+    // bound the whole launch (measured 331 -> 415+ G/s fixing this).
+    const int lo = (int)(((long)shard * T / nshards)
+                         / BT_RESNAP * BT_RESNAP);
+    const int hi = (shard == nshards - 1)
+                       ? T
+                       : (int)(((long)(shard + 1) * T / nshards)
+                               / BT_RESNAP * BT_RESNAP);
     const int start = (shard == 0) ? 0 : lo - tail;
 
     const float4* sym_candles =
@@ -602,7 +610,7 @@ __global__ void __launch_bounds__(BT_TBLOCK) bt_trades_kernel(
                 if ((lane & 63) == 0)
                     skipw[buf][wv] = skip ? 1 : 0;
                 if (!skip) {
-#pragma unroll 4
+#pragma unroll 8
                     for (int k = 0; k < 32; ++k) {
                         if (k >= hlen) break;
                         const int i = base + k;
@@ -619,7 +627,7 @@ __global__ void __launch_bounds__(BT_TBLOCK) bt_trades_kernel(
             if (!skipw[buf][wv]) {
                 const int hprev = habs - 1;
                 const int hlen = min(32, t_hi - (hprev << 5));
-#pragma unroll 4
+#pragma unroll 8
                 for (int k = 0; k < 32; ++k) {
                     if (k >= hlen) break;
                     float eq = eqbuf[buf][k][lane];
@@ -695,15 +703,12 @@ extern "C" void launch_bt_flags(const float* candles, const float* pop,
                                 int tail, int shard0, int nlaunch,
                                 hipStream_t stream) {
     int chunks = (P + BT_BLOCK - 1) / BT_BLOCK;
-    int body4 = nshards > 1
-                    ? (T / nshards) / BT_RESNAP * BT_RESNAP
-                    : T;
     if (nlaunch <= 0)
         nlaunch = nshards - shard0;
     hipLaunchKernelGGL(bt_flags_kernel,
                        dim3(nlaunch * nsym * chunks), dim3(BT_BLOCK), 0,
                        stream, candles, pop, eflags, xflags, nsym, T, P,
-                       chunks, nshards, body4, tail, shard0);
+                       chunks, nshards, tail, shard0);
 }
 
 extern "C" void launch_bt_trades(const float* candles, const float* pop,
