@@ -152,22 +152,11 @@ def run_backtest_continuous_gpu(
     def shard_lo(s):
         return (s * T // nshards) // RESNAP * RESNAP if s < nshards else T
 
-    # geometric group sizes (1,1,2,4,...): the first trades chunk starts
-    # after one shard of flags (~1/nshards of the flags time) while the
-    # later, bigger flags launches keep the chip oversubscribed
-    sizes = []
-    left, s = nshards, 1
-    while left > 0 and len(sizes) < time_groups - 1:
-        s = min(s, left)
-        sizes.append(s)
-        left -= s
-        s *= 2
-    if left > 0:
-        sizes.append(left)
-    sbounds = [0]
-    for s in sizes:
-        sbounds.append(sbounds[-1] + s)
-    time_groups = len(sizes)
+    # even shard groups: trades chunks cover equal time slices, so the
+    # tail chunk that runs after the last flags launch is 1/G of the
+    # timeline (geometric sizing was measured worse — it parks half the
+    # timeline behind the final flags group)
+    sbounds = [nshards * g // time_groups for g in range(time_groups + 1)]
     evs = []
     for g in range(time_groups):
         s0, s1 = sbounds[g], sbounds[g + 1]
