@@ -46,12 +46,13 @@ def run_backtest_gpu(
 _flag_cache: dict = {}
 
 
-def pick_nshards(nsym: int, T: int, P: int, *, target_blocks: int = 4096,
+def pick_nshards(nsym: int, T: int, P: int, *, target_blocks: int = 6144,
                  tail: int = 2048) -> int:
-    """Time shards for bt_flags: enough blocks to oversubscribe every CU
-    (measured sweep at 1024x64x1M: 8 shards = 399, 16 = 419, 32 = 394
-    Gcandles/s -> target 16 blocks/CU), bounded so each shard body is
-    >= one RESNAP period and >= 4x the warm tail."""
+    """Time shards for bt_flags: enough blocks to oversubscribe every CU,
+    bounded so each shard body is >= one RESNAP period and >= 4x the
+    warm tail. 24 shards x 8 even time groups measured best end-to-end
+    at the flagship shape (234 G/s; full sweep in
+    profiles/backtest_continuous_pmc.json)."""
     from ..backtesting.strategy import RESNAP
 
     chunks = (P + 255) // 256
