@@ -140,7 +140,11 @@ def run_backtest_continuous_gpu(
 
     ss = _pipe_streams.get(dev.index)
     if ss is None:
-        ss = (torch.cuda.Stream(dev), torch.cuda.Stream(dev))
+        # trades gets the high-priority stream: its serial chain paces
+        # the pipeline, so its waves should win scheduling over the
+        # issue-bound flags waves when co-resident
+        ss = (torch.cuda.Stream(dev),
+              torch.cuda.Stream(dev, priority=-1))
         _pipe_streams[dev.index] = ss
     sf, st = ss
     cur = torch.cuda.current_stream(dev)
