@@ -140,9 +140,9 @@ def run_backtest_continuous_gpu(
 
     ss = _pipe_streams.get(dev.index)
     if ss is None:
-        # trades gets the high-priority stream: its serial chain paces
-        # the pipeline, so its waves should win scheduling over the
-        # issue-bound flags waves when co-resident
+        # trades on the high-priority stream (measured neutral on
+        # MI355X — CU arbitration ignores stream priority — but it
+        # documents intent and costs nothing)
         ss = (torch.cuda.Stream(dev),
               torch.cuda.Stream(dev, priority=-1))
         _pipe_streams[dev.index] = ss
