@@ -60,11 +60,38 @@ class StrategySelectionService(Service):
         self.min_confidence = min_confidence
         self.current = "default"
         self.performance: dict[str, dict] = {}
+        # regime -> strategy -> {"pnl_sum", "n"} — the reference's
+        # per-regime strategy performance table that drives switching
+        # (market_regime_service.py:637-1113)
+        self.regime_performance: dict[str, dict[str, dict]] = {}
+        self._regime_now = "ranging"
         self.switches = 0
+
+    def record_trade(self, strategy: str, regime: str, pnl_pct: float):
+        """Feed a closed trade into the per-regime performance table."""
+        row = self.regime_performance.setdefault(regime, {}) \
+            .setdefault(strategy, {"pnl_sum": 0.0, "n": 0})
+        row["pnl_sum"] += pnl_pct
+        row["n"] += 1
+
+    def record_performance(self, strategy: str, perf: dict):
+        """Feed an evolution/backtest performance report (reference's
+        strategy_performance_{id} keys)."""
+        self.performance[strategy] = dict(perf)
 
     # --- factor scorers (:299-688) ---------------------------------------
     def score_regime(self, strat: str, regime: str) -> float:
-        return REGIME_FIT.get(regime, REGIME_FIT["ranging"]).get(strat, 0.3)
+        """Static regime fit blended with the MEASURED per-regime trade
+        performance once enough outcomes exist (the reference switches
+        on its regime-bucketed performance table)."""
+        fit = REGIME_FIT.get(regime, REGIME_FIT["ranging"]).get(strat, 0.3)
+        row = self.regime_performance.get(regime, {}).get(strat)
+        if not row or row["n"] < 3:
+            return fit
+        avg_pnl = row["pnl_sum"] / row["n"]        # pct per trade
+        measured = max(0.0, min(1.0, 0.5 + avg_pnl / 2.0))
+        w = min(row["n"] / 10.0, 0.6)              # up to 60% measured
+        return (1.0 - w) * fit + w * measured
 
     def score_history(self, strat: str) -> float:
         p = self.performance.get(strat)
@@ -141,11 +168,36 @@ class StrategySelectionService(Service):
         return imp >= self.min_improvement and \
             scores[best] >= self.min_confidence * max(scores.values())
 
+    def run_tasks(self):
+        return [self._consume_outcomes(), self.run()]
+
+    async def _consume_outcomes(self):
+        """Feed the per-regime performance table from closed trades and
+        evolution performance reports."""
+        sub = self.bus.subscribe(Channels.TRADE_EXECUTIONS,
+                                 Channels.STRATEGY_EVOLUTION_UPDATES)
+
+        def on_msg(chan, m):
+            if not isinstance(m, dict):
+                return
+            if chan == Channels.TRADE_EXECUTIONS:
+                if m.get("side") == "SELL" and "pnl_pct" in m:
+                    self.record_trade(self.current, self._regime_now,
+                                      float(m["pnl_pct"]))
+            else:
+                perf = m.get("performance")
+                if perf:
+                    self.record_performance(
+                        m.get("strategy_id", self.current), perf)
+
+        await self.consume(sub, on_msg)
+
     async def run(self):
         while self.running:
             regime_d = await self.bus.get_json(
                 Keys.CURRENT_MARKET_REGIME) or {}
             regime = regime_d.get("regime", "ranging")
+            self._regime_now = regime
             vol = regime_d.get("volatility", 0.5)
             sent = 0.5
             metrics = await self.bus.hgetall(Keys.SOCIAL_METRICS)

@@ -926,3 +926,35 @@ def test_runner_tcp_health_endpoint():
 
     h = asyncio.run(run())
     assert h["service"] == "model_registry" and h["healthy"]
+
+
+def test_regime_bucketed_performance_drives_selection():
+    """The per-regime strategy performance table (reference
+    market_regime_service.py:637-1113) blends into regime scoring:
+    measured losses in a regime demote the statically-favored strategy."""
+    from ai_crypto_trader_amd.bus.message_bus import InProcessBus
+    from ai_crypto_trader_amd.config import AppConfig
+    from ai_crypto_trader_amd.services.strategy_selection import (
+        StrategySelectionService,
+    )
+
+    sel = StrategySelectionService(InProcessBus(), AppConfig())
+    base = sel.score_regime("momentum", "bull")
+    assert base == 1.0                     # static fit favors momentum
+    # momentum keeps losing in bull markets
+    for _ in range(10):
+        sel.record_trade("momentum", "bull", -2.0)
+    demoted = sel.score_regime("momentum", "bull")
+    assert demoted < base - 0.2
+    # and consistent wins promote a statically-weak strategy
+    for _ in range(10):
+        sel.record_trade("mean_reversion", "bull", +2.0)
+    promoted = sel.score_regime("mean_reversion", "bull")
+    assert promoted > sel.score_regime("conservative", "bull")
+    # other regimes unaffected
+    assert sel.score_regime("momentum", "bear") == \
+        __import__("ai_crypto_trader_amd.services.strategy_selection",
+                   fromlist=["REGIME_FIT"]).REGIME_FIT["bear"]["momentum"]
+    # evolution performance reports feed score_history
+    sel.record_performance("momentum", {"sharpe": 2.0})
+    assert sel.score_history("momentum") == 1.0
