@@ -124,3 +124,78 @@ def test_resnap_keeps_engine_consistent():
     assert np.isfinite(m).all()
     m2 = run_backtest_cpu(candles, pop)
     assert np.array_equal(m, m2)
+
+
+# ---------------------------------------------------------------------------
+# Property-based invariants (hypothesis): the sequential engine's
+# financial invariants must hold for ANY parameter set and market.
+# ---------------------------------------------------------------------------
+try:
+    from hypothesis import given, settings
+    from hypothesis import strategies as hst
+    HAVE_HYP = True
+except ImportError:        # pragma: no cover
+    HAVE_HYP = False
+
+if HAVE_HYP:
+    @settings(max_examples=12, deadline=None)
+    @given(seed=hst.integers(0, 10_000), pop_seed=hst.integers(0, 10_000),
+           T=hst.integers(300, 2500))
+    def test_engine_invariants_hold_for_any_market(seed, pop_seed, T):
+        candles = candles_chl_v(generate_ohlcv(T, 1, seed=seed))
+        pop = random_population(6, seed=pop_seed)
+        m, curves = run_backtest_cpu(candles, pop, record_equity=True)
+        eq = curves[:, 0]
+        # equity stays positive and finite at every candle
+        assert np.isfinite(eq).all() and (eq > 0).all()
+        # wins never exceed trades; gross sums non-negative
+        assert (m[..., 2] <= m[..., 1] + 1e-6).all()
+        assert (m[..., 3] >= 0).all() and (m[..., 4] >= 0).all()
+        # drawdown is a fraction of peak
+        assert ((m[..., 5] >= 0) & (m[..., 5] < 1.0)).all()
+        # final equity consistent with the recorded curve
+        np.testing.assert_allclose(m[:, 0, 0], eq[:, -1] if eq.ndim > 1
+                                   else curves[:, 0, -1], rtol=1e-6)
+
+    @settings(max_examples=8, deadline=None)
+    @given(seed=hst.integers(0, 10_000))
+    def test_fees_never_create_money(seed):
+        """Zero-fee equity >= with-fee equity for the same decisions is
+        NOT guaranteed (different fills), but a round-trip at constant
+        price must lose exactly the fees."""
+        candles = candles_chl_v(generate_ohlcv(400, 1, seed=seed))
+        candles[0, :, 0] = 1.0          # constant close
+        candles[0, :, 1] = 1.0
+        candles[0, :, 2] = 1.0
+        p = dict_to_params({"entry_votes": 1, "exit_votes": 1,
+                            "stop_loss_pct": 0.2,
+                            "take_profit_pct": 0.4,
+                            "position_size_pct": 1.0})
+        m = run_backtest_cpu(candles, clip_params(p[None, :]))
+        # flat market: equity can only be <= initial (fees) and >= the
+        # double-fee floor per trade
+        fe = m[0, 0, 0]
+        n = m[0, 0, 1]
+        assert fe <= 1.0 + 1e-6
+        assert fe >= (1 - 0.001) ** (2 * max(n, 1)) - 1e-4
+
+
+if HAVE_HYP:
+    @settings(max_examples=6, deadline=None)
+    @given(seed=hst.integers(0, 10_000), pop_seed=hst.integers(0, 10_000))
+    def test_flag_decomposition_bitwise_any_seed(seed, pop_seed):
+        """The flags->trades decomposition (the continuous GPU path's
+        foundation) is bitwise for ANY market/population, not just the
+        fixed-seed case."""
+        from ai_crypto_trader_amd.backtesting.engine_cpu import (
+            run_trades_from_flags_cpu,
+        )
+
+        candles = candles_chl_v(generate_ohlcv(1200, 1, seed=seed))
+        pop = random_population(8, seed=pop_seed)
+        m_ref, nets = run_backtest_cpu(candles, pop, record_net=True)
+        entry_v = pop[:, 10].astype(np.int32)[:, None, None]
+        exit_v = pop[:, 11].astype(np.int32)[:, None, None]
+        m_flags = run_trades_from_flags_cpu(
+            candles, pop, nets >= entry_v, nets <= -exit_v)
+        assert np.array_equal(m_ref, m_flags)
