@@ -199,3 +199,26 @@ if HAVE_HYP:
         m_flags = run_trades_from_flags_cpu(
             candles, pop, nets >= entry_v, nets <= -exit_v)
         assert np.array_equal(m_ref, m_flags)
+
+
+if HAVE_HYP:
+    @settings(max_examples=10, deadline=None)
+    @given(seed=hst.integers(0, 10_000), gen=hst.integers(0, 50))
+    def test_ga_evolution_invariants_any_seed(seed, gen):
+        """ga_evolve_cpu: children in bounds, integer slots integral,
+        ema_slow > ema_fast, elites carried verbatim — for any seed."""
+        from ai_crypto_trader_amd.backtesting.strategy import PARAM_BOUNDS
+        from ai_crypto_trader_amd.ops.ga import ga_evolve_cpu
+
+        pop = random_population(64, seed=seed)
+        fitness = np.linspace(1.0, -1.0, 64).astype(np.float32)
+        child = ga_evolve_cpu(pop, fitness, elite_k=4, tournament=3,
+                              cx_rate=0.5, mut_rate=0.3, mut_scale=0.2,
+                              seed=seed, gen=gen)
+        lo, hi = PARAM_BOUNDS[:, 0], PARAM_BOUNDS[:, 1]
+        assert (child >= lo - 1e-6).all() and (child <= hi + 1e-6).all()
+        is_int = PARAM_BOUNDS[:, 2] > 0
+        assert np.allclose(child[:, is_int], np.rint(child[:, is_int]))
+        assert (child[:, 4] > child[:, 3]).all()      # ema_slow > fast
+        # elites = the top-fitness individuals, unchanged
+        np.testing.assert_array_equal(child[:4], pop[:4])
