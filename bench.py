@@ -16,9 +16,12 @@ Run (driver contract):
       --master-addr 127.0.0.1 bench.py --gpus 8 --steps 5 --warmup 2
 
 Synthetic data (seeded GBM, data/synthetic.py), random-init population.
-Also measures the Monte-Carlo VaR path rate (10M correlated-GBM paths x 64
-assets per GPU) outside the timed GA loop and reports it under
-"secondary".
+Outside the timed GA loop the same run also measures (and reports under
+"config"): the CONTINUOUS unsegmented backtest via the time-parallel
+kernel pair (all ranks), and under "secondary": the Monte-Carlo VaR path
+rate (10M correlated-GBM paths x 64 assets), the LSTM-predictor train
+rate (BASELINE #2) and the PPO train rate (BASELINE #4) — so the
+driver's clock attests every BASELINE config.
 """
 
 from __future__ import annotations
