@@ -111,6 +111,9 @@ class DCAConfig(BaseModel):
 
 
 class SocialConfig(BaseModel):
+    # metrics source: synthetic (offline) | lunarcrush (live, needs
+    # LUNARCRUSH_API_KEY)
+    source: str = "synthetic"
     update_interval_s: float = 300.0
     sentiment_half_life_h: float = 6.0        # social_risk_adjuster half-life
     source_weights: dict[str, float] = Field(
@@ -149,6 +152,9 @@ class OrderBookConfig(BaseModel):
 
 
 class NewsConfig(BaseModel):
+    # headline source: synthetic (offline) | live (CryptoPanic + RSS +
+    # LunarCrush; keys optional)
+    source: str = "synthetic"
     interval_s: float = 2.0                   # reference: 300 s against live feeds
     max_items_per_source: int = 20
 

@@ -74,7 +74,14 @@ def build_service(name: str, bus, cfg: AppConfig, args):
         return TradeExecutorService(bus, exchange(), cfg)
     if name == "social_monitor":
         from .social import SocialMonitorService
-        return SocialMonitorService(bus, cfg)
+        src = None
+        if getattr(cfg.social, "source", "synthetic") == "lunarcrush":
+            import os
+
+            from ..live.social_news import LunarCrushSocialSource
+            src = LunarCrushSocialSource(
+                api_key=os.environ.get("LUNARCRUSH_API_KEY", ""))
+        return SocialMonitorService(bus, cfg, source=src)
     if name == "enhanced_social":
         from .social import EnhancedSocialMonitorService
         return EnhancedSocialMonitorService(bus, cfg)
@@ -83,7 +90,15 @@ def build_service(name: str, bus, cfg: AppConfig, args):
         return SocialRiskAdjuster(bus, cfg)
     if name == "news_analysis":
         from .news import NewsAnalysisService
-        return NewsAnalysisService(bus, cfg)
+        src = None
+        if getattr(cfg.news, "source", "synthetic") == "live":
+            import os
+
+            from ..live.social_news import LiveNewsHeadlines
+            src = LiveNewsHeadlines(
+                cryptopanic_key=os.environ.get("CRYPTOPANIC_API_KEY", ""),
+                lunarcrush_key=os.environ.get("LUNARCRUSH_API_KEY", ""))
+        return NewsAnalysisService(bus, cfg, source=src)
     if name == "order_book":
         from .order_book import OrderBookAnalysisService
         return OrderBookAnalysisService(bus, exchange(), cfg)
