@@ -9,7 +9,7 @@ Examples:
   python run_backtest.py backtest --symbol BTCUSDC --strategy momentum --plot
   python run_backtest.py optimize --symbol BTCUSDC --pop 256 --generations 10
   python run_backtest.py list
-  python run_backtest.py analyze
+  python run_backtest.py analyze [--metric sharpe_ratio]
 """
 
 from __future__ import annotations
@@ -57,7 +57,13 @@ def main():
     o.add_argument("--cpu", action="store_true")
 
     sub.add_parser("list", help="list stored data and results")
-    sub.add_parser("analyze", help="summary report over stored results")
+    an = sub.add_parser("analyze",
+                        help="summary report over stored results")
+    an.add_argument("--metric", default=None,
+                    help="rank results by this metric only (reference "
+                         "run_backtest.py `analyze --metric "
+                         "sharpe_ratio`); aliases: sharpe_ratio, "
+                         "total_return, win_rate, profit_factor")
 
     args = ap.parse_args()
     dm = HistoricalDataManager(args.data_dir)
@@ -109,10 +115,25 @@ def main():
         return
 
     if args.cmd == "analyze":
+        _METRIC_ALIASES = {"sharpe_ratio": "sharpe",
+                           "total_return": "total_return_pct",
+                           "return": "total_return_pct"}
         eng = BacktestEngine(args.data_dir, device="cpu")
         results = eng.list_results()
         ra = ResultAnalyzer(f"{args.data_dir}/analysis")
-        print(json.dumps(ra.summary_report(results), indent=2))
+        report = ra.summary_report(results)
+        if args.metric:
+            m = _METRIC_ALIASES.get(args.metric, args.metric)
+            best = report.get("best_by", {}).get(m)
+            if best is None:
+                known = sorted(report.get("best_by", {}))
+                raise SystemExit(f"unknown metric {args.metric!r} "
+                                 f"(known: {known})")
+            print(json.dumps({"metric": m, "best": best,
+                              "n_results": report.get("n_results")},
+                             indent=2))
+        else:
+            print(json.dumps(report, indent=2))
         if results:
             p = ra.comparison_chart(results)
             print(f"comparison chart -> {p}")
