@@ -958,3 +958,30 @@ def test_regime_bucketed_performance_drives_selection():
     # evolution performance reports feed score_history
     sel.record_performance("momentum", {"sharpe": 2.0})
     assert sel.score_history("momentum") == 1.0
+
+
+def test_serving_endpoints_cpu_smoke(tmp_path):
+    """serve.py's ASGI app responds on CPU: health, analyze, and a
+    backtest request (the GPU latency profile is
+    profiles/serve_latency.json)."""
+    import pytest as _pytest
+    _pytest.importorskip("fastapi")
+    from fastapi.testclient import TestClient
+
+    from serve import build_server
+
+    app, _nn = build_server(str(tmp_path), device="cpu")
+    with TestClient(app) as c:
+        h = c.get("/healthz").json()
+        assert h["ok"] and h["device"] == "cpu"
+        r = c.post("/analyze", json={
+            "symbol": "BTCUSDC", "current_price": 100.0, "rsi": 25.0,
+            "stoch_k": 15.0, "williams_r": -85.0, "macd": 0.5,
+            "trend": "uptrend", "trend_strength": 40.0,
+            "price_change_1m": 0.4, "price_change_5m": 1.0,
+            "price_change_15m": 2.0, "bb_position": 0.1,
+        })
+        assert r.status_code == 200
+        out = r.json()
+        assert out["decision"] in ("BUY", "SELL", "HOLD")
+        assert 0 <= out["confidence"] <= 1
