@@ -208,10 +208,28 @@ class StrategyEvolutionService(Service):
             p[i] *= mult
         return clip_params(p[None])[0]
 
+    def similar_version(self, params: np.ndarray,
+                        threshold: float = 0.95) -> str | None:
+        """Model-version similarity dedup (reference :1295-1400): a new
+        parameter set within `threshold` cosine-style similarity (1 -
+        normalized L2 over the bounds span) of an existing version
+        reuses that version's id instead of minting a new one."""
+        from ..backtesting.strategy import PARAM_BOUNDS, dict_to_params
+
+        span = (PARAM_BOUNDS[:, 1] - PARAM_BOUNDS[:, 0]).astype(np.float64)
+        for v in reversed(self.model_versions[-50:]):
+            other = dict_to_params(v["params"]).astype(np.float64)
+            d = np.abs(params.astype(np.float64) - other) / span
+            sim = 1.0 - float(np.sqrt((d ** 2).mean()))
+            if sim >= threshold:
+                return v["id"]
+        return None
+
     async def hot_swap(self, params: np.ndarray, perf: dict, regime: str):
         """strategy_params key + 'reload' publish (reference :349-362)."""
         self.current_params = params
-        sid = f"evolved-{uuid.uuid4().hex[:8]}"
+        sid = self.similar_version(params) \
+            or f"evolved-{uuid.uuid4().hex[:8]}"
         old = self.current_strategy_id
         self.current_strategy_id = sid
         await self.bus.set(Keys.STRATEGY_PARAMS, params_to_dict(params))
@@ -220,13 +238,25 @@ class StrategyEvolutionService(Service):
             Channels.STRATEGY_EVOLUTION_UPDATES,
             EvolutionUpdate(sid, params_to_dict(params), perf,
                             market_regime=regime).to_dict())
-        self.model_versions.append({
-            "id": sid, "previous": old, "params": params_to_dict(params),
-            "performance": perf, "at": time.time(),
-        })
-        await self.bus.publish(Channels.MODEL_REGISTRY_EVENTS, {
-            "event": "strategy_registered", "id": sid,
-        })
+        existing = next((v for v in self.model_versions
+                         if v["id"] == sid), None)
+        if existing is not None:
+            # near-duplicate params: refresh the existing version's
+            # performance rather than minting a new entry (:1295-1400)
+            existing["performance"] = perf
+            existing["at"] = time.time()
+            await self.bus.publish(Channels.MODEL_REGISTRY_EVENTS, {
+                "event": "strategy_version_reused", "id": sid,
+            })
+        else:
+            self.model_versions.append({
+                "id": sid, "previous": old,
+                "params": params_to_dict(params),
+                "performance": perf, "at": time.time(),
+            })
+            await self.bus.publish(Channels.MODEL_REGISTRY_EVENTS, {
+                "event": "strategy_registered", "id": sid,
+            })
 
     async def evolve_once(self) -> dict:
         regime_d = await self.bus.get_json(Keys.CURRENT_MARKET_REGIME) or {}

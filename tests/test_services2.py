@@ -985,3 +985,51 @@ def test_serving_endpoints_cpu_smoke(tmp_path):
         out = r.json()
         assert out["decision"] in ("BUY", "SELL", "HOLD")
         assert 0 <= out["confidence"] <= 1
+
+
+def test_strategy_version_similarity_dedup():
+    """Near-duplicate evolved parameter sets reuse the existing version
+    id (reference model-version similarity dedup,
+    strategy_evolution_service.py:1295-1400); distinct sets register
+    new versions."""
+    import asyncio
+
+    import numpy as np
+
+    from ai_crypto_trader_amd.backtesting.strategy import (
+        DEFAULT_PARAMS, clip_params,
+    )
+    from ai_crypto_trader_amd.bus.message_bus import InProcessBus
+    from ai_crypto_trader_amd.config import AppConfig
+    from ai_crypto_trader_amd.data.synthetic import (
+        candles_chl_v, generate_ohlcv,
+    )
+    from ai_crypto_trader_amd.services.strategy_evolution import (
+        StrategyEvolutionService,
+    )
+
+    candles = candles_chl_v(generate_ohlcv(2000, 1, seed=0))
+    svc = StrategyEvolutionService(InProcessBus(), AppConfig(),
+                                   candles=candles)
+
+    async def run():
+        p1 = DEFAULT_PARAMS.copy()
+        await svc.hot_swap(p1, {"sharpe": 1.0}, "bull")
+        n1 = len(svc.model_versions)
+        # tiny perturbation -> same version reused
+        p2 = p1.copy()
+        p2[1] += 0.01
+        await svc.hot_swap(clip_params(p2[None])[0],
+                           {"sharpe": 1.1}, "bull")
+        assert len(svc.model_versions) == n1
+        assert svc.model_versions[-1]["performance"]["sharpe"] == 1.1
+        # clearly different params -> new version
+        p3 = p1.copy()
+        p3[0] = 60.0
+        p3[13] = 0.19
+        p3[12] = 0.06
+        await svc.hot_swap(clip_params(p3[None])[0],
+                           {"sharpe": 0.5}, "bear")
+        assert len(svc.model_versions) == n1 + 1
+
+    asyncio.run(run())
