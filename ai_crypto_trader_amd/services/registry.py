@@ -219,8 +219,25 @@ class FeatureImportanceAnalyzer(Service):
                 "categories_to_prioritize": cat_sorted[:2],
                 "categories_to_reconsider": cat_sorted[-1:],
             },
+            # pruned "optimized model" on the top features only
+            # (reference feature_importance_analyzer.py:550): shows how
+            # much predictive power the top-5 retain vs the full set
+            "optimized_model": self._pruned_model(X, y, rf, top[:5]),
         }
         return self.report
+
+    def _pruned_model(self, X, y, full_rf, keep: list[str]) -> dict:
+        from sklearn.ensemble import RandomForestRegressor
+
+        idx = [self.FEATURES.index(f) for f in keep]
+        pruned = RandomForestRegressor(
+            n_estimators=40, random_state=0, max_depth=6,
+        ).fit(X[:, idx], y)
+        return {
+            "features": keep,
+            "score_full": round(float(full_rf.score(X, y)), 4),
+            "score_pruned": round(float(pruned.score(X[:, idx], y)), 4),
+        }
 
     async def run(self):
         sub = self.bus.subscribe(Channels.TRADING_SIGNALS,
