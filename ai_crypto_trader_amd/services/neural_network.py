@@ -77,6 +77,9 @@ class NeuralNetworkService(Service):
         self.device = torch.device(device)
         self.models: dict[str, LSTMPricePredictor] = {}
         self.scalers: dict[str, MinMaxScaler] = {}
+        # (sym, regime) -> snapshot (reference :1445-1473)
+        self.regime_models: dict[tuple, LSTMPricePredictor] = {}
+        self.regime_scalers: dict[tuple, MinMaxScaler] = {}
         self.val_loss: dict[str, float] = {}
         self.candles: dict[str, list] = {}
         self.trained = 0
@@ -149,6 +152,30 @@ class NeuralNetworkService(Service):
         self.trained += 1
         return best_val
 
+    def snapshot_for_regime(self, sym: str, regime: str):
+        """Regime-specific model snapshot (reference
+        neural_network_service.py:1445-1473): keep a copy of the
+        current trained model tagged with the regime it was trained
+        under; predict() prefers the snapshot matching the current
+        regime."""
+        import copy
+
+        model = self.models.get(sym)
+        if model is None:
+            return None
+        key = (sym, regime)
+        self.regime_models[key] = copy.deepcopy(model)
+        self.regime_scalers[key] = self.scalers.get(sym)
+        return key
+
+    def model_for(self, sym: str, regime: str | None = None):
+        """Current-regime snapshot if present, else the live model."""
+        if regime is not None:
+            key = (sym, regime)
+            if key in self.regime_models:
+                return self.regime_models[key], self.regime_scalers[key]
+        return self.models.get(sym), self.scalers.get(sym)
+
     def save(self, directory: str | None = None):
         d = Path(directory or self.config.neural_network.model_dir)
         d.mkdir(parents=True, exist_ok=True)
@@ -169,9 +196,9 @@ class NeuralNetworkService(Service):
         return n
 
     # --- prediction ------------------------------------------------------
-    def predict(self, sym: str, candles: np.ndarray) -> dict | None:
-        model = self.models.get(sym)
-        scaler = self.scalers.get(sym)
+    def predict(self, sym: str, candles: np.ndarray,
+                regime: str | None = None) -> dict | None:
+        model, scaler = self.model_for(sym, regime)
         if model is None or scaler is None:
             return None
         cfg = self.config.neural_network

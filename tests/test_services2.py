@@ -1033,3 +1033,40 @@ def test_strategy_version_similarity_dedup():
         assert len(svc.model_versions) == n1 + 1
 
     asyncio.run(run())
+
+
+def test_nn_regime_specific_model_snapshots():
+    """Regime-tagged model snapshots (reference
+    neural_network_service.py:1445-1473): predict() prefers the model
+    trained under the matching regime; unknown regimes fall back to the
+    live model."""
+    import numpy as np
+
+    from ai_crypto_trader_amd.bus.message_bus import InProcessBus
+    from ai_crypto_trader_amd.config import AppConfig
+    from ai_crypto_trader_amd.data.synthetic import (
+        candles_chl_v, generate_ohlcv,
+    )
+    from ai_crypto_trader_amd.services.neural_network import (
+        NeuralNetworkService,
+    )
+
+    cfg = AppConfig()
+    cfg.neural_network.seq_len = 16
+    svc = NeuralNetworkService(InProcessBus(), cfg)
+    candles = candles_chl_v(generate_ohlcv(500, 1, seed=1))[0]
+    svc.train("BTCUSDC", candles, epochs=1)
+    key = svc.snapshot_for_regime("BTCUSDC", "bull")
+    assert key == ("BTCUSDC", "bull")
+    # retrain changes the live model; the bull snapshot is frozen
+    svc.train("BTCUSDC", candles, epochs=1)
+    m_live, _ = svc.model_for("BTCUSDC")
+    m_bull, _ = svc.model_for("BTCUSDC", "bull")
+    assert m_bull is not m_live
+    p_bull = svc.predict("BTCUSDC", candles, regime="bull")
+    p_live = svc.predict("BTCUSDC", candles)
+    assert p_bull is not None and p_live is not None
+    # unknown regime -> live model path
+    p_unknown = svc.predict("BTCUSDC", candles, regime="sideways")
+    assert np.isclose(p_unknown["predicted_change_pct"],
+                      p_live["predicted_change_pct"])
