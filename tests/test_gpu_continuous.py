@@ -145,10 +145,12 @@ def test_continuous_random_shape_fuzz(dev):
     from ai_crypto_trader_amd.ops.backtest import run_backtest_continuous_gpu
 
     rng = np.random.default_rng(77)
-    for case in range(6):
+    for case in range(7):
         T = int(rng.integers(2_000, 60_000))
         nsym = int(rng.integers(1, 4))
-        P = int(rng.integers(3, 70))
+        # last case: multi-chunk population (P > 256 exercises the
+        # chunk indexing in both kernels)
+        P = 300 if case == 6 else int(rng.integers(3, 70))
         nshards = int(rng.integers(1, 5))
         tgroups = int(rng.integers(1, 5))
         tail = 256 * int(rng.integers(4, 9))
