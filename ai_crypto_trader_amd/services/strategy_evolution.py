@@ -254,6 +254,7 @@ class StrategyEvolutionService(Service):
                 "params": params_to_dict(params),
                 "performance": perf, "at": time.time(),
             })
+            del self.model_versions[:-200]      # bounded history
             await self.bus.publish(Channels.MODEL_REGISTRY_EVENTS, {
                 "event": "strategy_registered", "id": sid,
             })
